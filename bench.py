@@ -1,0 +1,130 @@
+"""Flagship benchmark — LeNet-5 training step throughput (BASELINE.json).
+
+Single GPU:   python bench.py --gpus 1 --steps 200 --warmup 20
+Multi-GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Per-rank batch is 64 (the BASELINE.json config); weak scaling (per-GPU work
+fixed).  Synthetic 28x28x1 data, random-init weights (the reference's MNIST
+image blobs are absent and the baseline is defined on synthetic data).
+Timing: W untimed warmup steps, then exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX time over ranks;
+rank 0 prints one JSON line.
+
+Reference headline being compared against: the CUDA variant's
+2,996.99 ms / 60k-sample epoch on a T4 ~= 20,020 images/sec.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+BASELINE_IMAGES_PER_SEC = 60000.0 / 2.9969857  # BASELINE.md CUDA headline
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--batch-size", type=int, default=64,
+                   help="per-GPU batch (BASELINE config: 64)")
+    p.add_argument("--act-dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--device", type=str, default="auto")
+    args = p.parse_args()
+
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.data.mnist import synthetic_mnist
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.parallel import dist as pdist
+
+    cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
+                      device=args.device, log_interval=0, data="synthetic")
+    device = cfg.resolved_device()
+    ctx = pdist.init_from_env(device)
+    n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
+    if ctx.world_size == 1 and args.gpus > 1:
+        raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
+
+    trainer = Trainer(cfg, ctx=ctx)
+    B = args.batch_size
+
+    # Device-resident synthetic epoch pool (no H2D inside the timed loop; the
+    # pool is one epoch's worth of batches, cycled).
+    n_pool_batches = min(args.steps + args.warmup, 60000 // max(1, B)) or 1
+    x_host, y_host = synthetic_mnist(n_pool_batches * B,
+                                     seed=1234 + ctx.rank, structured=False)
+    xs, ys = [], []
+    for i in range(n_pool_batches):
+        xb, yb = trainer.stage_batch(x_host[i * B:(i + 1) * B],
+                                     y_host[i * B:(i + 1) * B])
+        xs.append(xb)
+        ys.append(yb)
+    if device == "cuda":
+        torch.cuda.synchronize()
+
+    def run(n_steps: int):
+        for s in range(n_steps):
+            i = s % n_pool_batches
+            trainer.step(xs[i], ys[i])
+
+    run(args.warmup)
+    pdist.barrier()
+    if device == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    run(args.steps)
+    if device == "cuda":
+        torch.cuda.synchronize()
+    pdist.barrier()
+    if device == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if ctx.world_size > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device == "cuda" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    global_batch = B * n_gpus
+    images_per_sec = args.steps * global_batch / elapsed
+    dtype = args.act_dtype if device == "cuda" else "fp32"
+    result = {
+        "metric": "training images/sec (whole node)",
+        "value": images_per_sec,
+        "unit": "images/sec",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": images_per_sec / BASELINE_IMAGES_PER_SEC,
+        "dtype": dtype,
+        "data": "synthetic",
+        "config": {
+            "model": "LeNet-5 28x28x1 (conv6x5x5 -> trainable-pool4x4 -> fc216x10)",
+            "global_batch": global_batch,
+            "per_gpu_batch": B,
+            "input": "28x28x1",
+            "parallelism": f"dp{n_gpus}",
+            "backend": trainer.backend,
+        },
+    }
+    if ctx.is_main:
+        print(json.dumps(result), flush=True)
+    pdist.barrier()
+    if pdist.is_distributed():
+        torch.distributed.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,235 @@
+"""Training/eval driver — the framework equivalent of the reference's
+Main.cpp `learn()` / `test()` / `classify()`.
+
+Backends:
+  * "hip"      — the gfx950 HIP kernels (3 fused launches per step);
+  * "cpu"      — the native C++ reference ops (threaded, race-free);
+  * "torchref" — the pure-PyTorch fp32 oracle (tests/debugging).
+
+A training step is: fused fwd+bwd-data -> weight-grad -> [RCCL all-reduce of
+the flat gradient bucket] -> fused SGD update.  Loss (sum over samples of
+||onehot - y||_2, the reference's metric) accumulates device-side and is
+only synced on readout.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ..config import TrainConfig
+from ..models.lenet import LeNet5
+from ..ops import native, torch_ref
+from ..ops import shapes as S
+from ..parallel import dist as pdist
+
+MODE_TRAIN, MODE_EVAL, MODE_INFER = 0, 1, 2
+
+
+class Workspace:
+    """Persistent activation / gradient-data buffers for a max batch size."""
+
+    def __init__(self, max_batch: int, device: torch.device,
+                 act_dtype: torch.dtype):
+        B = max_batch
+        dev = device
+        self.max_batch = B
+        self.act_dtype = act_dtype
+        self.a1 = torch.empty(B, S.C1_OUT, dtype=act_dtype, device=dev)
+        self.a2 = torch.empty(B, S.S1_OUT, dtype=act_dtype, device=dev)
+        self.y = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=dev)
+        self.dz = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=dev)
+        self.dz2 = torch.empty(B, S.S1_OUT, dtype=torch.float32, device=dev)
+        self.dz1 = torch.empty(B, S.C1_OUT, dtype=torch.float32, device=dev)
+        self.loss_accum = torch.zeros(1, dtype=torch.float32, device=dev)
+        self.correct_accum = torch.zeros(1, dtype=torch.int32, device=dev)
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, model: Optional[LeNet5] = None,
+                 ctx: Optional[pdist.DistContext] = None,
+                 max_batch: Optional[int] = None):
+        self.cfg = cfg
+        self.ctx = ctx or pdist.DistContext()
+        self.device = torch.device(cfg.resolved_device())
+        self.backend = cfg.resolved_backend()
+        if self.backend == "hip" and self.device.type != "cuda":
+            raise RuntimeError("hip backend requires a GPU device")
+        if self.backend in ("hip", "cpu"):
+            self._C = native.require()
+        self.model = model or LeNet5(self.device, seed=cfg.seed)
+        act_dtype = (torch.bfloat16
+                     if (cfg.act_dtype == "bf16" and self.backend == "hip")
+                     else torch.float32)
+        self.act_dtype = act_dtype
+        self.ws = Workspace(max_batch or cfg.batch_size, self.device,
+                            act_dtype)
+        self._loss_host = 0.0       # cpu/torchref backends accumulate here
+        self._samples_seen = 0
+        self.global_step = 0
+
+    # ------------------------------------------------------------------ util
+    def _update_scale(self, local_batch: int) -> float:
+        if self.cfg.grad_reduction == "mean":
+            return 1.0 / float(local_batch * self.ctx.world_size)
+        return 1.0
+
+    def stage_batch(self, x: torch.Tensor, labels: torch.Tensor
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Move a host fp32 batch to the execution device/dtype."""
+        if self.backend == "hip":
+            xd = x.to(self.device, dtype=self.act_dtype, non_blocking=True)
+            ld = labels.to(self.device, dtype=torch.int32, non_blocking=True)
+        else:
+            xd = x.to(torch.float32)
+            ld = labels.to(torch.int64)
+        return xd, ld
+
+    # ------------------------------------------------------------------ step
+    def step(self, x: torch.Tensor, labels: torch.Tensor) -> None:
+        """One training step on an already-staged batch."""
+        B = x.shape[0]
+        assert B <= self.ws.max_batch
+        m, w = self.model, self.ws
+        scale = self._update_scale(B)
+        if self.backend == "hip":
+            stream = native.current_stream_handle()
+            self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
+                               w.dz1, labels, w.loss_accum, w.correct_accum,
+                               B, MODE_TRAIN, stream)
+            self._C.hip_wgrad(x, w.a1, w.a2, w.dz, w.dz2, w.dz1, m.grads, B,
+                              stream)
+            pdist.allreduce_grads(m.grads)
+            self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
+                               native.current_stream_handle())
+        elif self.backend == "cpu":
+            # CPU path keeps fp32 activations in the workspace directly.
+            a1 = self._cpu_view(w.a1, B)
+            a2 = self._cpu_view(w.a2, B)
+            self._C.cpu_forward(x, m.params, a1, a2, w.y[:B])
+            loss = self._C.cpu_backward(x, m.params, a1, a2, w.y[:B], labels,
+                                        w.dz[:B], w.dz2[:B], w.dz1[:B],
+                                        m.grads)
+            self._loss_host += loss
+            pdist.allreduce_grads(m.grads)
+            self._C.cpu_update(m.params, m.grads, self.cfg.dt, scale)
+        else:  # torchref
+            a1, a2, y = torch_ref.forward(x, m.params)
+            dz, dz2, dz1, grads, loss = torch_ref.backward(
+                x, m.params, a1, a2, y, labels)
+            self._loss_host += loss
+            m.grads += grads
+            pdist.allreduce_grads(m.grads)
+            torch_ref.update(m.params, m.grads, self.cfg.dt, scale)
+        self._samples_seen += B * self.ctx.world_size
+        self.global_step += 1
+
+    def _cpu_view(self, t: torch.Tensor, B: int) -> torch.Tensor:
+        assert t.dtype == torch.float32, \
+            "cpu backend requires fp32 activation workspace"
+        return t[:B]
+
+    # ------------------------------------------------------------ loss/eval
+    def consume_loss(self) -> Tuple[float, int]:
+        """Returns (sum of per-sample error norms across ranks, samples seen)
+        since the last call.  Syncs the device."""
+        if self.backend == "hip":
+            local = float(self.ws.loss_accum.item())
+            self.ws.loss_accum.zero_()
+        else:
+            local = self._loss_host
+            self._loss_host = 0.0
+        dev = self.device if self.device.type == "cuda" else None
+        total = pdist.allreduce_scalar(local, device=dev)
+        n = self._samples_seen
+        self._samples_seen = 0
+        return total, n
+
+    @torch.no_grad()
+    def evaluate(self, x: torch.Tensor, labels: torch.Tensor,
+                 batch_size: Optional[int] = None) -> float:
+        """Error rate (%) over a dataset, sharded across ranks."""
+        bs = batch_size or self.ws.max_batch
+        n = x.shape[0]
+        correct = 0
+        w = self.ws
+        if self.backend == "hip":
+            w.correct_accum.zero_()
+        # contiguous rank shard
+        per = (n + self.ctx.world_size - 1) // self.ctx.world_size
+        lo, hi = self.ctx.rank * per, min(n, (self.ctx.rank + 1) * per)
+        for i in range(lo, hi, bs):
+            xb, lb = self.stage_batch(x[i:i + bs], labels[i:i + bs])
+            B = xb.shape[0]
+            if self.backend == "hip":
+                self._C.hip_fwdbwd(xb, self.model.params, w.a1, w.a2, w.y,
+                                   w.dz, w.dz2, w.dz1, lb, w.loss_accum,
+                                   w.correct_accum, B, MODE_EVAL,
+                                   native.current_stream_handle())
+            else:
+                preds = self._forward_preds(xb, B)
+                correct += int((preds == lb).sum().item())
+        if self.backend == "hip":
+            correct = int(w.correct_accum.item())
+            w.correct_accum.zero_()
+        dev = self.device if self.device.type == "cuda" else None
+        correct = int(pdist.allreduce_scalar(float(correct), device=dev))
+        return 100.0 * (1.0 - correct / float(n))
+
+    def _forward_preds(self, xb: torch.Tensor, B: int) -> torch.Tensor:
+        w = self.ws
+        if self.backend == "cpu":
+            a1, a2 = self._cpu_view(w.a1, B), self._cpu_view(w.a2, B)
+            self._C.cpu_forward(xb, self.model.params, a1, a2, w.y[:B])
+            y = w.y[:B]
+        else:
+            _, _, y = torch_ref.forward(xb, self.model.params)
+        return y.argmax(dim=1)
+
+    @torch.no_grad()
+    def classify(self, x: torch.Tensor) -> torch.Tensor:
+        """Predicted labels for a host fp32 batch [N, 784] (the reference's
+        `classify` entry point, Sequential/Main.cpp:186-200)."""
+        w = self.ws
+        preds = []
+        bs = w.max_batch
+        for i in range(0, x.shape[0], bs):
+            xb = x[i:i + bs]
+            B = xb.shape[0]
+            if self.backend == "hip":
+                xd = xb.to(self.device, dtype=self.act_dtype)
+                dummy = torch.zeros(B, dtype=torch.int32, device=self.device)
+                self._C.hip_fwdbwd(xd, self.model.params, w.a1, w.a2, w.y,
+                                   w.dz, w.dz2, w.dz1, dummy, w.loss_accum,
+                                   w.correct_accum, B, MODE_INFER,
+                                   native.current_stream_handle())
+                preds.append(w.y[:B].argmax(dim=1).cpu())
+            else:
+                xd = xb.to(torch.float32)
+                preds.append(self._forward_preds(xd, B).cpu())
+        return torch.cat(preds)
+
+    # ------------------------------------------------------------- training
+    def train_epoch(self, x: torch.Tensor, labels: torch.Tensor,
+                    log=print) -> float:
+        """One epoch over a host dataset, DP-sharded; returns the mean
+        per-sample error norm (the reference's per-epoch `error` print)."""
+        Bl = self.ws.max_batch
+        Bg = Bl * self.ctx.world_size
+        n = (x.shape[0] // Bg) * Bg  # drop ragged tail like a fixed-shape bench
+        total_loss, total_n = 0.0, 0
+        for s in range(0, n, Bg):
+            lo = s + self.ctx.rank * Bl
+            xb, lb = self.stage_batch(x[lo:lo + Bl], labels[lo:lo + Bl])
+            self.step(xb, lb)
+            if self.cfg.log_interval and \
+                    self.global_step % self.cfg.log_interval == 0:
+                loss, cnt = self.consume_loss()
+                total_loss += loss
+                total_n += cnt
+                if self.ctx.is_main and cnt:
+                    log(f"step {self.global_step}: error {loss / cnt:e}")
+        loss, cnt = self.consume_loss()
+        total_loss += loss
+        total_n += cnt
+        return total_loss / max(1, total_n)

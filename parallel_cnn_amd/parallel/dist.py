@@ -1,0 +1,75 @@
+"""Data parallelism over RCCL/xGMI (GPU) or gloo (CPU tests).
+
+Design: the whole gradient is ~9.4 KB, so the collective is
+pure-latency — ONE fused all-reduce of the flat gradient bucket per step
+(the reference MPI variant issued 16 blocking reduces per sample).  All
+ranks apply the identical deterministic update afterwards, so weights stay
+bit-identical across ranks (eliminating the reference's divergent non-root
+state bug class).
+
+One process per GPU; torch.distributed backend "nccl" IS RCCL on ROCm.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+
+@dataclass
+class DistContext:
+    rank: int = 0
+    world_size: int = 1
+    local_rank: int = 0
+    backend: str = ""
+
+    @property
+    def is_main(self) -> bool:
+        return self.rank == 0
+
+
+def is_distributed() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def init_from_env(device: str = "auto") -> DistContext:
+    """Initialise from torchrun env vars; no-op single-process context if
+    WORLD_SIZE is absent or 1."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return DistContext()
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_gpu = device != "cpu" and torch.cuda.is_available()
+    backend = "nccl" if use_gpu else "gloo"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend, timeout=datetime.timedelta(seconds=120))
+    return DistContext(rank=rank, world_size=world, local_rank=local_rank,
+                       backend=backend)
+
+
+def allreduce_grads(grads: torch.Tensor) -> None:
+    """Sum the flat gradient bucket across ranks (the engine folds the
+    1/world factor into the update scale)."""
+    if is_distributed():
+        dist.all_reduce(grads, op=dist.ReduceOp.SUM)
+
+
+def allreduce_scalar(value: float, device=None) -> float:
+    if not is_distributed():
+        return value
+    t = torch.tensor([value], dtype=torch.float64,
+                     device=device if device is not None else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return float(t.item())
+
+
+def barrier() -> None:
+    if is_distributed():
+        dist.barrier()

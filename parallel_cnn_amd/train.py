@@ -1,0 +1,78 @@
+"""CLI training driver — the framework equivalent of the reference's
+`main -> loaddata -> learn -> test`, with the same stdout
+shape ("Learning", per-epoch "error: %e", "Error Rate: %.2f%%") plus real
+flags, checkpointing and DP.
+
+Run:   python -m parallel_cnn_amd.train [--batch-size 64 ...]
+DP:    torchrun --standalone --nproc-per-node 8 -m parallel_cnn_amd.train ...
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import time
+
+import torch
+
+from .config import TrainConfig
+from .data.mnist import load_mnist, synthetic_mnist
+from .engine.trainer import Trainer
+from .parallel import dist as pdist
+
+
+def load_datasets(cfg: TrainConfig):
+    if cfg.data == "mnist":
+        xtr, ytr = load_mnist(
+            os.path.join(cfg.data_dir, "train-images.idx3-ubyte"),
+            os.path.join(cfg.data_dir, "train-labels.idx1-ubyte"))
+        xte, yte = load_mnist(
+            os.path.join(cfg.data_dir, "t10k-images.idx3-ubyte"),
+            os.path.join(cfg.data_dir, "t10k-labels.idx1-ubyte"))
+    else:
+        xtr, ytr = synthetic_mnist(cfg.train_count, seed=cfg.seed)
+        xte, yte = synthetic_mnist(cfg.test_count, seed=cfg.seed + 1)
+    return xtr, ytr, xte, yte
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(description=__doc__)
+    TrainConfig.add_cli_args(p)
+    args = p.parse_args(argv)
+    cfg = TrainConfig.from_args(args)
+
+    ctx = pdist.init_from_env(cfg.resolved_device())
+    trainer = Trainer(cfg, ctx=ctx)
+    if cfg.ckpt_load:
+        trainer.model.load(cfg.ckpt_load)
+
+    xtr, ytr, xte, yte = load_datasets(cfg)
+
+    if ctx.is_main:
+        print("Learning", flush=True)
+    t0 = time.perf_counter()
+    for epoch in range(cfg.epochs):
+        err = trainer.train_epoch(xtr, ytr)
+        dt = time.perf_counter() - t0
+        if ctx.is_main:
+            print(f"error: {err:e}, time_on_gpu: {dt:f}", flush=True)
+        if err < cfg.threshold:
+            break
+    if trainer.device.type == "cuda":
+        torch.cuda.synchronize()
+    total = time.perf_counter() - t0
+    if ctx.is_main:
+        print(f"\n Time - {total * 1e3:f} ms", flush=True)
+
+    if cfg.ckpt_save and ctx.is_main:
+        trainer.model.save(cfg.ckpt_save)
+        print(f"saved checkpoint: {cfg.ckpt_save}", flush=True)
+
+    err_rate = trainer.evaluate(xte, yte)
+    if ctx.is_main:
+        print(f"Error Rate: {err_rate:.2f}%", flush=True)
+    pdist.barrier()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

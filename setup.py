@@ -1,0 +1,78 @@
+"""In-tree build of parallel_cnn_amd._C.
+
+Compiles the gfx950 HIP kernels with hipcc (cross-compiles fine on a
+GPU-less box) and links them into a torch CppExtension together with the
+CPU reference ops.  Build with:
+
+    python setup.py build_ext --inplace
+"""
+import os
+import subprocess
+import sys
+
+from setuptools import setup
+
+from torch.utils.cpp_extension import BuildExtension, CppExtension
+
+ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")
+ARCH = os.environ.get("PCNN_GFX_ARCH", "gfx950")
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+HIP_SOURCES = [
+    os.path.join(HERE, "csrc", "hip", "lenet_kernels.hip"),
+]
+
+
+def compile_hip_objects():
+    objs = []
+    for src in HIP_SOURCES:
+        obj = os.path.splitext(src)[0] + ".o"
+        stale = (not os.path.exists(obj)
+                 or os.path.getmtime(obj) < os.path.getmtime(src)
+                 or os.path.getmtime(obj) < os.path.getmtime(
+                     os.path.join(HERE, "csrc", "lenet_dims.h")))
+        if stale:
+            cmd = [
+                os.path.join(ROCM, "bin", "hipcc"),
+                f"--offload-arch={ARCH}",
+                "-O3",
+                "-std=c++17",
+                "-fPIC",
+                "-c",
+                src,
+                "-o",
+                obj,
+            ]
+            print("[pcnn build]", " ".join(cmd), flush=True)
+            subprocess.check_call(cmd)
+        objs.append(obj)
+    return objs
+
+
+ext = CppExtension(
+    name="parallel_cnn_amd._C",
+    sources=["csrc/bindings.cpp", "csrc/cpu_ops.cpp"],
+    extra_objects=compile_hip_objects(),
+    libraries=["amdhip64"],
+    library_dirs=[os.path.join(ROCM, "lib")],
+    extra_compile_args=["-O3", "-std=c++17"],
+    extra_link_args=[f"-Wl,-rpath,{os.path.join(ROCM, 'lib')}"],
+)
+
+setup(
+    name="parallel_cnn_amd",
+    version="0.1.0",
+    description="MI355X-native CNN training framework "
+    "(capability parity with Tamerkobba/Parallel-CNN)",
+    packages=[
+        "parallel_cnn_amd",
+        "parallel_cnn_amd.data",
+        "parallel_cnn_amd.models",
+        "parallel_cnn_amd.ops",
+        "parallel_cnn_amd.parallel",
+        "parallel_cnn_amd.engine",
+        "parallel_cnn_amd.utils",
+    ],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)

@@ -1,0 +1,48 @@
+"""Config system tests."""
+import argparse
+
+import pytest
+
+from parallel_cnn_amd.config import TrainConfig
+
+
+def test_defaults_match_reference():
+    cfg = TrainConfig()
+    assert cfg.dt == 0.1
+    assert cfg.threshold == 1e-2
+    assert cfg.epochs == 1
+
+
+def test_yaml_roundtrip(tmp_path):
+    p = tmp_path / "cfg.yaml"
+    p.write_text("batch_size: 128\ndt: 0.05\nbackend: cpu\n")
+    cfg = TrainConfig.from_yaml(str(p))
+    assert cfg.batch_size == 128
+    assert cfg.dt == 0.05
+    assert cfg.backend == "cpu"
+
+
+def test_yaml_unknown_key(tmp_path):
+    p = tmp_path / "cfg.yaml"
+    p.write_text("not_a_key: 1\n")
+    with pytest.raises(ValueError, match="unknown config keys"):
+        TrainConfig.from_yaml(str(p))
+
+
+def test_cli_overrides_yaml(tmp_path):
+    p = tmp_path / "cfg.yaml"
+    p.write_text("batch_size: 128\n")
+    ap = argparse.ArgumentParser()
+    TrainConfig.add_cli_args(ap)
+    args = ap.parse_args(["--config", str(p), "--batch-size", "256",
+                          "--dt", "0.2"])
+    cfg = TrainConfig.from_args(args)
+    assert cfg.batch_size == 256
+    assert cfg.dt == 0.2
+
+
+def test_resolution():
+    cfg = TrainConfig(device="cpu", backend="auto")
+    assert cfg.resolved_backend() == "cpu"
+    cfg2 = TrainConfig(device="cpu", backend="torchref")
+    assert cfg2.resolved_backend() == "torchref"

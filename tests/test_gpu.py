@@ -1,0 +1,180 @@
+"""GPU (MI355X) tests: every HIP kernel against the fp32 PyTorch/native-CPU
+oracle, end-to-end step parity, convergence, and the bench path."""
+import json
+import subprocess
+import sys
+
+import pytest
+import torch
+
+from parallel_cnn_amd import _C
+from parallel_cnn_amd.config import TrainConfig
+from parallel_cnn_amd.data.mnist import synthetic_mnist
+from parallel_cnn_amd.engine.trainer import Trainer
+from parallel_cnn_amd.models.lenet import LeNet5
+from parallel_cnn_amd.ops import native
+from parallel_cnn_amd.ops import shapes as S
+from parallel_cnn_amd.ops import torch_ref
+
+pytestmark = pytest.mark.gpu
+
+
+def make_case(B, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.rand(B, S.IN_PIX, generator=g)
+    labels = torch.randint(0, 10, (B,), generator=g)
+    params = (0.5 - torch.rand(S.N_PARAMS, generator=g)).float()
+    return x, labels, params
+
+
+def run_gpu_step_pieces(x, labels, params, act_dtype, device):
+    """Run kernel A (fwdbwd) + kernel B (wgrad) on the GPU; return all
+    intermediates on the host."""
+    B = x.shape[0]
+    ad = act_dtype
+    xd = x.to(device, dtype=ad)
+    ld = labels.to(device, dtype=torch.int32)
+    pd = params.to(device)
+    a1 = torch.empty(B, S.C1_OUT, dtype=ad, device=device)
+    a2 = torch.empty(B, S.S1_OUT, dtype=ad, device=device)
+    y = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=device)
+    dz = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=device)
+    dz2 = torch.empty(B, S.S1_OUT, dtype=torch.float32, device=device)
+    dz1 = torch.empty(B, S.C1_OUT, dtype=torch.float32, device=device)
+    loss = torch.zeros(1, dtype=torch.float32, device=device)
+    corr = torch.zeros(1, dtype=torch.int32, device=device)
+    grads = torch.zeros(S.N_PARAMS, dtype=torch.float32, device=device)
+    stream = native.current_stream_handle()
+    _C.hip_fwdbwd(xd, pd, a1, a2, y, dz, dz2, dz1, ld, loss, corr, B, 0,
+                  stream)
+    _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, stream)
+    torch.cuda.synchronize()
+    return (a1.float().cpu(), a2.float().cpu(), y.cpu(), dz.cpu(), dz2.cpu(),
+            dz1.cpu(), grads.cpu(), float(loss.item()))
+
+
+def ref_step_pieces(x, labels, params):
+    a1, a2, y = torch_ref.forward(x, params)
+    dz, dz2, dz1, grads, loss = torch_ref.backward(x, params, a1, a2, y,
+                                                   labels)
+    return (a1.reshape(x.shape[0], -1), a2.reshape(x.shape[0], -1), y, dz,
+            dz2, dz1, grads, loss)
+
+
+@pytest.mark.parametrize("B", [1, 64, 100])
+def test_fwdbwd_wgrad_fp32_matches_oracle(B, device):
+    x, labels, params = make_case(B, seed=B)
+    got = run_gpu_step_pieces(x, labels, params, torch.float32, device)
+    want = ref_step_pieces(x, labels, params)
+    names = ["a1", "a2", "y", "dz", "dz2", "dz1", "grads"]
+    for n, a, b in zip(names, got[:7], want[:7]):
+        diff = (a.reshape(-1) - b.reshape(-1)).abs().max().item()
+        assert diff < 2e-4, f"{n}: max abs diff {diff}"
+    assert abs(got[7] - want[7]) < 1e-2 * max(1.0, abs(want[7]))
+
+
+@pytest.mark.parametrize("B", [64])
+def test_fwdbwd_wgrad_bf16_acts_close(B, device):
+    """bf16 activation storage: arithmetic is fp32 so tolerances are the
+    bf16 storage quantum (~0.8% relative)."""
+    x, labels, params = make_case(B, seed=5)
+    got = run_gpu_step_pieces(x, labels, params, torch.bfloat16, device)
+    want = ref_step_pieces(x, labels, params)
+    for n, a, b, tol in [("y", got[2], want[2], 2e-2),
+                         ("dz", got[3], want[3], 2e-2),
+                         ("dz2", got[4], want[4], 2e-2),
+                         ("dz1", got[5], want[5], 2e-2)]:
+        diff = (a.reshape(-1) - b.reshape(-1)).abs().max().item()
+        assert diff < tol, f"{n}: max abs diff {diff}"
+    gdiff = (got[6] - want[6]).abs().max().item()
+    gref = want[6].abs().max().item()
+    assert gdiff < 2e-2 * max(1.0, gref), f"grads: {gdiff} vs max {gref}"
+
+
+def test_update_kernel(device):
+    params = torch.zeros(S.N_PARAMS, device=device)
+    grads = torch.ones(S.N_PARAMS, device=device)
+    _C.hip_update(params, grads, 0.05, native.current_stream_handle())
+    torch.cuda.synchronize()
+    assert torch.allclose(params.cpu(), torch.full((S.N_PARAMS,), 0.05))
+    assert grads.abs().sum().item() == 0
+
+
+def test_eval_mode_correct_count(device):
+    B = 32
+    x, labels, params = make_case(B, seed=9)
+    # oracle predictions
+    _, _, y = torch_ref.forward(x, params)
+    want_correct = int((y.argmax(1) == labels).sum().item())
+    xd = x.to(device, dtype=torch.float32)
+    ld = labels.to(device, dtype=torch.int32)
+    pd = params.to(device)
+    a1 = torch.empty(B, S.C1_OUT, dtype=torch.float32, device=device)
+    a2 = torch.empty(B, S.S1_OUT, dtype=torch.float32, device=device)
+    yd = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=device)
+    e = torch.empty(0, device=device)
+    corr = torch.zeros(1, dtype=torch.int32, device=device)
+    loss = torch.zeros(1, dtype=torch.float32, device=device)
+    _C.hip_fwdbwd(xd, pd, a1, a2, yd, e, e, e, ld, loss, corr, B, 1,
+                  native.current_stream_handle())
+    torch.cuda.synchronize()
+    assert int(corr.item()) == want_correct
+
+
+def test_trainer_step_trajectory_matches_cpu(device):
+    """5 fp32 steps on GPU == 5 steps of the native CPU engine."""
+    cfg_g = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                        act_dtype="fp32", log_interval=0)
+    cfg_c = TrainConfig(batch_size=16, device="cpu", backend="cpu",
+                        log_interval=0)
+    tg, tc = Trainer(cfg_g), Trainer(cfg_c)
+    x, y = synthetic_mnist(80, seed=3)
+    for s in range(5):
+        xb, yb = x[s * 16:(s + 1) * 16], y[s * 16:(s + 1) * 16]
+        tg.step(*tg.stage_batch(xb, yb))
+        tc.step(*tc.stage_batch(xb, yb))
+    torch.cuda.synchronize()
+    diff = (tg.model.params.cpu() - tc.model.params).abs().max().item()
+    assert diff < 5e-4, diff
+    lg, ng = tg.consume_loss()
+    lc, nc = tc.consume_loss()
+    assert ng == nc == 80
+    assert abs(lg - lc) < 1e-2 * max(1.0, lc)
+
+
+def test_gpu_convergence_bf16(device):
+    """The flagship bf16 path must actually learn the structured bands."""
+    xtr, ytr = synthetic_mnist(4096, seed=2)
+    xte, yte = synthetic_mnist(512, seed=3)
+    cfg = TrainConfig(batch_size=32, device="cuda", backend="hip",
+                      act_dtype="bf16", log_interval=0)
+    t = Trainer(cfg)
+    before = t.evaluate(xte, yte)
+    for _ in range(3):
+        t.train_epoch(xtr, ytr, log=lambda *a: None)
+    after = t.evaluate(xte, yte)
+    assert after < before
+    assert after < 5.0, (before, after)
+
+
+def test_native_extension_is_loaded_on_gpu(device):
+    """The HIP path must be the one that runs (no silent eager fallback)."""
+    assert native.available()
+    assert _C.HAS_HIP_KERNELS
+    cfg = TrainConfig(batch_size=4, device="cuda", log_interval=0)
+    t = Trainer(cfg)
+    assert t.backend == "hip"
+
+
+def test_bench_contract_single_gpu(device):
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "1", "--steps", "30",
+         "--warmup", "5"],
+        capture_output=True, text=True, timeout=600, check=True)
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["n_gpus"] == 1 and r["steps"] == 30
+    assert r["unit"] == "images/sec" and r["higher_is_better"]
+    assert r["value"] > 0 and r["ms_per_step"] > 0
+    assert r["dtype"] == "bf16"
+    assert r["config"]["global_batch"] == 64
