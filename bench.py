@@ -36,6 +36,7 @@ def main() -> int:
     p.add_argument("--act-dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default="auto")
+    p.add_argument("--wgrad-chunk", type=int, default=8)
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
@@ -44,7 +45,8 @@ def main() -> int:
     from parallel_cnn_amd.parallel import dist as pdist
 
     cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
-                      device=args.device, log_interval=0, data="synthetic")
+                      device=args.device, log_interval=0, data="synthetic",
+                      wgrad_chunk=args.wgrad_chunk)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
@@ -56,22 +58,17 @@ def main() -> int:
 
     # Device-resident synthetic epoch pool (no H2D inside the timed loop; the
     # pool is one epoch's worth of batches, cycled).
-    n_pool_batches = min(args.steps + args.warmup, 60000 // max(1, B)) or 1
+    n_pool_batches = max(1, min(args.steps + args.warmup,
+                                60000 // max(1, B)))
     x_host, y_host = synthetic_mnist(n_pool_batches * B,
                                      seed=1234 + ctx.rank, structured=False)
-    xs, ys = [], []
-    for i in range(n_pool_batches):
-        xb, yb = trainer.stage_batch(x_host[i * B:(i + 1) * B],
-                                     y_host[i * B:(i + 1) * B])
-        xs.append(xb)
-        ys.append(yb)
+    x_pool, y_pool = trainer.stage_batch(x_host, y_host)
+    x_pool, y_pool = x_pool.contiguous(), y_pool.contiguous()
     if device == "cuda":
         torch.cuda.synchronize()
 
     def run(n_steps: int):
-        for s in range(n_steps):
-            i = s % n_pool_batches
-            trainer.step(xs[i], ys[i])
+        trainer.run_steps_pooled(x_pool, y_pool, n_steps)
 
     run(args.warmup)
     pdist.barrier()

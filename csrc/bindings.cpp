@@ -27,7 +27,8 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
                        int B, int act_is_bf16, int mode, void* stream);
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
-                      float* grads, int B, int act_is_bf16, void* stream);
+                      float* grads, int B, int act_is_bf16, int chunk_imgs,
+                      void* stream);
 int pcnn_launch_update(float* params, float* grads, float step, void* stream);
 const char* pcnn_hip_error_string(int err);
 }
@@ -70,12 +71,12 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
 
 void hip_wgrad(at::Tensor x, at::Tensor a1, at::Tensor a2, at::Tensor dz,
                at::Tensor dz2, at::Tensor dz1, at::Tensor grads, int64_t B,
-               int64_t stream) {
+               int64_t chunk_imgs, int64_t stream) {
   int f = act_flag(x);
   check_hip(pcnn_launch_wgrad(x.data_ptr(), a1.data_ptr(), a2.data_ptr(),
                               dz.data_ptr<float>(), dz2.data_ptr<float>(),
                               dz1.data_ptr<float>(), grads.data_ptr<float>(),
-                              (int)B, f, (void*)stream),
+                              (int)B, f, (int)chunk_imgs, (void*)stream),
             "wgrad");
 }
 
@@ -85,6 +86,49 @@ void hip_update(at::Tensor params, at::Tensor grads, double step,
                                grads.data_ptr<float>(), (float)step,
                                (void*)stream),
             "update");
+}
+
+// Single-GPU fused training loop: enqueues `steps` full training steps
+// (fwd+bwd-data -> wgrad -> update) from C++, cycling a device-resident
+// batch pool.  Asynchronous — caller syncs the stream.  The distributed
+// path keeps the per-step Python loop (the all-reduce sits between wgrad
+// and update there).
+void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
+                     at::Tensor params, at::Tensor grads, at::Tensor a1,
+                     at::Tensor a2, at::Tensor y, at::Tensor dz,
+                     at::Tensor dz2, at::Tensor dz1, at::Tensor loss_accum,
+                     int64_t B, int64_t steps, int64_t chunk_imgs,
+                     double step_scale, int64_t stream) {
+  TORCH_CHECK(x_pool.is_cuda() && x_pool.dim() == 2, "x_pool [P*B, 784]");
+  TORCH_CHECK(labels_pool.scalar_type() == at::kInt, "labels must be int32");
+  const int64_t pool_rows = x_pool.size(0);
+  TORCH_CHECK(pool_rows % B == 0, "pool rows must be a multiple of B");
+  const int64_t P = pool_rows / B;
+  const int f = act_flag(x_pool);
+  const size_t esz = f ? 2 : 4;
+  const char* xp = (const char*)x_pool.data_ptr();
+  const int* lp = labels_pool.data_ptr<int>();
+  float* pp = params.data_ptr<float>();
+  float* gp = grads.data_ptr<float>();
+  void* s = (void*)stream;
+  for (int64_t st = 0; st < steps; ++st) {
+    const int64_t i = st % P;
+    const void* xb = xp + (size_t)i * B * pcnn::IN_PIX * esz;
+    const int* lb = lp + i * B;
+    check_hip(pcnn_launch_fwdbwd(xb, pp, a1.data_ptr(), a2.data_ptr(),
+                                 y.data_ptr<float>(), dz.data_ptr<float>(),
+                                 dz2.data_ptr<float>(), dz1.data_ptr<float>(),
+                                 lb, loss_accum.data_ptr<float>(), nullptr,
+                                 (int)B, f, 0, s),
+              "train_steps/fwdbwd");
+    check_hip(pcnn_launch_wgrad(xb, a1.data_ptr(), a2.data_ptr(),
+                                dz.data_ptr<float>(), dz2.data_ptr<float>(),
+                                dz1.data_ptr<float>(), gp, (int)B, f,
+                                (int)chunk_imgs, s),
+              "train_steps/wgrad");
+    check_hip(pcnn_launch_update(pp, gp, (float)step_scale, s),
+              "train_steps/update");
+  }
 }
 
 }  // namespace
@@ -97,6 +141,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hip_fwdbwd", &hip_fwdbwd);
   m.def("hip_wgrad", &hip_wgrad);
   m.def("hip_update", &hip_update);
+  m.def("hip_train_steps", &hip_train_steps);
   m.attr("N_PARAMS") = pcnn::N_PARAMS;
   m.attr("OFF_C1W") = pcnn::OFF_C1W;
   m.attr("OFF_C1B") = pcnn::OFF_C1B;

@@ -157,7 +157,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     float s = 0.f;
 #pragma unroll
     for (int k = 0; k < FC_OUT; ++k) s += L.sq[k];
-    atomicAdd(loss_accum, sqrtf(s));
+    unsafeAtomicAdd(loss_accum, sqrtf(s));
   }
 
   // ---- fc backward-data -> pool preact gradient ----
@@ -198,14 +198,14 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 // the flat fp32 gradient bucket (conflicts only across chunks).
 // ---------------------------------------------------------------------------
 
-constexpr int WG_CHUNK = 8;  // images per chunk
+constexpr int WG_CHUNK_DEFAULT = 8;  // images per chunk (runtime-tunable)
 
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad(
     const act_t* __restrict__ x, const act_t* __restrict__ a1g,
     const act_t* __restrict__ a2g, const float* __restrict__ dzg,
     const float* __restrict__ dz2g, const float* __restrict__ dz1g,
-    float* __restrict__ grads, int B, int NC) {
+    float* __restrict__ grads, int B, int NC, int CIMG) {
   const int tid = threadIdx.x;
   const int blk = blockIdx.x;
 
@@ -213,8 +213,8 @@ __global__ __launch_bounds__(256) void k_wgrad(
     // ---- conv1: dW[o,i,j] = sum_{b,r,c} dz1[b,o,r,c] * x[b,r+i,c+j] / 576
     const int o = blk / NC;
     const int chunk = blk - o * NC;
-    const int b_lo = chunk * WG_CHUNK;
-    const int b_hi = min(B, b_lo + WG_CHUNK);
+    const int b_lo = chunk * CIMG;
+    const int b_hi = min(B, b_lo + CIMG);
     __shared__ float S[IN_PIX + C1_K * C1_K + 1];
     float* xs = S;
     float* wacc = S + IN_PIX;     // [25] block-level accumulators
@@ -249,13 +249,13 @@ __global__ __launch_bounds__(256) void k_wgrad(
     __syncthreads();
     constexpr float inv_pix = 1.0f / (float)C1_PIX;
     if (tid < C1_K * C1_K)
-      atomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + tid], wacc[tid] * inv_pix);
-    if (tid == C1_K * C1_K) atomicAdd(&grads[OFF_C1B + o], bacc[0] * inv_pix);
+      unsafeAtomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + tid], wacc[tid] * inv_pix);
+    if (tid == C1_K * C1_K) unsafeAtomicAdd(&grads[OFF_C1B + o], bacc[0] * inv_pix);
   } else if (blk < (C1_CH + 1) * NC) {
     // ---- pool: dW[i,j] = sum_{b,o,p,q} dz2[b,o,p,q] * a1[b,o,4p+i,4q+j]
     const int chunk = blk - C1_CH * NC;
-    const int b_lo = chunk * WG_CHUNK;
-    const int b_hi = min(B, b_lo + WG_CHUNK);
+    const int b_lo = chunk * CIMG;
+    const int b_hi = min(B, b_lo + CIMG);
     __shared__ float S[C1_OUT + S1_WSZ + 1];
     float* a1s = S;
     float* wacc = S + C1_OUT;
@@ -290,14 +290,14 @@ __global__ __launch_bounds__(256) void k_wgrad(
     for (int w = 0; w < S1_WSZ; ++w) atomicAdd(&wacc[w], acc[w]);
     atomicAdd(bacc, bsum);
     __syncthreads();
-    if (tid < S1_WSZ) atomicAdd(&grads[OFF_S1W + tid], wacc[tid]);
+    if (tid < S1_WSZ) unsafeAtomicAdd(&grads[OFF_S1W + tid], wacc[tid]);
     if (tid == S1_WSZ)
-      atomicAdd(&grads[OFF_S1B], bacc[0] / (float)S1_OUT);
+      unsafeAtomicAdd(&grads[OFF_S1B], bacc[0] / (float)S1_OUT);
   } else {
     // ---- fc: dW[k,m] = sum_b dz[b,k] * a2[b,m];  db[k] = sum_b dz[b,k]
     const int chunk = blk - (C1_CH + 1) * NC;
-    const int b_lo = chunk * WG_CHUNK;
-    const int b_hi = min(B, b_lo + WG_CHUNK);
+    const int b_lo = chunk * CIMG;
+    const int b_hi = min(B, b_lo + CIMG);
     __shared__ float S[S1_OUT + FC_OUT];
     float* a2s = S;
     float* dzs = S + S1_OUT;
@@ -327,9 +327,9 @@ __global__ __launch_bounds__(256) void k_wgrad(
 #pragma unroll
     for (int u = 0; u < NACC; ++u) {
       const int q = tid + u * 256;
-      if (q < FC_WSZ) atomicAdd(&grads[OFF_FW + q], acc[u]);
+      if (q < FC_WSZ) unsafeAtomicAdd(&grads[OFF_FW + q], acc[u]);
     }
-    if (tid < FC_OUT) atomicAdd(&grads[OFF_FB + tid], bsum);
+    if (tid < FC_OUT) unsafeAtomicAdd(&grads[OFF_FB + tid], bsum);
   }
 }
 
@@ -400,18 +400,20 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
 
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
-                      float* grads, int B, int act_is_bf16, void* stream) {
-  const int NC = (B + WG_CHUNK - 1) / WG_CHUNK;
+                      float* grads, int B, int act_is_bf16, int chunk_imgs,
+                      void* stream) {
+  const int CIMG = chunk_imgs > 0 ? chunk_imgs : WG_CHUNK_DEFAULT;
+  const int NC = (B + CIMG - 1) / CIMG;
   dim3 grid((C1_CH + 2) * NC), block(256);
   hipStream_t s = (hipStream_t)stream;
   if (act_is_bf16) {
     hipLaunchKernelGGL((k_wgrad<bf16>), grid, block, 0, s, (const bf16*)x,
                        (const bf16*)a1, (const bf16*)a2, dz, dz2, dz1, grads,
-                       B, NC);
+                       B, NC, CIMG);
   } else {
     hipLaunchKernelGGL((k_wgrad<float>), grid, block, 0, s, (const float*)x,
                        (const float*)a1, (const float*)a2, dz, dz2, dz1, grads,
-                       B, NC);
+                       B, NC, CIMG);
   }
   return (int)hipGetLastError();
 }

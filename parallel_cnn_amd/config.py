@@ -38,6 +38,7 @@ class TrainConfig:
     # execution
     backend: str = "auto"            # auto | hip | cpu | torchref
     device: str = "auto"             # auto | cuda | cpu
+    wgrad_chunk: int = 8             # images per weight-grad workgroup chunk
 
     # io / observability
     log_interval: int = 100          # steps between loss readouts

@@ -98,7 +98,7 @@ class Trainer:
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream)
             self._C.hip_wgrad(x, w.a1, w.a2, w.dz, w.dz2, w.dz1, m.grads, B,
-                              stream)
+                              self.cfg.wgrad_chunk, stream)
             pdist.allreduce_grads(m.grads)
             self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
                                native.current_stream_handle())
@@ -123,6 +123,28 @@ class Trainer:
             torch_ref.update(m.params, m.grads, self.cfg.dt, scale)
         self._samples_seen += B * self.ctx.world_size
         self.global_step += 1
+
+    def run_steps_pooled(self, x_pool: torch.Tensor,
+                         labels_pool: torch.Tensor, steps: int) -> None:
+        """Run `steps` training steps over a device-resident batch pool
+        (pool rows = P*B, cycled).  Single-GPU non-distributed runs use the
+        C++ fused enqueue loop (no per-step Python overhead); distributed
+        runs keep the per-step loop with the gradient all-reduce."""
+        B = self.ws.max_batch
+        P = x_pool.shape[0] // B
+        if self.backend == "hip" and self.ctx.world_size == 1:
+            w = self.ws
+            self._C.hip_train_steps(
+                x_pool, labels_pool, self.model.params, self.model.grads,
+                w.a1, w.a2, w.y, w.dz, w.dz2, w.dz1, w.loss_accum, B, steps,
+                self.cfg.wgrad_chunk, self.cfg.dt * self._update_scale(B),
+                native.current_stream_handle())
+            self._samples_seen += B * steps
+            self.global_step += steps
+        else:
+            for s in range(steps):
+                i = (s % P) * B
+                self.step(x_pool[i:i + B], labels_pool[i:i + B])
 
     def _cpu_view(self, t: torch.Tensor, B: int) -> torch.Tensor:
         assert t.dtype == torch.float32, \

@@ -47,7 +47,7 @@ def run_gpu_step_pieces(x, labels, params, act_dtype, device):
     stream = native.current_stream_handle()
     _C.hip_fwdbwd(xd, pd, a1, a2, y, dz, dz2, dz1, ld, loss, corr, B, 0,
                   stream)
-    _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, stream)
+    _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, 8, stream)
     torch.cuda.synchronize()
     return (a1.float().cpu(), a2.float().cpu(), y.cpu(), dz.cpu(), dz2.cpu(),
             dz1.cpu(), grads.cpu(), float(loss.item()))
@@ -69,7 +69,10 @@ def test_fwdbwd_wgrad_fp32_matches_oracle(B, device):
     names = ["a1", "a2", "y", "dz", "dz2", "dz1", "grads"]
     for n, a, b in zip(names, got[:7], want[:7]):
         diff = (a.reshape(-1) - b.reshape(-1)).abs().max().item()
-        assert diff < 2e-4, f"{n}: max abs diff {diff}"
+        # grads are B-way sums accumulated with atomics: allow summation-
+        # order roundoff to scale with the magnitude of the reduction
+        tol = 2e-4 * max(1.0, b.abs().max().item()) if n == "grads" else 2e-4
+        assert diff < tol, f"{n}: max abs diff {diff}"
     assert abs(got[7] - want[7]) < 1e-2 * max(1.0, abs(want[7]))
 
 
@@ -155,6 +158,56 @@ def test_gpu_convergence_bf16(device):
     after = t.evaluate(xte, yte)
     assert after < before
     assert after < 5.0, (before, after)
+
+
+def test_fused_step_loop_matches_python_loop(device):
+    """The C++ multi-step enqueue path must produce the same trajectory as
+    per-step Python calls."""
+    x, y = synthetic_mnist(64, seed=11)
+    cfg = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0)
+    t1, t2 = Trainer(cfg), Trainer(cfg)
+    xp, yp = t1.stage_batch(x, y)
+    t1.run_steps_pooled(xp.contiguous(), yp.contiguous(), 4)  # C++ loop
+    for s in range(4):                                        # Python loop
+        t2.step(xp[s * 16:(s + 1) * 16], yp[s * 16:(s + 1) * 16])
+    torch.cuda.synchronize()
+    diff = (t1.model.params - t2.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+    l1, n1 = t1.consume_loss()
+    l2, n2 = t2.consume_loss()
+    assert n1 == n2 == 64
+    assert abs(l1 - l2) < 1e-3 * max(1.0, l2)
+
+
+def test_wgrad_chunk_sizes_agree(device):
+    """The wgrad chunk knob must not change the math."""
+    B = 64
+    x, labels, params = make_case(B, seed=21)
+    base = None
+    for chunk in (2, 8, 32, 64):
+        xd = x.to(device)
+        pd = params.to(device)
+        ld = labels.to(device, dtype=torch.int32)
+        a1 = torch.empty(B, S.C1_OUT, device=device)
+        a2 = torch.empty(B, S.S1_OUT, device=device)
+        yv = torch.empty(B, S.FC_OUT, device=device)
+        dz = torch.empty(B, S.FC_OUT, device=device)
+        dz2 = torch.empty(B, S.S1_OUT, device=device)
+        dz1 = torch.empty(B, S.C1_OUT, device=device)
+        loss = torch.zeros(1, device=device)
+        corr = torch.zeros(1, dtype=torch.int32, device=device)
+        grads = torch.zeros(S.N_PARAMS, device=device)
+        st = native.current_stream_handle()
+        _C.hip_fwdbwd(xd, pd, a1, a2, yv, dz, dz2, dz1, ld, loss, corr, B, 0,
+                      st)
+        _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, chunk, st)
+        torch.cuda.synchronize()
+        g = grads.cpu()
+        if base is None:
+            base = g
+        else:
+            assert torch.allclose(g, base, atol=1e-3), chunk
 
 
 def test_native_extension_is_loaded_on_gpu(device):
