@@ -2,22 +2,26 @@
 // training step.  Design notes:
 //
 // The whole network is ~1 MFLOP per image forward+backward, so on MI355X the
-// bound is kernel-launch count and memory latency, never FLOPs (SURVEY.md §7
-// "hard parts").  The reference CUDA variant spends 17 launches + 8 memsets
-// per *sample* (SURVEY.md §2.3); here one training step of a whole batch is
-// THREE kernels:
+// bound is kernel-launch count and LATENCY (barrier phases, dependent memory
+// chains), never FLOPs (SURVEY.md §7 "hard parts").  The reference CUDA
+// variant spends 17 launches + 8 memsets per *sample* (SURVEY.md §2.3); here
+// one training step of a whole batch is THREE kernels:
 //
 //   1. k_fwdbwd  — fused forward + backward-data.  One 256-thread workgroup
-//      per image; every activation lives in LDS (≈28 KB incl. a parameter
-//      stage), global traffic is one read of x and one write of the tensors
-//      the weight-grad kernel needs.  Fuses what the reference ran as 12
-//      separate kernels (fp_c1, sigmoid, fp_s1, sigmoid, fp_f, sigmoid,
-//      makeError, nrm2, bp_output_s1, bp_preact_s1, bp_output_c1,
-//      bp_preact_c1).
-//   2. k_wgrad   — all weight/bias gradients, batch-reduced.  Grid is
-//      (role × batch-chunk); per-thread register accumulation over the
-//      chunk, one fp32 atomicAdd per weight per block into the flat
-//      gradient bucket (which is also the RCCL all-reduce payload).
+//      per image; activations live in LDS; 5 barrier phases
+//      ({stage x+params} {conv1} {pool} {fc+loss-residual} {fc-bwd}
+//      {pool-bwd}); the fc dot products are 16-lane shuffle reductions so no
+//      phase has a >32-step dependency chain.  Fuses what the reference ran
+//      as 12 separate kernels per sample.
+//   2. k_wgrad   — all weight/bias gradients, batch-reduced.  Measured
+//      lesson (profiles/, round 1): a chunk-serial design with LDS staging
+//      and per-image barriers ran 62 us — latency-chained.  v2 is
+//      barrier-free: grid-stride over flattened (image, position) work
+//      items, per-thread register accumulators, wave-level __shfl_down
+//      reduction, then a handful of hardware fp32 atomics
+//      (unsafeAtomicAdd -> global_atomic_add_f32) into the flat gradient
+//      bucket.  The fc weight grads use exclusive per-thread ownership of
+//      (k,m) pairs — no atomics, plain accumulate-stores.
 //   3. k_update  — SGD apply (p += dt*scale*g) fused with gradient zeroing.
 //
 // Numerics: activations are stored bf16 (or fp32, template) in global
@@ -76,13 +80,13 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   if (b >= B) return;
   const int tid = threadIdx.x;
 
-  // Stage parameters and the input image into LDS.
+  // ---- phase 0: stage parameters and the input image into LDS ----
   for (int i = tid; i < N_PARAMS; i += 256) L.ps[i] = params[i];
   const act_t* xb = x + (size_t)b * IN_PIX;
   for (int i = tid; i < IN_PIX; i += 256) L.xs[i] = ldf(xb + i);
   __syncthreads();
 
-  // ---- conv1 (5x5 valid, 6 ch) + sigmoid ----
+  // ---- phase 1: conv1 (5x5 valid, 6 ch) + sigmoid ----
   for (int t = tid; t < C1_OUT; t += 256) {
     const int o = t / C1_PIX;
     const int rc = t - o * C1_PIX;
@@ -101,7 +105,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   }
   __syncthreads();
 
-  // ---- trainable pool (4x4 stride 4, shared kernel) + sigmoid ----
+  // ---- phase 2: trainable pool (4x4 stride 4, shared kernel) + sigmoid ----
   if (tid < S1_OUT) {
     const int o = tid / S1_PIX;
     const int pq = tid - o * S1_PIX;
@@ -120,15 +124,33 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   }
   __syncthreads();
 
-  // ---- fc (216 -> 10) + sigmoid ----
-  if (tid < FC_OUT) {
-    float acc = L.ps[OFF_FB + tid];
-    const float* wk = &L.ps[OFF_FW + tid * FC_IN];
-#pragma unroll 8
-    for (int m = 0; m < FC_IN; ++m) acc += wk[m] * L.a2s[m];
-    const float v = sigmoidf_dev(acc);
-    L.ys[tid] = v;
-    if (yg != nullptr) yg[(size_t)b * FC_OUT + tid] = v;
+  // ---- phase 3: fc (216 -> 10) + sigmoid [+ loss residual] ----
+  // 16 lanes per output class: lane l of group k sums m = l, l+16, ...
+  // then a 4-step shuffle tree; the group leader applies bias+sigmoid and
+  // (TRAIN) computes the residual immediately — no extra barrier phase.
+  if (tid < FC_OUT * 16) {
+    const int k = tid >> 4;
+    const int l = tid & 15;
+    const float* wk = &L.ps[OFF_FW + k * FC_IN];
+    float p = 0.f;
+#pragma unroll
+    for (int u = 0; u < (FC_IN + 15) / 16; ++u) {
+      const int m = l + u * 16;
+      if (m < FC_IN) p += wk[m] * L.a2s[m];
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) p += __shfl_down(p, off, 16);
+    if (l == 0) {
+      const float v = sigmoidf_dev(p + L.ps[OFF_FB + k]);
+      L.ys[k] = v;
+      if (yg != nullptr) yg[(size_t)b * FC_OUT + k] = v;
+      if (MODE == MODE_TRAIN) {
+        const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
+        L.dzs[k] = d;
+        L.sq[k] = d * d;
+        dzg[(size_t)b * FC_OUT + k] = d;
+      }
+    }
   }
   __syncthreads();
 
@@ -145,22 +167,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   }
   if (MODE == MODE_INFER) return;
 
-  // ---- residual loss gradient + loss metric ----
-  if (tid < FC_OUT) {
-    const float d = (tid == labels[b] ? 1.0f : 0.0f) - L.ys[tid];
-    L.dzs[tid] = d;
-    L.sq[tid] = d * d;
-    dzg[(size_t)b * FC_OUT + tid] = d;
-  }
-  __syncthreads();
-  if (tid == 0 && loss_accum != nullptr) {
-    float s = 0.f;
-#pragma unroll
-    for (int k = 0; k < FC_OUT; ++k) s += L.sq[k];
-    unsafeAtomicAdd(loss_accum, sqrtf(s));
-  }
-
-  // ---- fc backward-data -> pool preact gradient ----
+  // ---- phase 4: fc backward-data -> pool preact gradient ----
   if (tid < S1_OUT) {
     float da = 0.f;
 #pragma unroll
@@ -170,10 +177,15 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     const float d = da * v * (1.0f - v);
     L.dz2s[tid] = d;
     dz2g[(size_t)b * S1_OUT + tid] = d;
+  } else if (tid == S1_OUT && loss_accum != nullptr) {
+    float s = 0.f;
+#pragma unroll
+    for (int k = 0; k < FC_OUT; ++k) s += L.sq[k];
+    unsafeAtomicAdd(loss_accum, sqrtf(s));
   }
   __syncthreads();
 
-  // ---- pool backward-data -> conv1 preact gradient ----
+  // ---- phase 5: pool backward-data -> conv1 preact gradient ----
   // stride == kernel: each conv1 output feeds exactly one pool cell (gather).
   for (int t = tid; t < C1_OUT; t += 256) {
     const int o = t / C1_PIX;
@@ -188,148 +200,117 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 }
 
 // ---------------------------------------------------------------------------
-// Weight gradients, batch-reduced.
+// Weight gradients, batch-reduced (v2 — barrier-free grid-stride).
 //
-// Grid role layout with NC = ceil(B / CHUNK) batch-chunks:
-//   blocks [0, 6*NC)      : conv1 wgrad+bgrad, one block per (channel, chunk)
-//   blocks [6*NC, 7*NC)   : pool wgrad+bgrad, one block per chunk
-//   blocks [7*NC, 8*NC)   : fc wgrad+bgrad, one block per chunk
-// Each block accumulates over its chunk in registers, then atomicAdds into
-// the flat fp32 gradient bucket (conflicts only across chunks).
+// Grid role layout (GC = blocks per conv channel, GS = pool blocks):
+//   blocks [0, 6*GC)            : conv1 wgrad+bgrad, blk -> (channel, slice)
+//   blocks [6*GC, 6*GC+GS)      : pool wgrad+bgrad
+//   blocks [6*GC+GS, +9)        : fc wgrad+bgrad (2160 weights + 10 biases,
+//                                 one owner thread each — no atomics)
 // ---------------------------------------------------------------------------
 
-constexpr int WG_CHUNK_DEFAULT = 8;  // images per chunk (runtime-tunable)
+constexpr int WG_GC_DEFAULT = 8;   // conv1 slices per channel (runtime knob)
+constexpr int FC_BLOCKS = (FC_WSZ + 255) / 256;  // 9
+
+// Cross-lane sum over the full 64-lane wave.
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
 
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad(
     const act_t* __restrict__ x, const act_t* __restrict__ a1g,
     const act_t* __restrict__ a2g, const float* __restrict__ dzg,
     const float* __restrict__ dz2g, const float* __restrict__ dz1g,
-    float* __restrict__ grads, int B, int NC, int CIMG) {
+    float* __restrict__ grads, int B, int GC, int GS) {
   const int tid = threadIdx.x;
   const int blk = blockIdx.x;
+  const int lane = tid & 63;
 
-  if (blk < C1_CH * NC) {
+  if (blk < C1_CH * GC) {
     // ---- conv1: dW[o,i,j] = sum_{b,r,c} dz1[b,o,r,c] * x[b,r+i,c+j] / 576
-    const int o = blk / NC;
-    const int chunk = blk - o * NC;
-    const int b_lo = chunk * CIMG;
-    const int b_hi = min(B, b_lo + CIMG);
-    __shared__ float S[IN_PIX + C1_K * C1_K + 1];
-    float* xs = S;
-    float* wacc = S + IN_PIX;     // [25] block-level accumulators
-    float* bacc = wacc + C1_K * C1_K;
-    if (tid < C1_K * C1_K + 1) wacc[tid] = 0.f;
+    const int o = blk / GC;
+    const int slice = blk - o * GC;
     float acc[C1_K * C1_K];
 #pragma unroll
     for (int w = 0; w < C1_K * C1_K; ++w) acc[w] = 0.f;
-    float bsum = 0.f;
-    for (int b = b_lo; b < b_hi; ++b) {
-      __syncthreads();  // protect xs reload
-      const act_t* xb = x + (size_t)b * IN_PIX;
-      for (int i = tid; i < IN_PIX; i += 256) xs[i] = ldf(xb + i);
-      __syncthreads();
-      const float* dz1b = dz1g + (size_t)b * C1_OUT + o * C1_PIX;
-      for (int t = tid; t < C1_PIX; t += 256) {
-        const int r = t / C1_W;
-        const int c = t - r * C1_W;
-        const float d = dz1b[t];
-        bsum += d;
+    float bacc = 0.f;
+    const int N = B * C1_PIX;
+    for (int it = slice * 256 + tid; it < N; it += GC * 256) {
+      const int b = it / C1_PIX;
+      const int pos = it - b * C1_PIX;
+      const int r = pos / C1_W;
+      const int c = pos - r * C1_W;
+      const float d = dz1g[(size_t)b * C1_OUT + o * C1_PIX + pos];
+      bacc += d;
+      const act_t* xb = x + (size_t)b * IN_PIX + r * IN_W + c;
 #pragma unroll
-        for (int i = 0; i < C1_K; ++i)
+      for (int i = 0; i < C1_K; ++i)
 #pragma unroll
-          for (int j = 0; j < C1_K; ++j)
-            acc[i * C1_K + j] += d * xs[(r + i) * IN_W + (c + j)];
-      }
+        for (int j = 0; j < C1_K; ++j)
+          acc[i * C1_K + j] += d * ldf(xb + i * IN_W + j);
     }
-    __syncthreads();
-#pragma unroll
-    for (int w = 0; w < C1_K * C1_K; ++w) atomicAdd(&wacc[w], acc[w]);
-    atomicAdd(bacc, bsum);
-    __syncthreads();
     constexpr float inv_pix = 1.0f / (float)C1_PIX;
-    if (tid < C1_K * C1_K)
-      unsafeAtomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + tid], wacc[tid] * inv_pix);
-    if (tid == C1_K * C1_K) unsafeAtomicAdd(&grads[OFF_C1B + o], bacc[0] * inv_pix);
-  } else if (blk < (C1_CH + 1) * NC) {
+#pragma unroll
+    for (int w = 0; w < C1_K * C1_K; ++w) {
+      const float s = wave_sum(acc[w]);
+      if (lane == 0)
+        unsafeAtomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + w], s * inv_pix);
+    }
+    const float bs = wave_sum(bacc);
+    if (lane == 0) unsafeAtomicAdd(&grads[OFF_C1B + o], bs * inv_pix);
+  } else if (blk < C1_CH * GC + GS) {
     // ---- pool: dW[i,j] = sum_{b,o,p,q} dz2[b,o,p,q] * a1[b,o,4p+i,4q+j]
-    const int chunk = blk - C1_CH * NC;
-    const int b_lo = chunk * CIMG;
-    const int b_hi = min(B, b_lo + CIMG);
-    __shared__ float S[C1_OUT + S1_WSZ + 1];
-    float* a1s = S;
-    float* wacc = S + C1_OUT;
-    float* bacc = wacc + S1_WSZ;
-    if (tid < S1_WSZ + 1) wacc[tid] = 0.f;
+    const int slice = blk - C1_CH * GC;
     float acc[S1_WSZ];
 #pragma unroll
     for (int w = 0; w < S1_WSZ; ++w) acc[w] = 0.f;
-    float bsum = 0.f;
-    for (int b = b_lo; b < b_hi; ++b) {
-      __syncthreads();
-      const act_t* a1b = a1g + (size_t)b * C1_OUT;
-      for (int i = tid; i < C1_OUT; i += 256) a1s[i] = ldf(a1b + i);
-      __syncthreads();
-      if (tid < S1_OUT) {
-        const int o = tid / S1_PIX;
-        const int pq = tid - o * S1_PIX;
-        const int pr = pq / S1_W;
-        const int pc = pq - pr * S1_W;
-        const float d = dz2g[(size_t)b * S1_OUT + tid];
-        bsum += d;
-        const float* base = &a1s[o * C1_PIX + pr * S1_K * C1_W + pc * S1_K];
+    float bacc = 0.f;
+    const int N = B * S1_OUT;
+    for (int it = slice * 256 + tid; it < N; it += GS * 256) {
+      const int b = it / S1_OUT;
+      const int opq = it - b * S1_OUT;
+      const int o = opq / S1_PIX;
+      const int pq = opq - o * S1_PIX;
+      const int pr = pq / S1_W;
+      const int pc = pq - pr * S1_W;
+      const float d = dz2g[it];
+      bacc += d;
+      const act_t* base = a1g + (size_t)b * C1_OUT + o * C1_PIX +
+                          pr * S1_K * C1_W + pc * S1_K;
 #pragma unroll
-        for (int i = 0; i < S1_K; ++i)
+      for (int i = 0; i < S1_K; ++i)
 #pragma unroll
-          for (int j = 0; j < S1_K; ++j)
-            acc[i * S1_K + j] += d * base[i * C1_W + j];
-      }
+        for (int j = 0; j < S1_K; ++j)
+          acc[i * S1_K + j] += d * ldf(base + i * C1_W + j);
     }
-    __syncthreads();
 #pragma unroll
-    for (int w = 0; w < S1_WSZ; ++w) atomicAdd(&wacc[w], acc[w]);
-    atomicAdd(bacc, bsum);
-    __syncthreads();
-    if (tid < S1_WSZ) unsafeAtomicAdd(&grads[OFF_S1W + tid], wacc[tid]);
-    if (tid == S1_WSZ)
-      unsafeAtomicAdd(&grads[OFF_S1B], bacc[0] / (float)S1_OUT);
+    for (int w = 0; w < S1_WSZ; ++w) {
+      const float s = wave_sum(acc[w]);
+      if (lane == 0) unsafeAtomicAdd(&grads[OFF_S1W + w], s);
+    }
+    const float bs = wave_sum(bacc);
+    if (lane == 0)
+      unsafeAtomicAdd(&grads[OFF_S1B], bs / (float)S1_OUT);
   } else {
     // ---- fc: dW[k,m] = sum_b dz[b,k] * a2[b,m];  db[k] = sum_b dz[b,k]
-    const int chunk = blk - (C1_CH + 1) * NC;
-    const int b_lo = chunk * CIMG;
-    const int b_hi = min(B, b_lo + CIMG);
-    __shared__ float S[S1_OUT + FC_OUT];
-    float* a2s = S;
-    float* dzs = S + S1_OUT;
-    constexpr int NACC = (FC_WSZ + 255) / 256;  // 9 (k,m) pairs per thread
-    float acc[NACC];
-#pragma unroll
-    for (int u = 0; u < NACC; ++u) acc[u] = 0.f;
-    float bsum = 0.f;  // threads 0..9 hold fc bias grad
-    for (int b = b_lo; b < b_hi; ++b) {
-      __syncthreads();
-      const act_t* a2b = a2g + (size_t)b * S1_OUT;
-      if (tid < S1_OUT) a2s[tid] = ldf(a2b + tid);
-      if (tid >= S1_OUT && tid < S1_OUT + FC_OUT)
-        dzs[tid - S1_OUT] = dzg[(size_t)b * FC_OUT + (tid - S1_OUT)];
-      __syncthreads();
-#pragma unroll
-      for (int u = 0; u < NACC; ++u) {
-        const int q = tid + u * 256;
-        if (q < FC_WSZ) {
-          const int k = q / FC_IN;
-          const int m = q - k * FC_IN;
-          acc[u] += dzs[k] * a2s[m];
-        }
-      }
-      if (tid < FC_OUT) bsum += dzs[tid];
+    // One owner thread per weight/bias: plain accumulate, no atomics.
+    const int q = (blk - C1_CH * GC - GS) * 256 + tid;
+    if (q < FC_WSZ) {
+      const int k = q / FC_IN;
+      const int m = q - k * FC_IN;
+      float acc = 0.f;
+      for (int b = 0; b < B; ++b)
+        acc += dzg[(size_t)b * FC_OUT + k] * ldf(a2g + (size_t)b * FC_IN + m);
+      grads[OFF_FW + q] += acc;
+    } else if (q < FC_WSZ + FC_OUT) {
+      const int k = q - FC_WSZ;
+      float acc = 0.f;
+      for (int b = 0; b < B; ++b) acc += dzg[(size_t)b * FC_OUT + k];
+      grads[OFF_FB + k] += acc;
     }
-#pragma unroll
-    for (int u = 0; u < NACC; ++u) {
-      const int q = tid + u * 256;
-      if (q < FC_WSZ) unsafeAtomicAdd(&grads[OFF_FW + q], acc[u]);
-    }
-    if (tid < FC_OUT) unsafeAtomicAdd(&grads[OFF_FB + tid], bsum);
   }
 }
 
@@ -398,22 +379,23 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
   }
 }
 
+// chunk_imgs is the conv1 slices-per-channel knob (GC); <=0 -> default.
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
                       void* stream) {
-  const int CIMG = chunk_imgs > 0 ? chunk_imgs : WG_CHUNK_DEFAULT;
-  const int NC = (B + CIMG - 1) / CIMG;
-  dim3 grid((C1_CH + 2) * NC), block(256);
+  const int GC = chunk_imgs > 0 ? chunk_imgs : WG_GC_DEFAULT;
+  const int GS = GC / 4 > 2 ? GC / 4 : 2;
+  dim3 grid(C1_CH * GC + GS + FC_BLOCKS), block(256);
   hipStream_t s = (hipStream_t)stream;
   if (act_is_bf16) {
     hipLaunchKernelGGL((k_wgrad<bf16>), grid, block, 0, s, (const bf16*)x,
                        (const bf16*)a1, (const bf16*)a2, dz, dz2, dz1, grads,
-                       B, NC, CIMG);
+                       B, GC, GS);
   } else {
     hipLaunchKernelGGL((k_wgrad<float>), grid, block, 0, s, (const float*)x,
                        (const float*)a1, (const float*)a2, dz, dz2, dz1, grads,
-                       B, NC, CIMG);
+                       B, GC, GS);
   }
   return (int)hipGetLastError();
 }
