@@ -36,7 +36,7 @@ def main() -> int:
     p.add_argument("--act-dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default="auto")
-    p.add_argument("--wgrad-chunk", type=int, default=8)
+    p.add_argument("--wgrad-chunk", type=int, default=0)
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig

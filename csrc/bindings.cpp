@@ -29,6 +29,10 @@ int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
                       void* stream);
+int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
+                         const float* dz, const float* dz2, const float* dz1,
+                         float* grads, int B, int act_is_bf16, int chunk_imgs,
+                         int roles, void* stream);
 int pcnn_launch_update(float* params, float* grads, float step, void* stream);
 const char* pcnn_hip_error_string(int err);
 }
@@ -78,6 +82,20 @@ void hip_wgrad(at::Tensor x, at::Tensor a1, at::Tensor a2, at::Tensor dz,
                               dz1.data_ptr<float>(), grads.data_ptr<float>(),
                               (int)B, f, (int)chunk_imgs, (void*)stream),
             "wgrad");
+}
+
+// roles bitmask (1=conv1, 2=pool, 4=fc) — ablation/diagnostics only.
+void hip_wgrad_roles(at::Tensor x, at::Tensor a1, at::Tensor a2,
+                     at::Tensor dz, at::Tensor dz2, at::Tensor dz1,
+                     at::Tensor grads, int64_t B, int64_t chunk_imgs,
+                     int64_t roles, int64_t stream) {
+  int f = act_flag(x);
+  check_hip(pcnn_launch_wgrad_ex(x.data_ptr(), a1.data_ptr(), a2.data_ptr(),
+                                 dz.data_ptr<float>(), dz2.data_ptr<float>(),
+                                 dz1.data_ptr<float>(), grads.data_ptr<float>(),
+                                 (int)B, f, (int)chunk_imgs, (int)roles,
+                                 (void*)stream),
+            "wgrad_roles");
 }
 
 void hip_update(at::Tensor params, at::Tensor grads, double step,
@@ -140,6 +158,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cpu_update", &pcnn::cpu_update);
   m.def("hip_fwdbwd", &hip_fwdbwd);
   m.def("hip_wgrad", &hip_wgrad);
+  m.def("hip_wgrad_roles", &hip_wgrad_roles);
   m.def("hip_update", &hip_update);
   m.def("hip_train_steps", &hip_train_steps);
   m.attr("N_PARAMS") = pcnn::N_PARAMS;

@@ -224,12 +224,13 @@ __global__ __launch_bounds__(256) void k_wgrad(
     const act_t* __restrict__ x, const act_t* __restrict__ a1g,
     const act_t* __restrict__ a2g, const float* __restrict__ dzg,
     const float* __restrict__ dz2g, const float* __restrict__ dz1g,
-    float* __restrict__ grads, int B, int GC, int GS) {
+    float* __restrict__ grads, int B, int GC, int GS, int FS, int roles) {
   const int tid = threadIdx.x;
   const int blk = blockIdx.x;
   const int lane = tid & 63;
 
   if (blk < C1_CH * GC) {
+    if (!(roles & 1)) return;
     // ---- conv1: dW[o,i,j] = sum_{b,r,c} dz1[b,o,r,c] * x[b,r+i,c+j] / 576
     const int o = blk / GC;
     const int slice = blk - o * GC;
@@ -262,6 +263,7 @@ __global__ __launch_bounds__(256) void k_wgrad(
     const float bs = wave_sum(bacc);
     if (lane == 0) unsafeAtomicAdd(&grads[OFF_C1B + o], bs * inv_pix);
   } else if (blk < C1_CH * GC + GS) {
+    if (!(roles & 2)) return;
     // ---- pool: dW[i,j] = sum_{b,o,p,q} dz2[b,o,p,q] * a1[b,o,4p+i,4q+j]
     const int slice = blk - C1_CH * GC;
     float acc[S1_WSZ];
@@ -295,21 +297,36 @@ __global__ __launch_bounds__(256) void k_wgrad(
     if (lane == 0)
       unsafeAtomicAdd(&grads[OFF_S1B], bs / (float)S1_OUT);
   } else {
+    if (!(roles & 4)) return;
     // ---- fc: dW[k,m] = sum_b dz[b,k] * a2[b,m];  db[k] = sum_b dz[b,k]
-    // One owner thread per weight/bias: plain accumulate, no atomics.
-    const int q = (blk - C1_CH * GC - GS) * 256 + tid;
+    // FS batch slices; one owner thread per (weight, slice).  With FS==1
+    // the owner accumulates with a plain store (no atomics); FS>1 (large
+    // batch) uses hardware atomics across slices.
+    const int fblk = blk - C1_CH * GC - GS;
+    const int slice = fblk / FC_BLOCKS;
+    const int q = (fblk - slice * FC_BLOCKS) * 256 + tid;
+    const int b_lo = (int)(((long)B * slice) / FS);
+    const int b_hi = (int)(((long)B * (slice + 1)) / FS);
     if (q < FC_WSZ) {
       const int k = q / FC_IN;
       const int m = q - k * FC_IN;
       float acc = 0.f;
-      for (int b = 0; b < B; ++b)
+#pragma unroll 8
+      for (int b = b_lo; b < b_hi; ++b)
         acc += dzg[(size_t)b * FC_OUT + k] * ldf(a2g + (size_t)b * FC_IN + m);
-      grads[OFF_FW + q] += acc;
+      if (FS == 1)
+        grads[OFF_FW + q] += acc;
+      else
+        unsafeAtomicAdd(&grads[OFF_FW + q], acc);
     } else if (q < FC_WSZ + FC_OUT) {
       const int k = q - FC_WSZ;
       float acc = 0.f;
-      for (int b = 0; b < B; ++b) acc += dzg[(size_t)b * FC_OUT + k];
-      grads[OFF_FB + k] += acc;
+#pragma unroll 8
+      for (int b = b_lo; b < b_hi; ++b) acc += dzg[(size_t)b * FC_OUT + k];
+      if (FS == 1)
+        grads[OFF_FB + k] += acc;
+      else
+        unsafeAtomicAdd(&grads[OFF_FB + k], acc);
     }
   }
 }
@@ -384,18 +401,33 @@ int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
                       void* stream) {
-  const int GC = chunk_imgs > 0 ? chunk_imgs : WG_GC_DEFAULT;
-  const int GS = GC / 4 > 2 ? GC / 4 : 2;
-  dim3 grid(C1_CH * GC + GS + FC_BLOCKS), block(256);
+  return pcnn_launch_wgrad_ex(x, a1, a2, dz, dz2, dz1, grads, B, act_is_bf16,
+                              chunk_imgs, 7, stream);
+}
+
+int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
+                         const float* dz, const float* dz2, const float* dz1,
+                         float* grads, int B, int act_is_bf16, int chunk_imgs,
+                         int roles, void* stream) {
+  // Batch-adaptive defaults: ~36 (image,position) items per conv thread,
+  // pool slices at GC/4, fc batch slices of ~64 images.
+  int GC = chunk_imgs > 0 ? chunk_imgs : (B + 15) / 16;
+  if (GC < 1) GC = 1;
+  if (GC > 384) GC = 384;
+  int GS = GC / 4 > 2 ? GC / 4 : 2;
+  int FS = B / 64;
+  if (FS < 1) FS = 1;
+  if (FS > 32) FS = 32;
+  dim3 grid(C1_CH * GC + GS + FS * FC_BLOCKS), block(256);
   hipStream_t s = (hipStream_t)stream;
   if (act_is_bf16) {
     hipLaunchKernelGGL((k_wgrad<bf16>), grid, block, 0, s, (const bf16*)x,
                        (const bf16*)a1, (const bf16*)a2, dz, dz2, dz1, grads,
-                       B, GC, GS);
+                       B, GC, GS, FS, roles);
   } else {
     hipLaunchKernelGGL((k_wgrad<float>), grid, block, 0, s, (const float*)x,
                        (const float*)a1, (const float*)a2, dz, dz2, dz1, grads,
-                       B, GC, GS);
+                       B, GC, GS, FS, roles);
   }
   return (int)hipGetLastError();
 }

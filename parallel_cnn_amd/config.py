@@ -38,7 +38,7 @@ class TrainConfig:
     # execution
     backend: str = "auto"            # auto | hip | cpu | torchref
     device: str = "auto"             # auto | cuda | cpu
-    wgrad_chunk: int = 8             # images per weight-grad workgroup chunk
+    wgrad_chunk: int = 0             # conv1 wgrad slices/channel (0 = auto)
 
     # io / observability
     log_interval: int = 100          # steps between loss readouts
