@@ -397,14 +397,6 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
 }
 
 // chunk_imgs is the conv1 slices-per-channel knob (GC); <=0 -> default.
-int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
-                      const float* dz, const float* dz2, const float* dz1,
-                      float* grads, int B, int act_is_bf16, int chunk_imgs,
-                      void* stream) {
-  return pcnn_launch_wgrad_ex(x, a1, a2, dz, dz2, dz1, grads, B, act_is_bf16,
-                              chunk_imgs, 7, stream);
-}
-
 int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
                          const float* dz, const float* dz2, const float* dz1,
                          float* grads, int B, int act_is_bf16, int chunk_imgs,
@@ -430,6 +422,14 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
                        B, GC, GS, FS, roles);
   }
   return (int)hipGetLastError();
+}
+
+int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
+                      const float* dz, const float* dz2, const float* dz1,
+                      float* grads, int B, int act_is_bf16, int chunk_imgs,
+                      void* stream) {
+  return pcnn_launch_wgrad_ex(x, a1, a2, dz, dz2, dz1, grads, B, act_is_bf16,
+                              chunk_imgs, 7, stream);
 }
 
 int pcnn_launch_update(float* params, float* grads, float step, void* stream) {
