@@ -53,12 +53,38 @@ __device__ __forceinline__ void stf(T* p, float v) {
   *p = (T)v;
 }
 
+// Vectorized loads of 4/8 consecutive activations as fp32.  Callers
+// guarantee 8-byte (bf16) / 16-byte (fp32) alignment of p.
+__device__ __forceinline__ void ld4f(const bf16* p, float* out) {
+  const ushort4 v = *reinterpret_cast<const ushort4*>(p);
+  const bf16* e = reinterpret_cast<const bf16*>(&v);
+  out[0] = (float)e[0];
+  out[1] = (float)e[1];
+  out[2] = (float)e[2];
+  out[3] = (float)e[3];
+}
+__device__ __forceinline__ void ld4f(const float* p, float* out) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  out[0] = v.x;
+  out[1] = v.y;
+  out[2] = v.z;
+  out[3] = v.w;
+}
+template <typename T>
+__device__ __forceinline__ void ld8f(const T* p, float* out) {
+  ld4f(p, out);
+  ld4f(p + 4, out + 4);
+}
+
 // Execution modes for the fused forward kernel.
 enum Mode { MODE_TRAIN = 0, MODE_EVAL = 1, MODE_INFER = 2 };
 
 // LDS image of the per-image state.  One single __shared__ object.
+// Only the conv/pool parameters (173 floats) are staged; the fc weight
+// matrix stays in global memory (L2-resident, read coalesced once per
+// phase) — staging it cost a full extra LDS round and 9.4 KB of occupancy.
 struct FwdLds {
-  float ps[N_PARAMS];   // staged parameters
+  float ps[OFF_FW];     // staged conv1+pool parameters
   float xs[IN_PIX];     // input image (fp32)
   float a1s[C1_OUT];    // conv1 activation
   float a2s[S1_OUT];    // pool activation
@@ -80,8 +106,8 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   if (b >= B) return;
   const int tid = threadIdx.x;
 
-  // ---- phase 0: stage parameters and the input image into LDS ----
-  for (int i = tid; i < N_PARAMS; i += 256) L.ps[i] = params[i];
+  // ---- phase 0: stage conv/pool parameters and the input image ----
+  if (tid < OFF_FW) L.ps[tid] = params[tid];
   const act_t* xb = x + (size_t)b * IN_PIX;
   for (int i = tid; i < IN_PIX; i += 256) L.xs[i] = ldf(xb + i);
   __syncthreads();
@@ -131,7 +157,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   if (tid < FC_OUT * 16) {
     const int k = tid >> 4;
     const int l = tid & 15;
-    const float* wk = &L.ps[OFF_FW + k * FC_IN];
+    const float* wk = params + OFF_FW + k * FC_IN;
     float p = 0.f;
 #pragma unroll
     for (int u = 0; u < (FC_IN + 15) / 16; ++u) {
@@ -141,7 +167,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 #pragma unroll
     for (int off = 8; off > 0; off >>= 1) p += __shfl_down(p, off, 16);
     if (l == 0) {
-      const float v = sigmoidf_dev(p + L.ps[OFF_FB + k]);
+      const float v = sigmoidf_dev(p + params[OFF_FB + k]);
       L.ys[k] = v;
       if (yg != nullptr) yg[(size_t)b * FC_OUT + k] = v;
       if (MODE == MODE_TRAIN) {
@@ -172,7 +198,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     float da = 0.f;
 #pragma unroll
     for (int k = 0; k < FC_OUT; ++k)
-      da += L.ps[OFF_FW + k * FC_IN + tid] * L.dzs[k];
+      da += params[OFF_FW + k * FC_IN + tid] * L.dzs[k];
     const float v = L.a2s[tid];
     const float d = da * v * (1.0f - v);
     L.dz2s[tid] = d;
@@ -232,26 +258,36 @@ __global__ __launch_bounds__(256) void k_wgrad(
   if (blk < C1_CH * GC) {
     if (!(roles & 1)) return;
     // ---- conv1: dW[o,i,j] = sum_{b,r,c} dz1[b,o,r,c] * x[b,r+i,c+j] / 576
+    // Work item = 4 consecutive output columns of one row: 5 vectorized
+    // 8-wide x row loads + one float4 dz1 load replace 104 scalar loads
+    // (the v2 scalar form was address-issue-bound: profiles/ r1).
     const int o = blk / GC;
     const int slice = blk - o * GC;
+    constexpr int GROUPS_PER_ROW = C1_W / 4;  // 6
     float acc[C1_K * C1_K];
 #pragma unroll
     for (int w = 0; w < C1_K * C1_K; ++w) acc[w] = 0.f;
     float bacc = 0.f;
-    const int N = B * C1_PIX;
+    const int N = B * C1_H * GROUPS_PER_ROW;
     for (int it = slice * 256 + tid; it < N; it += GC * 256) {
-      const int b = it / C1_PIX;
-      const int pos = it - b * C1_PIX;
-      const int r = pos / C1_W;
-      const int c = pos - r * C1_W;
-      const float d = dz1g[(size_t)b * C1_OUT + o * C1_PIX + pos];
-      bacc += d;
-      const act_t* xb = x + (size_t)b * IN_PIX + r * IN_W + c;
+      const int b = it / (C1_H * GROUPS_PER_ROW);
+      const int rg = it - b * (C1_H * GROUPS_PER_ROW);
+      const int r = rg / GROUPS_PER_ROW;
+      const int c0 = (rg - r * GROUPS_PER_ROW) * 4;
+      float d4[4];
+      ld4f(dz1g + (size_t)b * C1_OUT + o * C1_PIX + r * C1_W + c0, d4);
+      bacc += d4[0] + d4[1] + d4[2] + d4[3];
+      const act_t* xb = x + (size_t)b * IN_PIX + r * IN_W + c0;
+      float xr[8];
 #pragma unroll
-      for (int i = 0; i < C1_K; ++i)
+      for (int i = 0; i < C1_K; ++i) {
+        ld8f(xb + i * IN_W, xr);
 #pragma unroll
-        for (int j = 0; j < C1_K; ++j)
-          acc[i * C1_K + j] += d * ldf(xb + i * IN_W + j);
+        for (int pp = 0; pp < 4; ++pp)
+#pragma unroll
+          for (int j = 0; j < C1_K; ++j)
+            acc[i * C1_K + j] += d4[pp] * xr[pp + j];
+      }
     }
     constexpr float inv_pix = 1.0f / (float)C1_PIX;
 #pragma unroll
@@ -282,11 +318,13 @@ __global__ __launch_bounds__(256) void k_wgrad(
       bacc += d;
       const act_t* base = a1g + (size_t)b * C1_OUT + o * C1_PIX +
                           pr * S1_K * C1_W + pc * S1_K;
+      float ar[4];
 #pragma unroll
-      for (int i = 0; i < S1_K; ++i)
+      for (int i = 0; i < S1_K; ++i) {
+        ld4f(base + i * C1_W, ar);
 #pragma unroll
-        for (int j = 0; j < S1_K; ++j)
-          acc[i * S1_K + j] += d * ldf(base + i * C1_W + j);
+        for (int j = 0; j < S1_K; ++j) acc[i * S1_K + j] += d * ar[j];
+      }
     }
 #pragma unroll
     for (int w = 0; w < S1_WSZ; ++w) {
@@ -403,9 +441,9 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
                          int roles, void* stream) {
   // Batch-adaptive defaults: ~36 (image,position) items per conv thread,
   // pool slices at GC/4, fc batch slices of ~64 images.
-  int GC = chunk_imgs > 0 ? chunk_imgs : (B + 15) / 16;
-  if (GC < 1) GC = 1;
-  if (GC > 384) GC = 384;
+  int GC = chunk_imgs > 0 ? chunk_imgs : (int)(2.0f * __builtin_cbrtf((float)B) + 0.5f);
+  if (GC < 2) GC = 2;
+  if (GC > 256) GC = 256;
   int GS = GC / 4 > 2 ? GC / 4 : 2;
   int FS = B / 64;
   if (FS < 1) FS = 1;
