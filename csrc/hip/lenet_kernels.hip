@@ -82,18 +82,25 @@ enum Mode { MODE_TRAIN = 0, MODE_EVAL = 1, MODE_INFER = 2 };
 // LDS image of the per-image state.  One single __shared__ object.
 // Only the conv/pool parameters (173 floats) are staged; the fc weight
 // matrix stays in global memory (L2-resident, read coalesced once per
-// phase) — staging it cost a full extra LDS round and 9.4 KB of occupancy.
+// phase).
 struct FwdLds {
   float ps[OFF_FW];     // staged conv1+pool parameters
   float xs[IN_PIX];     // input image (fp32)
-  float a1s[C1_OUT];    // conv1 activation
   float a2s[S1_OUT];    // pool activation
   float ys[FC_OUT];     // logits (post-sigmoid)
   float dzs[FC_OUT];    // residual gradient
   float sq[FC_OUT];     // per-class squared error
-  float dz2s[S1_OUT];   // pool preact gradient
 };
 
+// Fused forward + backward-data, one 256-thread workgroup per image.
+//
+// Dataflow: thread t < 216 OWNS pool cell (o, pr, pc) — it computes the
+// cell's 16 conv1 outputs (from the LDS image + staged weights), applies
+// sigmoid, keeps them in REGISTERS, computes the pool output, and later
+// runs the pool backward for the same 16 positions from those registers.
+// That ownership removes the conv->pool and fc-bwd->pool-bwd barriers and
+// all cross-thread a1 LDS traffic: the kernel has 3 barriers total
+// ({stage} {conv+pool} {fc+residual} {fc-bwd + pool-bwd + loss}).
 template <typename act_t, int MODE>
 __global__ __launch_bounds__(256) void k_fwdbwd(
     const act_t* __restrict__ x, const float* __restrict__ params,
@@ -112,48 +119,62 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   for (int i = tid; i < IN_PIX; i += 256) L.xs[i] = ldf(xb + i);
   __syncthreads();
 
-  // ---- phase 1: conv1 (5x5 valid, 6 ch) + sigmoid ----
-  for (int t = tid; t < C1_OUT; t += 256) {
-    const int o = t / C1_PIX;
-    const int rc = t - o * C1_PIX;
-    const int r = rc / C1_W;
-    const int c = rc - r * C1_W;
-    const float* w = &L.ps[OFF_C1W + o * C1_K * C1_K];
-    float acc = L.ps[OFF_C1B + o];
-#pragma unroll
-    for (int i = 0; i < C1_K; ++i)
-#pragma unroll
-      for (int j = 0; j < C1_K; ++j)
-        acc += w[i * C1_K + j] * L.xs[(r + i) * IN_W + (c + j)];
-    const float v = sigmoidf_dev(acc);
-    L.a1s[t] = v;
-    if (MODE == MODE_TRAIN) stf(a1g + (size_t)b * C1_OUT + t, v);
-  }
-  __syncthreads();
-
-  // ---- phase 2: trainable pool (4x4 stride 4, shared kernel) + sigmoid ----
+  // ---- phase 1: conv1 + sigmoid + pool + sigmoid (no barrier between) ----
+  const int o = tid / S1_PIX;
+  const int pq = tid - o * S1_PIX;
+  const int pr = pq / S1_W;
+  const int pc = pq - pr * S1_W;
+  float a1v[S1_K * S1_K];  // this cell's conv activations, kept live to bwd
+  float a2v = 0.f;
   if (tid < S1_OUT) {
-    const int o = tid / S1_PIX;
-    const int pq = tid - o * S1_PIX;
-    const int pr = pq / S1_W;
-    const int pc = pq - pr * S1_W;
-    const float* base = &L.a1s[o * C1_PIX + pr * S1_K * C1_W + pc * S1_K];
-    float acc = L.ps[OFF_S1B];
+    // load the cell's 8x8 input window into registers (16 x b128 LDS reads)
+    float xw[8][8];
 #pragma unroll
-    for (int i = 0; i < S1_K; ++i)
+    for (int u = 0; u < 8; ++u)
 #pragma unroll
-      for (int j = 0; j < S1_K; ++j)
-        acc += L.ps[OFF_S1W + i * S1_K + j] * base[i * C1_W + j];
-    const float v = sigmoidf_dev(acc);
-    L.a2s[tid] = v;
-    if (MODE == MODE_TRAIN) stf(a2g + (size_t)b * S1_OUT + tid, v);
+      for (int v4 = 0; v4 < 2; ++v4)
+        ld4f(&L.xs[(pr * S1_K + u) * IN_W + pc * S1_K + v4 * 4],
+             &xw[u][v4 * 4]);
+    const float* w = &L.ps[OFF_C1W + o * C1_K * C1_K];
+    const float cb = L.ps[OFF_C1B + o];
+    float pacc = L.ps[OFF_S1B];
+#pragma unroll
+    for (int i = 0; i < S1_K; ++i) {
+#pragma unroll
+      for (int j = 0; j < S1_K; ++j) {
+        float acc = cb;
+#pragma unroll
+        for (int u = 0; u < C1_K; ++u)
+#pragma unroll
+          for (int v = 0; v < C1_K; ++v)
+            acc += w[u * C1_K + v] * xw[i + u][j + v];
+        const float av = sigmoidf_dev(acc);
+        a1v[i * S1_K + j] = av;
+        pacc += L.ps[OFF_S1W + i * S1_K + j] * av;
+      }
+    }
+    if (MODE == MODE_TRAIN) {
+      // 4-wide activation stores per conv row
+      act_t packed[S1_K];
+#pragma unroll
+      for (int i = 0; i < S1_K; ++i) {
+#pragma unroll
+        for (int j = 0; j < S1_K; ++j) packed[j] = (act_t)a1v[i * S1_K + j];
+        *reinterpret_cast<uint2*>(
+            a1g + (size_t)b * C1_OUT + o * C1_PIX +
+            (pr * S1_K + i) * C1_W + pc * S1_K) =
+            *reinterpret_cast<const uint2*>(packed);
+      }
+    }
+    a2v = sigmoidf_dev(pacc);
+    L.a2s[tid] = a2v;
+    if (MODE == MODE_TRAIN) stf(a2g + (size_t)b * S1_OUT + tid, a2v);
   }
   __syncthreads();
 
-  // ---- phase 3: fc (216 -> 10) + sigmoid [+ loss residual] ----
-  // 16 lanes per output class: lane l of group k sums m = l, l+16, ...
-  // then a 4-step shuffle tree; the group leader applies bias+sigmoid and
-  // (TRAIN) computes the residual immediately — no extra barrier phase.
+  // ---- phase 2: fc (216 -> 10) + sigmoid [+ loss residual] ----
+  // 16 lanes per output class, 4-step shuffle tree; the group leader
+  // applies bias+sigmoid and computes the residual in-phase.
   if (tid < FC_OUT * 16) {
     const int k = tid >> 4;
     const int l = tid & 15;
@@ -193,35 +214,32 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   }
   if (MODE == MODE_INFER) return;
 
-  // ---- phase 4: fc backward-data -> pool preact gradient ----
+  // ---- phase 3: fc bwd -> pool preact grad -> pool bwd (no barriers) ----
   if (tid < S1_OUT) {
     float da = 0.f;
 #pragma unroll
     for (int k = 0; k < FC_OUT; ++k)
       da += params[OFF_FW + k * FC_IN + tid] * L.dzs[k];
-    const float v = L.a2s[tid];
-    const float d = da * v * (1.0f - v);
-    L.dz2s[tid] = d;
-    dz2g[(size_t)b * S1_OUT + tid] = d;
-  } else if (tid == S1_OUT && loss_accum != nullptr) {
-    float s = 0.f;
+    const float d2 = da * a2v * (1.0f - a2v);
+    dz2g[(size_t)b * S1_OUT + tid] = d2;
+    // pool backward for this thread's own 16 conv positions (registers)
 #pragma unroll
-    for (int k = 0; k < FC_OUT; ++k) s += L.sq[k];
-    unsafeAtomicAdd(loss_accum, sqrtf(s));
-  }
-  __syncthreads();
-
-  // ---- phase 5: pool backward-data -> conv1 preact gradient ----
-  // stride == kernel: each conv1 output feeds exactly one pool cell (gather).
-  for (int t = tid; t < C1_OUT; t += 256) {
-    const int o = t / C1_PIX;
-    const int rc = t - o * C1_PIX;
-    const int r = rc / C1_W;
-    const int c = rc - r * C1_W;
-    const float da = L.dz2s[o * S1_PIX + (r / S1_K) * S1_W + (c / S1_K)] *
-                     L.ps[OFF_S1W + (r % S1_K) * S1_K + (c % S1_K)];
-    const float v = L.a1s[t];
-    dz1g[(size_t)b * C1_OUT + t] = da * v * (1.0f - v);
+    for (int i = 0; i < S1_K; ++i) {
+      float row[S1_K];
+#pragma unroll
+      for (int j = 0; j < S1_K; ++j) {
+        const float av = a1v[i * S1_K + j];
+        row[j] = d2 * L.ps[OFF_S1W + i * S1_K + j] * av * (1.0f - av);
+      }
+      *reinterpret_cast<float4*>(
+          dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
+          pc * S1_K) = *reinterpret_cast<const float4*>(row);
+    }
+  } else if (tid == S1_OUT && loss_accum != nullptr) {
+    float ssum = 0.f;
+#pragma unroll
+    for (int k = 0; k < FC_OUT; ++k) ssum += L.sq[k];
+    unsafeAtomicAdd(loss_accum, sqrtf(ssum));
   }
 }
 
@@ -260,7 +278,10 @@ __global__ __launch_bounds__(256) void k_wgrad(
     // ---- conv1: dW[o,i,j] = sum_{b,r,c} dz1[b,o,r,c] * x[b,r+i,c+j] / 576
     // Work item = 4 consecutive output columns of one row: 5 vectorized
     // 8-wide x row loads + one float4 dz1 load replace 104 scalar loads
-    // (the v2 scalar form was address-issue-bound: profiles/ r1).
+    // (the v2 scalar form was address-issue-bound: profiles/ r1).  Two
+    // groups are processed per loop iteration so the second group's loads
+    // issue before the first group's FMAs — one memory-latency stall per
+    // TWO groups instead of one per group.
     const int o = blk / GC;
     const int slice = blk - o * GC;
     constexpr int GROUPS_PER_ROW = C1_W / 4;  // 6
@@ -269,35 +290,73 @@ __global__ __launch_bounds__(256) void k_wgrad(
     for (int w = 0; w < C1_K * C1_K; ++w) acc[w] = 0.f;
     float bacc = 0.f;
     const int N = B * C1_H * GROUPS_PER_ROW;
-    for (int it = slice * 256 + tid; it < N; it += GC * 256) {
-      const int b = it / (C1_H * GROUPS_PER_ROW);
-      const int rg = it - b * (C1_H * GROUPS_PER_ROW);
-      const int r = rg / GROUPS_PER_ROW;
-      const int c0 = (rg - r * GROUPS_PER_ROW) * 4;
-      float d4[4];
-      ld4f(dz1g + (size_t)b * C1_OUT + o * C1_PIX + r * C1_W + c0, d4);
-      bacc += d4[0] + d4[1] + d4[2] + d4[3];
-      const act_t* xb = x + (size_t)b * IN_PIX + r * IN_W + c0;
-      float xr[8];
+    const int stride = GC * 256;
+    for (int it = slice * 256 + tid; it < N; it += 2 * stride) {
+      const int it2 = it + stride;
+      float d4a[4], xra[C1_K][8];
+      float d4b[4], xrb[C1_K][8];
+      {
+        const int bi = it / (C1_H * GROUPS_PER_ROW);
+        const int rg = it - bi * (C1_H * GROUPS_PER_ROW);
+        const int r = rg / GROUPS_PER_ROW;
+        const int c0 = (rg - r * GROUPS_PER_ROW) * 4;
+        ld4f(dz1g + (size_t)bi * C1_OUT + o * C1_PIX + r * C1_W + c0, d4a);
+        const act_t* xb = x + (size_t)bi * IN_PIX + r * IN_W + c0;
 #pragma unroll
-      for (int i = 0; i < C1_K; ++i) {
-        ld8f(xb + i * IN_W, xr);
+        for (int i = 0; i < C1_K; ++i) ld8f(xb + i * IN_W, xra[i]);
+      }
+      if (it2 < N) {
+        const int bi = it2 / (C1_H * GROUPS_PER_ROW);
+        const int rg = it2 - bi * (C1_H * GROUPS_PER_ROW);
+        const int r = rg / GROUPS_PER_ROW;
+        const int c0 = (rg - r * GROUPS_PER_ROW) * 4;
+        ld4f(dz1g + (size_t)bi * C1_OUT + o * C1_PIX + r * C1_W + c0, d4b);
+        const act_t* xb = x + (size_t)bi * IN_PIX + r * IN_W + c0;
+#pragma unroll
+        for (int i = 0; i < C1_K; ++i) ld8f(xb + i * IN_W, xrb[i]);
+      }
+      bacc += d4a[0] + d4a[1] + d4a[2] + d4a[3];
+#pragma unroll
+      for (int i = 0; i < C1_K; ++i)
 #pragma unroll
         for (int pp = 0; pp < 4; ++pp)
 #pragma unroll
           for (int j = 0; j < C1_K; ++j)
-            acc[i * C1_K + j] += d4[pp] * xr[pp + j];
+            acc[i * C1_K + j] += d4a[pp] * xra[i][pp + j];
+      if (it2 < N) {
+        bacc += d4b[0] + d4b[1] + d4b[2] + d4b[3];
+#pragma unroll
+        for (int i = 0; i < C1_K; ++i)
+#pragma unroll
+          for (int pp = 0; pp < 4; ++pp)
+#pragma unroll
+            for (int j = 0; j < C1_K; ++j)
+              acc[i * C1_K + j] += d4b[pp] * xrb[i][pp + j];
       }
     }
-    constexpr float inv_pix = 1.0f / (float)C1_PIX;
+    // cross-wave pre-reduce in LDS: one hardware atomic per weight per
+    // BLOCK (was one per wave -> 4x the same-address atomic traffic).
+    __shared__ float red[4][C1_K * C1_K + 1];
+    const int wv = tid >> 6;
 #pragma unroll
     for (int w = 0; w < C1_K * C1_K; ++w) {
-      const float s = wave_sum(acc[w]);
-      if (lane == 0)
-        unsafeAtomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + w], s * inv_pix);
+      const float v = wave_sum(acc[w]);
+      if (lane == 0) red[wv][w] = v;
     }
-    const float bs = wave_sum(bacc);
-    if (lane == 0) unsafeAtomicAdd(&grads[OFF_C1B + o], bs * inv_pix);
+    {
+      const float v = wave_sum(bacc);
+      if (lane == 0) red[wv][C1_K * C1_K] = v;
+    }
+    __syncthreads();
+    constexpr float inv_pix = 1.0f / (float)C1_PIX;
+    if (tid < C1_K * C1_K) {
+      const float v =
+          red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+      unsafeAtomicAdd(&grads[OFF_C1W + o * C1_K * C1_K + tid], v * inv_pix);
+    } else if (tid == C1_K * C1_K) {
+      const float v = red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+      unsafeAtomicAdd(&grads[OFF_C1B + o], v * inv_pix);
+    }
   } else if (blk < C1_CH * GC + GS) {
     if (!(roles & 2)) return;
     // ---- pool: dW[i,j] = sum_{b,o,p,q} dz2[b,o,p,q] * a1[b,o,4p+i,4q+j]
@@ -307,33 +366,68 @@ __global__ __launch_bounds__(256) void k_wgrad(
     for (int w = 0; w < S1_WSZ; ++w) acc[w] = 0.f;
     float bacc = 0.f;
     const int N = B * S1_OUT;
-    for (int it = slice * 256 + tid; it < N; it += GS * 256) {
-      const int b = it / S1_OUT;
-      const int opq = it - b * S1_OUT;
-      const int o = opq / S1_PIX;
-      const int pq = opq - o * S1_PIX;
-      const int pr = pq / S1_W;
-      const int pc = pq - pr * S1_W;
-      const float d = dz2g[it];
-      bacc += d;
-      const act_t* base = a1g + (size_t)b * C1_OUT + o * C1_PIX +
-                          pr * S1_K * C1_W + pc * S1_K;
-      float ar[4];
+    const int stride = GS * 256;
+    for (int it = slice * 256 + tid; it < N; it += 2 * stride) {
+      const int it2 = it + stride;
+      float da = 0.f, db = 0.f, ara[S1_K][S1_K], arb[S1_K][S1_K];
+      {
+        const int bi = it / S1_OUT;
+        const int opq = it - bi * S1_OUT;
+        const int oo = opq / S1_PIX;
+        const int pq = opq - oo * S1_PIX;
+        const int prr = pq / S1_W;
+        const int pcc = pq - prr * S1_W;
+        da = dz2g[it];
+        const act_t* base = a1g + (size_t)bi * C1_OUT + oo * C1_PIX +
+                            prr * S1_K * C1_W + pcc * S1_K;
 #pragma unroll
-      for (int i = 0; i < S1_K; ++i) {
-        ld4f(base + i * C1_W, ar);
+        for (int i = 0; i < S1_K; ++i) ld4f(base + i * C1_W, ara[i]);
+      }
+      if (it2 < N) {
+        const int bi = it2 / S1_OUT;
+        const int opq = it2 - bi * S1_OUT;
+        const int oo = opq / S1_PIX;
+        const int pq = opq - oo * S1_PIX;
+        const int prr = pq / S1_W;
+        const int pcc = pq - prr * S1_W;
+        db = dz2g[it2];
+        const act_t* base = a1g + (size_t)bi * C1_OUT + oo * C1_PIX +
+                            prr * S1_K * C1_W + pcc * S1_K;
 #pragma unroll
-        for (int j = 0; j < S1_K; ++j) acc[i * S1_K + j] += d * ar[j];
+        for (int i = 0; i < S1_K; ++i) ld4f(base + i * C1_W, arb[i]);
+      }
+      bacc += da;
+#pragma unroll
+      for (int i = 0; i < S1_K; ++i)
+#pragma unroll
+        for (int j = 0; j < S1_K; ++j) acc[i * S1_K + j] += da * ara[i][j];
+      if (it2 < N) {
+        bacc += db;
+#pragma unroll
+        for (int i = 0; i < S1_K; ++i)
+#pragma unroll
+          for (int j = 0; j < S1_K; ++j) acc[i * S1_K + j] += db * arb[i][j];
       }
     }
+    __shared__ float red[4][S1_WSZ + 1];
+    const int wv = tid >> 6;
 #pragma unroll
     for (int w = 0; w < S1_WSZ; ++w) {
-      const float s = wave_sum(acc[w]);
-      if (lane == 0) unsafeAtomicAdd(&grads[OFF_S1W + w], s);
+      const float v = wave_sum(acc[w]);
+      if (lane == 0) red[wv][w] = v;
     }
-    const float bs = wave_sum(bacc);
-    if (lane == 0)
-      unsafeAtomicAdd(&grads[OFF_S1B], bs / (float)S1_OUT);
+    {
+      const float v = wave_sum(bacc);
+      if (lane == 0) red[wv][S1_WSZ] = v;
+    }
+    __syncthreads();
+    if (tid < S1_WSZ) {
+      const float v = red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+      unsafeAtomicAdd(&grads[OFF_S1W + tid], v);
+    } else if (tid == S1_WSZ) {
+      const float v = red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+      unsafeAtomicAdd(&grads[OFF_S1B], v / (float)S1_OUT);
+    }
   } else {
     if (!(roles & 4)) return;
     // ---- fc: dW[k,m] = sum_b dz[b,k] * a2[b,m];  db[k] = sum_b dz[b,k]
@@ -444,7 +538,9 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   int GC = chunk_imgs > 0 ? chunk_imgs : (int)(2.0f * __builtin_cbrtf((float)B) + 0.5f);
   if (GC < 2) GC = 2;
   if (GC > 256) GC = 256;
-  int GS = GC / 4 > 2 ? GC / 4 : 2;
+  int GS = (B * S1_OUT + 256 * 48 - 1) / (256 * 48);
+  if (GS < 2) GS = 2;
+  if (GS > 96) GS = 96;
   int FS = B / 64;
   if (FS < 1) FS = 1;
   if (FS > 32) FS = 32;
