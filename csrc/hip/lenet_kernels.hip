@@ -76,6 +76,16 @@ __device__ __forceinline__ void ld8f(const T* p, float* out) {
   ld4f(p + 4, out + 4);
 }
 
+// Vectorized store of 4 consecutive activations (8-byte-aligned bf16 /
+// 16-byte-aligned fp32 destination).
+__device__ __forceinline__ void st4f(bf16* p, const float* v) {
+  bf16 t[4] = {(bf16)v[0], (bf16)v[1], (bf16)v[2], (bf16)v[3]};
+  *reinterpret_cast<uint2*>(p) = *reinterpret_cast<const uint2*>(t);
+}
+__device__ __forceinline__ void st4f(float* p, const float* v) {
+  *reinterpret_cast<float4*>(p) = make_float4(v[0], v[1], v[2], v[3]);
+}
+
 // Execution modes for the fused forward kernel.
 enum Mode { MODE_TRAIN = 0, MODE_EVAL = 1, MODE_INFER = 2 };
 
@@ -155,16 +165,11 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     }
     if (MODE == MODE_TRAIN) {
       // 4-wide activation stores per conv row
-      act_t packed[S1_K];
 #pragma unroll
-      for (int i = 0; i < S1_K; ++i) {
-#pragma unroll
-        for (int j = 0; j < S1_K; ++j) packed[j] = (act_t)a1v[i * S1_K + j];
-        *reinterpret_cast<uint2*>(
-            a1g + (size_t)b * C1_OUT + o * C1_PIX +
-            (pr * S1_K + i) * C1_W + pc * S1_K) =
-            *reinterpret_cast<const uint2*>(packed);
-      }
+      for (int i = 0; i < S1_K; ++i)
+        st4f(a1g + (size_t)b * C1_OUT + o * C1_PIX +
+                 (pr * S1_K + i) * C1_W + pc * S1_K,
+             &a1v[i * S1_K]);
     }
     a2v = sigmoidf_dev(pacc);
     L.a2s[tid] = a2v;
