@@ -540,7 +540,7 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
                          int roles, void* stream) {
   // Batch-adaptive defaults: ~36 (image,position) items per conv thread,
   // pool slices at GC/4, fc batch slices of ~64 images.
-  int GC = chunk_imgs > 0 ? chunk_imgs : (int)(2.0f * __builtin_cbrtf((float)B) + 0.5f);
+  int GC = chunk_imgs > 0 ? chunk_imgs : (int)(4.0f * __builtin_cbrtf((float)B) + 0.5f);
   if (GC < 2) GC = 2;
   if (GC > 256) GC = 256;
   int GS = (B * S1_OUT + 256 * 48 - 1) / (256 * 48);
