@@ -34,7 +34,7 @@ def main() -> int:
     p.add_argument("--batch-size", type=int, default=64,
                    help="per-GPU batch (BASELINE config: 64)")
     p.add_argument("--act-dtype", type=str, default="bf16",
-                   choices=["bf16", "fp32"])
+                   choices=["bf16", "fp16", "fp32"])
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--wgrad-chunk", type=int, default=0)
     args = p.parse_args()
@@ -84,11 +84,8 @@ def main() -> int:
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks
-    if ctx.world_size > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64,
-                         device=device if device == "cuda" else "cpu")
-        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-        elapsed = float(t.item())
+    elapsed = pdist.allreduce_max_scalar(
+        elapsed, device=device if device == "cuda" else None)
 
     global_batch = B * n_gpus
     images_per_sec = args.steps * global_batch / elapsed

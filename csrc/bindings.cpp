@@ -46,8 +46,9 @@ void check_hip(int err, const char* what) {
 
 int act_flag(const at::Tensor& t) {
   if (t.scalar_type() == at::kBFloat16) return 1;
+  if (t.scalar_type() == at::kHalf) return 2;
   TORCH_CHECK(t.scalar_type() == at::kFloat,
-              "activation tensors must be bf16 or fp32");
+              "activation tensors must be bf16, fp16 or fp32");
   return 0;
 }
 
@@ -123,7 +124,7 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
   TORCH_CHECK(pool_rows % B == 0, "pool rows must be a multiple of B");
   const int64_t P = pool_rows / B;
   const int f = act_flag(x_pool);
-  const size_t esz = f ? 2 : 4;
+  const size_t esz = f ? 2 : 4;  // bf16/fp16 vs fp32
   const char* xp = (const char*)x_pool.data_ptr();
   const int* lp = labels_pool.data_ptr<int>();
   float* pp = params.data_ptr<float>();

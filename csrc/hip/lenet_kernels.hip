@@ -32,6 +32,7 @@
 // Wave width is 64 (CDNA4); block size 256 = 4 waves.
 
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
 #include "../lenet_dims.h"
@@ -39,6 +40,7 @@
 namespace pcnn {
 
 using bf16 = __hip_bfloat16;
+using fp16 = __half;
 
 __device__ __forceinline__ float sigmoidf_dev(float v) {
   return 1.0f / (1.0f + __expf(-v));
@@ -63,6 +65,14 @@ __device__ __forceinline__ void ld4f(const bf16* p, float* out) {
   out[2] = (float)e[2];
   out[3] = (float)e[3];
 }
+__device__ __forceinline__ void ld4f(const fp16* p, float* out) {
+  const ushort4 v = *reinterpret_cast<const ushort4*>(p);
+  const fp16* e = reinterpret_cast<const fp16*>(&v);
+  out[0] = (float)e[0];
+  out[1] = (float)e[1];
+  out[2] = (float)e[2];
+  out[3] = (float)e[3];
+}
 __device__ __forceinline__ void ld4f(const float* p, float* out) {
   const float4 v = *reinterpret_cast<const float4*>(p);
   out[0] = v.x;
@@ -80,6 +90,10 @@ __device__ __forceinline__ void ld8f(const T* p, float* out) {
 // 16-byte-aligned fp32 destination).
 __device__ __forceinline__ void st4f(bf16* p, const float* v) {
   bf16 t[4] = {(bf16)v[0], (bf16)v[1], (bf16)v[2], (bf16)v[3]};
+  *reinterpret_cast<uint2*>(p) = *reinterpret_cast<const uint2*>(t);
+}
+__device__ __forceinline__ void st4f(fp16* p, const float* v) {
+  fp16 t[4] = {(fp16)v[0], (fp16)v[1], (fp16)v[2], (fp16)v[3]};
   *reinterpret_cast<uint2*>(p) = *reinterpret_cast<const uint2*>(t);
 }
 __device__ __forceinline__ void st4f(float* p, const float* v) {
@@ -497,9 +511,13 @@ int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, hipStream_t stream) {
   dim3 grid(B), block(256);
-  if (act_is_bf16) {
+  if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_fwdbwd<bf16, MODE>), grid, block, 0, stream,
                        (const bf16*)x, params, (bf16*)a1, (bf16*)a2, y, dz,
+                       dz2, dz1, labels, loss_accum, correct, B);
+  } else if (act_is_bf16 == 2) {
+    hipLaunchKernelGGL((k_fwdbwd<fp16, MODE>), grid, block, 0, stream,
+                       (const fp16*)x, params, (fp16*)a1, (fp16*)a2, y, dz,
                        dz2, dz1, labels, loss_accum, correct, B);
   } else {
     hipLaunchKernelGGL((k_fwdbwd<float, MODE>), grid, block, 0, stream,
@@ -551,9 +569,13 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   if (FS > 32) FS = 32;
   dim3 grid(C1_CH * GC + GS + FS * FC_BLOCKS), block(256);
   hipStream_t s = (hipStream_t)stream;
-  if (act_is_bf16) {
+  if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_wgrad<bf16>), grid, block, 0, s, (const bf16*)x,
                        (const bf16*)a1, (const bf16*)a2, dz, dz2, dz1, grads,
+                       B, GC, GS, FS, roles);
+  } else if (act_is_bf16 == 2) {
+    hipLaunchKernelGGL((k_wgrad<fp16>), grid, block, 0, s, (const fp16*)x,
+                       (const fp16*)a1, (const fp16*)a2, dz, dz2, dz1, grads,
                        B, GC, GS, FS, roles);
   } else {
     hipLaunchKernelGGL((k_wgrad<float>), grid, block, 0, s, (const float*)x,

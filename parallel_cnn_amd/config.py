@@ -17,7 +17,7 @@ from .ops.shapes import REF_DT, REF_THRESHOLD
 class TrainConfig:
     # model / numerics
     model: str = "lenet5"
-    act_dtype: str = "bf16"          # activation storage on GPU: bf16 | fp32
+    act_dtype: str = "bf16"          # activation storage on GPU: bf16 | fp16 | fp32
     seed: int = 0
 
     # optimizer (reference semantics: p += dt * grad)

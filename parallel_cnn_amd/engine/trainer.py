@@ -58,8 +58,9 @@ class Trainer:
         if self.backend in ("hip", "cpu"):
             self._C = native.require()
         self.model = model or LeNet5(self.device, seed=cfg.seed)
-        act_dtype = (torch.bfloat16
-                     if (cfg.act_dtype == "bf16" and self.backend == "hip")
+        act_map = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}
+        act_dtype = (act_map[cfg.act_dtype] if self.backend == "hip"
                      else torch.float32)
         self.act_dtype = act_dtype
         self.ws = Workspace(max_batch or cfg.batch_size, self.device,

@@ -37,16 +37,23 @@ def is_distributed() -> bool:
 
 def init_from_env(device: str = "auto") -> DistContext:
     """Initialise from torchrun env vars; no-op single-process context if
-    WORLD_SIZE is absent or 1."""
+    WORLD_SIZE is absent or 1.
+
+    Backend: RCCL ("nccl") when each rank has its own GPU; gloo otherwise.
+    PCNN_DIST_BACKEND overrides (e.g. gloo on a 1-GPU box to exercise the
+    multi-rank engine path with both ranks sharing the device)."""
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world <= 1:
         return DistContext()
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     use_gpu = device != "cpu" and torch.cuda.is_available()
-    backend = "nccl" if use_gpu else "gloo"
+    n_gpus = torch.cuda.device_count() if use_gpu else 0
+    backend = os.environ.get(
+        "PCNN_DIST_BACKEND",
+        "nccl" if (use_gpu and n_gpus >= world) else "gloo")
     if use_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(local_rank % n_gpus)
     if not dist.is_initialized():
         dist.init_process_group(
             backend=backend, timeout=datetime.timedelta(seconds=120))
@@ -56,17 +63,38 @@ def init_from_env(device: str = "auto") -> DistContext:
 
 def allreduce_grads(grads: torch.Tensor) -> None:
     """Sum the flat gradient bucket across ranks (the engine folds the
-    1/world factor into the update scale)."""
-    if is_distributed():
+    1/world factor into the update scale).  Over gloo with device tensors
+    (single-GPU multi-rank testing) the bucket bounces through host RAM —
+    correctness path only; real multi-GPU runs use RCCL directly."""
+    if not is_distributed():
+        return
+    if grads.is_cuda and dist.get_backend() == "gloo":
+        host = grads.cpu()
+        dist.all_reduce(host, op=dist.ReduceOp.SUM)
+        grads.copy_(host)
+    else:
         dist.all_reduce(grads, op=dist.ReduceOp.SUM)
 
 
 def allreduce_scalar(value: float, device=None) -> float:
     if not is_distributed():
         return value
+    if dist.get_backend() == "gloo":
+        device = "cpu"
     t = torch.tensor([value], dtype=torch.float64,
                      device=device if device is not None else "cpu")
     dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return float(t.item())
+
+
+def allreduce_max_scalar(value: float, device=None) -> float:
+    if not is_distributed():
+        return value
+    if dist.get_backend() == "gloo":
+        device = "cpu"
+    t = torch.tensor([value], dtype=torch.float64,
+                     device=device if device is not None else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return float(t.item())
 
 
