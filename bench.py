@@ -37,38 +37,53 @@ def main() -> int:
                    choices=["bf16", "fp16", "fp32"])
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--wgrad-chunk", type=int, default=0)
+    p.add_argument("--model", type=str, default="lenet5",
+                   choices=["lenet5", "deepcnn"])
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
-    from parallel_cnn_amd.data.mnist import synthetic_mnist
+    from parallel_cnn_amd.data.mnist import synthetic_images, synthetic_mnist
+    from parallel_cnn_amd.engine.deep import DeepTrainer
     from parallel_cnn_amd.engine.trainer import Trainer
     from parallel_cnn_amd.parallel import dist as pdist
 
     cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
                       device=args.device, log_interval=0, data="synthetic",
-                      wgrad_chunk=args.wgrad_chunk)
+                      wgrad_chunk=args.wgrad_chunk, model=args.model)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
     if ctx.world_size == 1 and args.gpus > 1:
         raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
 
-    trainer = Trainer(cfg, ctx=ctx)
+    trainer = (DeepTrainer(cfg, ctx=ctx) if args.model == "deepcnn"
+               else Trainer(cfg, ctx=ctx))
     B = args.batch_size
 
     # Device-resident synthetic epoch pool (no H2D inside the timed loop; the
     # pool is one epoch's worth of batches, cycled).
     n_pool_batches = max(1, min(args.steps + args.warmup,
                                 60000 // max(1, B)))
-    x_host, y_host = synthetic_mnist(n_pool_batches * B,
-                                     seed=1234 + ctx.rank, structured=False)
+    if args.model == "deepcnn":
+        x_host, y_host = synthetic_images(n_pool_batches * B, 32, 32, 3,
+                                          seed=1234 + ctx.rank,
+                                          structured=False)
+    else:
+        x_host, y_host = synthetic_mnist(n_pool_batches * B,
+                                         seed=1234 + ctx.rank,
+                                         structured=False)
     x_pool, y_pool = trainer.stage_batch(x_host, y_host)
     x_pool, y_pool = x_pool.contiguous(), y_pool.contiguous()
     if device == "cuda":
         torch.cuda.synchronize()
 
     def run(n_steps: int):
-        trainer.run_steps_pooled(x_pool, y_pool, n_steps)
+        if args.model == "deepcnn":
+            for st in range(n_steps):
+                i = (st % n_pool_batches) * B
+                trainer.step(x_pool[i:i + B], y_pool[i:i + B])
+        else:
+            trainer.run_steps_pooled(x_pool, y_pool, n_steps)
 
     run(args.warmup)
     pdist.barrier()
@@ -104,10 +119,13 @@ def main() -> int:
         "dtype": dtype,
         "data": "synthetic",
         "config": {
-            "model": "LeNet-5 28x28x1 (conv6x5x5 -> trainable-pool4x4 -> fc216x10)",
+            "model": ("DeepCNN 32x32x3 (3x(conv5x5+trainable-pool2x2) -> "
+                      "fc1024x10), im2col+MFMA GEMM path"
+                      if args.model == "deepcnn" else
+                      "LeNet-5 28x28x1 (conv6x5x5 -> trainable-pool4x4 -> fc216x10)"),
             "global_batch": global_batch,
             "per_gpu_batch": B,
-            "input": "28x28x1",
+            "input": "32x32x3" if args.model == "deepcnn" else "28x28x1",
             "parallelism": f"dp{n_gpus}",
             "backend": trainer.backend,
         },

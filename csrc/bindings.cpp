@@ -35,6 +35,42 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
                          int roles, void* stream);
 int pcnn_launch_update(float* params, float* grads, float step, void* stream);
 const char* pcnn_hip_error_string(int err);
+// deep (general conv) kernels — csrc/hip/conv_kernels.hip
+int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
+                     int K, int P, int KcP, int actf, void* stream);
+int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
+                   void* C, long long M, int K, int N, int ldA, int ldC,
+                   int b_kxn, int epilogue, int actf, void* stream);
+int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
+                         long long M, int KcP, int N, int MS, int actf,
+                         void* stream);
+int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
+                     int slices, int actf, void* stream);
+int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
+                            int B, int H, int W, int Cin, int K, int P,
+                            int KcP, int actf, void* stream);
+int pcnn_deep_pool_fwd(const void* a, const float* pw, void* pout, int B,
+                       int H, int W, int C, int K, int actf, void* stream);
+int pcnn_deep_pool_bwd(const void* dppre, const void* a, const float* pw,
+                       void* dapre, int B, int H, int W, int C, int K,
+                       int actf, void* stream);
+int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
+                         int H, int W, int C, int K, int G, int actf,
+                         void* stream);
+int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
+                     const int* labels, float* yg, float* dzg,
+                     float* loss_accum, int* correct_accum, int B, int FCIN,
+                     int NCLS, int mode, int actf, void* stream);
+int pcnn_deep_fc_bwd(const float* dzg, const void* flat, const float* fw,
+                     void* dflat, int B, int FCIN, int NCLS, int actf,
+                     void* stream);
+int pcnn_deep_fc_wgrad(const float* dzg, const void* flat, float* gfw,
+                       float* gfb, int B, int FCIN, int NCLS, int FS,
+                       int actf, void* stream);
+int pcnn_deep_update(float* params, float* grads, long long n, float step,
+                     void* stream);
+int pcnn_deep_mfma_selftest(const float* A, const float* Bm, float* D,
+                            void* stream);
 }
 
 namespace {
@@ -150,6 +186,139 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
   }
 }
 
+// ---------------------------------------------------------------------------
+// DeepCNN (general conv path) wrappers
+// ---------------------------------------------------------------------------
+void deep_im2col(at::Tensor x, at::Tensor cols, int64_t B, int64_t H,
+                 int64_t W, int64_t Cin, int64_t K, int64_t P, int64_t KcP,
+                 int64_t stream) {
+  check_hip(pcnn_deep_im2col(x.data_ptr(), cols.data_ptr(), (int)B, (int)H,
+                             (int)W, (int)Cin, (int)K, (int)P, (int)KcP,
+                             act_flag(x), (void*)stream),
+            "deep_im2col");
+}
+
+void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
+               int64_t M, int64_t K, int64_t N, int64_t ldA, int64_t ldC,
+               int64_t b_kxn, int64_t epilogue, int64_t stream) {
+  check_hip(pcnn_deep_gemm(A.data_ptr(), Bsrc.data_ptr<float>(),
+                           bias.numel() ? bias.data_ptr<float>() : nullptr,
+                           C.data_ptr(), M, (int)K, (int)N, (int)ldA,
+                           (int)ldC, (int)b_kxn, (int)epilogue, act_flag(A),
+                           (void*)stream),
+            "deep_gemm");
+}
+
+void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
+                     int64_t M, int64_t KcP, int64_t N, int64_t MS,
+                     int64_t stream) {
+  check_hip(pcnn_deep_wgrad_gemm(cols.data_ptr(), dpre.data_ptr(),
+                                 dW.data_ptr<float>(), M, (int)KcP, (int)N,
+                                 (int)MS, act_flag(cols), (void*)stream),
+            "deep_wgrad_gemm");
+}
+
+void deep_colsum(at::Tensor dpre, at::Tensor db, int64_t M, int64_t N,
+                 int64_t slices, int64_t stream) {
+  check_hip(pcnn_deep_colsum(dpre.data_ptr(), db.data_ptr<float>(), M,
+                             (int)N, (int)slices, act_flag(dpre),
+                             (void*)stream),
+            "deep_colsum");
+}
+
+void deep_col2im_sigbwd(at::Tensor dcols, at::Tensor pout, at::Tensor out,
+                        int64_t B, int64_t H, int64_t W, int64_t Cin,
+                        int64_t K, int64_t P, int64_t KcP, int64_t stream) {
+  check_hip(pcnn_deep_col2im_sigbwd(
+                dcols.data_ptr(), pout.numel() ? pout.data_ptr() : nullptr,
+                out.data_ptr(), (int)B, (int)H, (int)W, (int)Cin, (int)K,
+                (int)P, (int)KcP, act_flag(dcols), (void*)stream),
+            "deep_col2im_sigbwd");
+}
+
+void deep_pool_fwd(at::Tensor a, at::Tensor pw, at::Tensor pout, int64_t B,
+                   int64_t H, int64_t W, int64_t C, int64_t K,
+                   int64_t stream) {
+  check_hip(pcnn_deep_pool_fwd(a.data_ptr(), pw.data_ptr<float>(),
+                               pout.data_ptr(), (int)B, (int)H, (int)W,
+                               (int)C, (int)K, act_flag(a), (void*)stream),
+            "deep_pool_fwd");
+}
+
+void deep_pool_bwd(at::Tensor dppre, at::Tensor a, at::Tensor pw,
+                   at::Tensor dapre, int64_t B, int64_t H, int64_t W,
+                   int64_t C, int64_t K, int64_t stream) {
+  check_hip(pcnn_deep_pool_bwd(dppre.data_ptr(), a.data_ptr(),
+                               pw.data_ptr<float>(), dapre.data_ptr(),
+                               (int)B, (int)H, (int)W, (int)C, (int)K,
+                               act_flag(a), (void*)stream),
+            "deep_pool_bwd");
+}
+
+void deep_pool_wgrad(at::Tensor dppre, at::Tensor a, at::Tensor dpw,
+                     int64_t B, int64_t H, int64_t W, int64_t C, int64_t K,
+                     int64_t G, int64_t stream) {
+  check_hip(pcnn_deep_pool_wgrad(dppre.data_ptr(), a.data_ptr(),
+                                 dpw.data_ptr<float>(), (int)B, (int)H,
+                                 (int)W, (int)C, (int)K, (int)G,
+                                 act_flag(a), (void*)stream),
+            "deep_pool_wgrad");
+}
+
+void deep_fc_fwd(at::Tensor flat, at::Tensor fw, at::Tensor fb,
+                 at::Tensor labels, at::Tensor y, at::Tensor dz,
+                 at::Tensor loss_accum, at::Tensor correct_accum, int64_t B,
+                 int64_t FCIN, int64_t NCLS, int64_t mode, int64_t stream) {
+  check_hip(pcnn_deep_fc_fwd(
+                flat.data_ptr(), fw.data_ptr<float>(), fb.data_ptr<float>(),
+                labels.data_ptr<int>(),
+                y.numel() ? y.data_ptr<float>() : nullptr,
+                dz.numel() ? dz.data_ptr<float>() : nullptr,
+                loss_accum.numel() ? loss_accum.data_ptr<float>() : nullptr,
+                correct_accum.numel() ? correct_accum.data_ptr<int>()
+                                      : nullptr,
+                (int)B, (int)FCIN, (int)NCLS, (int)mode, act_flag(flat),
+                (void*)stream),
+            "deep_fc_fwd");
+}
+
+void deep_fc_bwd(at::Tensor dz, at::Tensor flat, at::Tensor fw,
+                 at::Tensor dflat, int64_t B, int64_t FCIN, int64_t NCLS,
+                 int64_t stream) {
+  check_hip(pcnn_deep_fc_bwd(dz.data_ptr<float>(), flat.data_ptr(),
+                             fw.data_ptr<float>(), dflat.data_ptr(), (int)B,
+                             (int)FCIN, (int)NCLS, act_flag(flat),
+                             (void*)stream),
+            "deep_fc_bwd");
+}
+
+void deep_fc_wgrad(at::Tensor dz, at::Tensor flat, at::Tensor gfw,
+                   at::Tensor gfb, int64_t B, int64_t FCIN, int64_t NCLS,
+                   int64_t FS, int64_t stream) {
+  check_hip(pcnn_deep_fc_wgrad(dz.data_ptr<float>(), flat.data_ptr(),
+                               gfw.data_ptr<float>(), gfb.data_ptr<float>(),
+                               (int)B, (int)FCIN, (int)NCLS, (int)FS,
+                               act_flag(flat), (void*)stream),
+            "deep_fc_wgrad");
+}
+
+void deep_update(at::Tensor params, at::Tensor grads, double step,
+                 int64_t stream) {
+  check_hip(pcnn_deep_update(params.data_ptr<float>(),
+                             grads.data_ptr<float>(), params.numel(),
+                             (float)step, (void*)stream),
+            "deep_update");
+}
+
+void deep_mfma_selftest(at::Tensor A, at::Tensor Bm, at::Tensor D,
+                        int64_t stream) {
+  check_hip(pcnn_deep_mfma_selftest(A.data_ptr<float>(),
+                                    Bm.data_ptr<float>(),
+                                    D.data_ptr<float>(), (void*)stream),
+            "deep_mfma_selftest");
+}
+
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -162,6 +331,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hip_wgrad_roles", &hip_wgrad_roles);
   m.def("hip_update", &hip_update);
   m.def("hip_train_steps", &hip_train_steps);
+  m.def("deep_im2col", &deep_im2col);
+  m.def("deep_gemm", &deep_gemm);
+  m.def("deep_wgrad_gemm", &deep_wgrad_gemm);
+  m.def("deep_colsum", &deep_colsum);
+  m.def("deep_col2im_sigbwd", &deep_col2im_sigbwd);
+  m.def("deep_pool_fwd", &deep_pool_fwd);
+  m.def("deep_pool_bwd", &deep_pool_bwd);
+  m.def("deep_pool_wgrad", &deep_pool_wgrad);
+  m.def("deep_fc_fwd", &deep_fc_fwd);
+  m.def("deep_fc_bwd", &deep_fc_bwd);
+  m.def("deep_fc_wgrad", &deep_fc_wgrad);
+  m.def("deep_update", &deep_update);
+  m.def("deep_mfma_selftest", &deep_mfma_selftest);
   m.attr("N_PARAMS") = pcnn::N_PARAMS;
   m.attr("OFF_C1W") = pcnn::OFF_C1W;
   m.attr("OFF_C1B") = pcnn::OFF_C1B;

@@ -1,0 +1,748 @@
+// General conv-net kernels for gfx950 (CDNA4 / MI355X): the im2col + MFMA
+// GEMM path (BASELINE.json config #4: 3x(conv5x5+pool)+FC on 32x32x3).
+//
+// Layout conventions (chosen MFMA-first):
+//   * Activations are NHWC:  a[B][H][W][C]  (act_t = bf16 / fp16 / fp32);
+//     the GEMM view of a conv output is C[M][N] with M = B*OH*OW rows in
+//     (b, oh, ow) order and N = Cout — i.e. exactly the NHWC tensor.
+//   * im2col columns:  cols[M][KcP] with kc = (i*K + j)*Cin + ci and KcP =
+//     Kc rounded up to 32 (zero-padded) so every GEMM K-loop is whole MFMA
+//     steps.  Consecutive kc == consecutive input channels == contiguous
+//     NHWC memory, so im2col reads and GEMM stages are coalesced.
+//   * Conv weights are stored [KcP][Cout] fp32 (master); pad rows are zero
+//     and stay zero (their gradients are identically zero).  Kernels cast
+//     fp32 -> bf16 while staging into LDS, so no separate cast pass exists.
+//
+// GEMM kernels use __builtin_amdgcn_mfma_f32_16x16x32_bf16 (gfx950 2xK
+// form), 64x64 C-tiles, BK=32, 4 waves x 4 fragments, single-buffered LDS
+// with +8 bf16 row padding (conflict-free ds_read_b128 fragment reads).
+// Correctness-first structure (the "step-0/1" shape of the CDNA GEMM
+// ladder); these GEMMs are microseconds at our sizes.
+//
+// Numerics: all accumulation fp32 (MFMA accumulators); activations and the
+// backward-data tensors stored act_t; parameters/gradients fp32.
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include "../lenet_dims.h"
+
+namespace pcnn_deep {
+
+using bf16 = __hip_bfloat16;
+using fp16 = __half;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ float sigmoidf_dev(float v) {
+  return 1.0f / (1.0f + __expf(-v));
+}
+
+template <typename T>
+__device__ __forceinline__ float ldf(const T* p) {
+  return (float)*p;
+}
+template <typename T>
+__device__ __forceinline__ void stf(T* p, float v) {
+  *p = (T)v;
+}
+
+// ---------------------------------------------------------------------------
+// im2col (NHWC, same-padding):
+//   cols[(b*OH+oh)*OW+ow][ (i*K+j)*Cin+ci ] = x[b][oh+i-P][ow+j-P][ci] or 0
+// One thread per cols element; kc is the fast axis (coalesced stores, and
+// coalesced loads since ci is the fast axis of NHWC x).
+// ---------------------------------------------------------------------------
+template <typename act_t>
+__global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
+                         int B, int H, int W, int Cin, int K, int P,
+                         int KcP) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)B * H * W * KcP;
+  if (idx >= total) return;
+  const int kc = (int)(idx % KcP);
+  const long long m = idx / KcP;
+  const int Kc = K * K * Cin;
+  act_t v = (act_t)0.f;
+  if (kc < Kc) {
+    const int ci = kc % Cin;
+    const int ij = kc / Cin;
+    const int i = ij / K;
+    const int j = ij - i * K;
+    const int ow = (int)(m % W);
+    const long long bh = m / W;
+    const int oh = (int)(bh % H);
+    const int b = (int)(bh / H);
+    const int ih = oh + i - P;
+    const int iw = ow + j - P;
+    if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+      v = x[(((long long)b * H + ih) * W + iw) * Cin + ci];
+  }
+  cols[m * KcP + kc] = v;
+}
+
+// ---------------------------------------------------------------------------
+// MFMA GEMM fragment maps for mfma_f32_16x16x32_bf16 (validated on-device
+// by k_mfma_selftest / tests):
+//   A[16x32]: lane l, reg r (0..7):  A[row = l&15][k = (l>>4)*8 + r]
+//   B[32x16]: lane l, reg r:         B[k = (l>>4)*8 + r][col = l&15]
+//   C[16x16]: lane l, reg r (0..3):  C[row = (l>>4)*4 + r][col = l&15]
+// ---------------------------------------------------------------------------
+
+constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int LDP = BK + 8;  // LDS row stride (bf16) — conflict-free b128
+
+struct GemmLds {
+  __bf16 As[BM][LDP];  // A tile, [m][k]
+  __bf16 Bs[BN][LDP];  // B tile, [n][k]  (transposed image: frag reads are
+                       // contiguous along k for both operands)
+};
+
+// Load an 8-element bf16 fragment from an LDS row.
+__device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
+  return *reinterpret_cast<const bf16x8*>(row + k0);
+}
+
+// Generic C[M][N] = epilogue(A[M][ldA] @ B [K][N] + bias):
+//   A: act_t, row-major, leading dim ldA (>= K, multiple of 32 used).
+//   Bsrc: fp32.  b_kxn = true: Bsrc is [K][N] (stage-transposed to LDS);
+//                 false: Bsrc is [N][K] (staged directly).
+//   epilogue: 0 = plain store, 1 = bias + sigmoid.
+//   C: act_t [M][ldC].
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_gemm(
+    const act_t* __restrict__ A, const float* __restrict__ Bsrc,
+    const float* __restrict__ bias, act_t* __restrict__ C, long long M,
+    int K, int N, int ldA, int ldC, int b_kxn, int epilogue) {
+  __shared__ GemmLds L;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int ntiles = (N + BN - 1) / BN;
+  const long long mtile = blockIdx.x / ntiles;
+  const int ntile = (int)(blockIdx.x % ntiles);
+  const long long m0 = mtile * BM;
+  const int n0 = ntile * BN;
+  const int nf = min(BN, N - n0) / 16;  // fragments along N (N % 16 == 0)
+
+  f32x4 acc[BN / 16];
+#pragma unroll
+  for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+
+  const int row_a = tid >> 2;            // 64 rows, 4 threads each
+  const int kq = (tid & 3) * 8;          // 8 k per thread
+  for (int kt = 0; kt < K; kt += BK) {
+    // stage A tile [BM][BK]
+    {
+      const long long m = m0 + row_a;
+      if (m < M) {
+        const act_t* src = A + m * ldA + kt + kq;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)(float)src[u];
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)0.f;
+      }
+    }
+    // stage B tile into [n][k] image
+    if (b_kxn) {
+      // Bsrc[K][N]: read rows k (coalesced along n), write transposed
+      const int k = tid >> 3;            // 32 k rows, 8 threads each
+      const int nq = (tid & 7) * 8;      // 8 n per thread
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int n = n0 + nq + u;
+        const float v = (n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
+        L.Bs[nq + u][k] = (__bf16)v;
+      }
+    } else {
+      // Bsrc[N][K]: row per n, staged directly
+      const int n = tid >> 2;            // 64 n rows, 4 threads each
+      const float* src = Bsrc + (long long)(n0 + n) * K + kt + kq;
+      const bool ok = (n0 + n) < N;
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        L.Bs[n][kq + u] = (__bf16)(ok ? src[u] : 0.f);
+    }
+    __syncthreads();
+    // wave wv owns C rows [wv*16, wv*16+16)
+    const bf16x8 a0 =
+        frag_from_lds(L.As[wv * 16 + (lane & 15)], (lane >> 4) * 8);
+#pragma unroll
+    for (int f = 0; f < BN / 16; ++f) {
+      if (f < nf) {
+        const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
+                                        (lane >> 4) * 8);
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0, 0,
+                                                         0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
+  const int crow = wv * 16 + (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int f = 0; f < BN / 16; ++f) {
+    if (f < nf) {
+      const int n = n0 + f * 16 + ccol;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long long m = m0 + crow + r;
+        if (m < M && n < N) {
+          float v = acc[f][r];
+          if (epilogue == 1) v = sigmoidf_dev(v + bias[n]);
+          stf(C + m * ldC + n, v);
+        }
+      }
+    }
+  }
+}
+
+// Weight-grad GEMM: dW[KcP][N] += cols^T[Kc x M-slice] @ dpre[M-slice x N].
+// Both operands are transpose-staged into the [row][k=m] LDS image; the
+// M dimension is the MFMA K axis.  Grid: (kc-tiles) x (n-tiles) x MS
+// M-slices; fp32 hardware atomics combine slices.
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_wgrad_gemm(
+    const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
+    float* __restrict__ dW, long long M, int KcP, int N, int MS) {
+  __shared__ GemmLds L;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int ntiles = (N + BN - 1) / BN;
+  const int ktiles = (KcP + BM - 1) / BM;
+  const int kct = blockIdx.x % max(1, ktiles);
+  const int rest = blockIdx.x / max(1, ktiles);
+  const int ntile = rest % ntiles;
+  const int slice = rest / ntiles;
+  const int kc0 = kct * BM;
+  const int n0 = ntile * BN;
+  const int nf = min(BN, N - n0) / 16;
+
+  const long long m_lo = (M * slice) / MS;
+  const long long m_hi = (M * (slice + 1)) / MS;
+
+  f32x4 acc[BN / 16];
+#pragma unroll
+  for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+
+  const int row_s = tid >> 3;        // source row m (32 rows, 8 thr each)
+  const int cq = (tid & 7) * 8;      // 8 columns per thread
+  for (long long mt = m_lo; mt < m_hi; mt += BK) {
+    // stage cols chunk [32m][64kc] -> LDS As[kc][m] (transposed)
+    {
+      const long long m = mt + row_s;
+      const act_t* src = cols + m * KcP + kc0 + cq;
+      const bool ok = m < m_hi && m < M;
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        L.As[cq + u][row_s] = (__bf16)(ok ? (float)src[u] : 0.f);
+    }
+    // stage dpre chunk [32m][<=64n] -> LDS Bs[n][m] (transposed)
+    {
+      const long long m = mt + row_s;
+      const act_t* src = dpre + m * N + n0 + cq;
+      const bool ok = m < m_hi && m < M;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const bool okn = ok && (n0 + cq + u) < N;
+        L.Bs[cq + u][row_s] = (__bf16)(okn ? (float)src[u] : 0.f);
+      }
+    }
+    __syncthreads();
+    const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
+                                    (lane >> 4) * 8);
+#pragma unroll
+    for (int f = 0; f < BN / 16; ++f) {
+      if (f < nf) {
+        const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
+                                        (lane >> 4) * 8);
+        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0, 0,
+                                                         0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int crow = wv * 16 + (lane >> 4) * 4;  // kc within tile
+  const int ccol = lane & 15;                  // n within fragment
+#pragma unroll
+  for (int f = 0; f < BN / 16; ++f) {
+    if (f < nf) {
+      const int n = n0 + f * 16 + ccol;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kc = kc0 + crow + r;
+        if (kc < KcP && n < N) {
+          const float v = acc[f][r];
+          if (v != 0.f) unsafeAtomicAdd(&dW[(long long)kc * N + n], v);
+        }
+      }
+    }
+  }
+}
+
+// Column-sum for the conv bias grad: db[n] += sum_m dpre[m][n].
+template <typename act_t>
+__global__ void k_colsum(const act_t* __restrict__ dpre,
+                         float* __restrict__ db, long long M, int N,
+                         int slices) {
+  const int n = blockIdx.x % ((N + 255) / 256) * 256 + threadIdx.x;
+  const int slice = blockIdx.x / ((N + 255) / 256);
+  if (n >= N) return;
+  const long long m_lo = (M * slice) / slices;
+  const long long m_hi = (M * (slice + 1)) / slices;
+  float acc = 0.f;
+  for (long long m = m_lo; m < m_hi; ++m) acc += ldf(dpre + m * N + n);
+  unsafeAtomicAdd(&db[n], acc);
+}
+
+// ---------------------------------------------------------------------------
+// col2im (gather form) fused with the pool-output sigmoid backward of the
+// PREVIOUS block:  for each input pixel of this conv,
+//   g = sum_{i,j valid} dcols[(b, h+P-i, w+P-j)][(i*K+j)*Cin+ci]
+//   out[b,h,w,ci] = g * pout*(1-pout)        (pout = that pixel's value)
+// When pout == nullptr the sigmoid factor is skipped (plain dX).
+// ---------------------------------------------------------------------------
+template <typename act_t>
+__global__ void k_col2im_sigbwd(const act_t* __restrict__ dcols,
+                                const act_t* __restrict__ pout,
+                                act_t* __restrict__ out, int B, int H, int W,
+                                int Cin, int K, int P, int KcP) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)B * H * W * Cin;
+  if (idx >= total) return;
+  const int ci = (int)(idx % Cin);
+  long long t = idx / Cin;
+  const int w = (int)(t % W);
+  t /= W;
+  const int h = (int)(t % H);
+  const int b = (int)(t / H);
+  float g = 0.f;
+  for (int i = 0; i < K; ++i) {
+    const int oh = h + P - i;
+    if (oh < 0 || oh >= H) continue;
+    for (int j = 0; j < K; ++j) {
+      const int ow = w + P - j;
+      if (ow < 0 || ow >= W) continue;
+      const long long m = ((long long)b * H + oh) * W + ow;
+      g += ldf(dcols + m * KcP + (i * K + j) * Cin + ci);
+    }
+  }
+  if (pout != nullptr) {
+    const float pv = ldf(pout + idx);
+    g *= pv * (1.0f - pv);
+  }
+  stf(out + idx, g);
+}
+
+// ---------------------------------------------------------------------------
+// Trainable pool (NHWC, shared KxK kernel, stride == K, scalar bias; the
+// framework's generalization of the reference's trainable 4x4 pool).
+// ---------------------------------------------------------------------------
+template <typename act_t>
+__global__ void k_pool_fwd(const act_t* __restrict__ a,
+                           const float* __restrict__ pw,  // [K*K] then bias
+                           act_t* __restrict__ pout, int B, int H, int W,
+                           int C, int K) {
+  const int OH = H / K, OW = W / K;
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)B * OH * OW * C;
+  if (idx >= total) return;
+  const int c = (int)(idx % C);
+  long long t = idx / C;
+  const int q = (int)(t % OW);
+  t /= OW;
+  const int p = (int)(t % OH);
+  const int b = (int)(t / OH);
+  float acc = pw[K * K];
+  for (int i = 0; i < K; ++i)
+    for (int j = 0; j < K; ++j)
+      acc += pw[i * K + j] *
+             ldf(a + (((long long)b * H + p * K + i) * W + q * K + j) * C + c);
+  stf(pout + idx, sigmoidf_dev(acc));
+}
+
+// dpre_conv[b,h,w,c] = dppre[b,h/K,w/K,c] * pw[h%K,w%K] * a*(1-a)
+template <typename act_t>
+__global__ void k_pool_bwd(const act_t* __restrict__ dppre,
+                           const act_t* __restrict__ a,
+                           const float* __restrict__ pw,
+                           act_t* __restrict__ dapre, int B, int H, int W,
+                           int C, int K) {
+  const int OH = H / K, OW = W / K;
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)B * H * W * C;
+  if (idx >= total) return;
+  const int c = (int)(idx % C);
+  long long t = idx / C;
+  const int w = (int)(t % W);
+  t /= W;
+  const int h = (int)(t % H);
+  const int b = (int)(t / H);
+  const float d =
+      ldf(dppre + (((long long)b * OH + h / K) * OW + w / K) * C + c);
+  const float av = ldf(a + idx);
+  stf(dapre + idx, d * pw[(h % K) * K + (w % K)] * av * (1.0f - av));
+}
+
+// pool wgrad: dpw[i,j] += sum dppre[b,p,q,c] * a[b,pK+i,qK+j,c];
+// bias += sum dppre.  Grid-stride, per-thread regs, wave reduce, atomics.
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_pool_wgrad(
+    const act_t* __restrict__ dppre, const act_t* __restrict__ a,
+    float* __restrict__ dpw, int B, int H, int W, int C, int K, int G) {
+  const int OH = H / K, OW = W / K;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  float acc[4 * 4];  // K <= 4 supported
+  for (int w = 0; w < K * K; ++w) acc[w] = 0.f;
+  float bacc = 0.f;
+  const long long N = (long long)B * OH * OW * C;
+  for (long long it = (long long)blockIdx.x * 256 + tid; it < N;
+       it += (long long)G * 256) {
+    const int c = (int)(it % C);
+    long long t = it / C;
+    const int q = (int)(t % OW);
+    t /= OW;
+    const int p = (int)(t % OH);
+    const int b = (int)(t / OH);
+    const float d = ldf(dppre + it);
+    bacc += d;
+    for (int i = 0; i < K; ++i)
+      for (int j = 0; j < K; ++j)
+        acc[i * K + j] +=
+            d * ldf(a + (((long long)b * H + p * K + i) * W + q * K + j) * C +
+                    c);
+  }
+  __shared__ float red[4][17];
+  const int wv = tid >> 6;
+  for (int w = 0; w < K * K; ++w) {
+    float v = acc[w];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][w] = v;
+  }
+  {
+    float v = bacc;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][16] = v;
+  }
+  __syncthreads();
+  if (tid < K * K)
+    unsafeAtomicAdd(&dpw[tid],
+                    red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid]);
+  if (tid == K * K)
+    unsafeAtomicAdd(&dpw[K * K],
+                    red[0][16] + red[1][16] + red[2][16] + red[3][16]);
+}
+
+// ---------------------------------------------------------------------------
+// FC head (FCIN -> 10) + residual loss, general fan-in.  One block per
+// sample; 16 lanes per class with shuffle reduction (the LeNet pattern).
+// mode: 0 train (emit dz + loss), 1 eval (argmax + correct), 2 infer.
+// ---------------------------------------------------------------------------
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_fc_fwd(
+    const act_t* __restrict__ flat, const float* __restrict__ fw,
+    const float* __restrict__ fb, const int* __restrict__ labels,
+    float* __restrict__ yg, float* __restrict__ dzg,
+    float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B,
+    int FCIN, int NCLS, int mode) {
+  __shared__ float ys[32];
+  __shared__ float sq[32];
+  const int b = blockIdx.x;
+  if (b >= B) return;
+  const int tid = threadIdx.x;
+  const act_t* xb = flat + (long long)b * FCIN;
+  if (tid < NCLS * 16) {
+    const int k = tid >> 4;
+    const int l = tid & 15;
+    const float* wk = fw + (long long)k * FCIN;
+    float p = 0.f;
+    for (int m = l; m < FCIN; m += 16) p += wk[m] * ldf(xb + m);
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) p += __shfl_down(p, off, 16);
+    if (l == 0) {
+      const float v = sigmoidf_dev(p + fb[k]);
+      ys[k] = v;
+      if (yg != nullptr) yg[(long long)b * NCLS + k] = v;
+      if (mode == 0) {
+        const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
+        sq[k] = d * d;
+        dzg[(long long)b * NCLS + k] = d;
+      }
+    }
+  }
+  __syncthreads();
+  if (mode == 1 && tid == 0) {
+    int best = 0;
+    for (int k = 1; k < NCLS; ++k)
+      if (ys[k] > ys[best]) best = k;
+    if (best == labels[b]) atomicAdd(correct_accum, 1);
+  }
+  if (mode == 0 && tid == 0 && loss_accum != nullptr) {
+    float s = 0.f;
+    for (int k = 0; k < NCLS; ++k) s += sq[k];
+    unsafeAtomicAdd(loss_accum, sqrtf(s));
+  }
+}
+
+// dflat[b,m] = (sum_k fw[k,m] dz[b,k]) * flat*(1-flat)   (act_t out)
+template <typename act_t>
+__global__ void k_fc_bwd(const float* __restrict__ dzg,
+                         const act_t* __restrict__ flat,
+                         const float* __restrict__ fw,
+                         act_t* __restrict__ dflat, int B, int FCIN,
+                         int NCLS) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long long)B * FCIN) return;
+  const int m = (int)(idx % FCIN);
+  const int b = (int)(idx / FCIN);
+  float da = 0.f;
+  for (int k = 0; k < NCLS; ++k)
+    da += fw[(long long)k * FCIN + m] * dzg[(long long)b * NCLS + k];
+  const float v = ldf(flat + idx);
+  stf(dflat + idx, da * v * (1.0f - v));
+}
+
+// fc wgrad: owner thread per (k, m) pair, batch-sliced like the LeNet fc.
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_fc_wgrad(
+    const float* __restrict__ dzg, const act_t* __restrict__ flat,
+    float* __restrict__ gfw, float* __restrict__ gfb, int B, int FCIN,
+    int NCLS, int FS) {
+  const int nw = NCLS * FCIN;
+  const int blocks_per_slice = (nw + 255 + NCLS) / 256 + 1;
+  const int slice = blockIdx.x / blocks_per_slice;
+  const int q = (blockIdx.x % blocks_per_slice) * 256 + threadIdx.x;
+  const int b_lo = (int)(((long long)B * slice) / FS);
+  const int b_hi = (int)(((long long)B * (slice + 1)) / FS);
+  if (q < nw) {
+    const int k = q / FCIN;
+    const int m = q - k * FCIN;
+    float acc = 0.f;
+    for (int b = b_lo; b < b_hi; ++b)
+      acc += dzg[(long long)b * NCLS + k] * ldf(flat + (long long)b * FCIN + m);
+    if (FS == 1)
+      gfw[q] += acc;
+    else
+      unsafeAtomicAdd(&gfw[q], acc);
+  } else if (q < nw + NCLS) {
+    const int k = q - nw;
+    float acc = 0.f;
+    for (int b = b_lo; b < b_hi; ++b) acc += dzg[(long long)b * NCLS + k];
+    if (FS == 1)
+      gfb[k] += acc;
+    else
+      unsafeAtomicAdd(&gfb[k], acc);
+  }
+}
+
+// Generic SGD apply + zero:  p += step*g; g = 0  over n params.
+__global__ void k_update_n(float* __restrict__ params,
+                           float* __restrict__ grads, long long n,
+                           float step) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    params[i] += step * grads[i];
+    grads[i] = 0.f;
+  }
+}
+
+// MFMA layout self-test: D[16][16] = A[16][32] @ B[32][16], plain fp32 I/O.
+__global__ void k_mfma_selftest(const float* __restrict__ A,
+                                const float* __restrict__ Bm,
+                                float* __restrict__ D) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+#pragma unroll
+  for (int r = 0; r < 8; ++r) {
+    a[r] = (__bf16)A[(lane & 15) * 32 + (lane >> 4) * 8 + r];
+    b[r] = (__bf16)Bm[((lane >> 4) * 8 + r) * 16 + (lane & 15)];
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = c[r];
+}
+
+}  // namespace pcnn_deep
+
+// ---------------------------------------------------------------------------
+// extern "C" launchers.  act flag: 0 = fp32, 1 = bf16, 2 = fp16.
+// ---------------------------------------------------------------------------
+using namespace pcnn_deep;
+
+#define PCNN_DISPATCH(flag, ...)                    \
+  do {                                               \
+    if ((flag) == 1) {                               \
+      using act_t = bf16;                            \
+      __VA_ARGS__;                                   \
+    } else if ((flag) == 2) {                        \
+      using act_t = fp16;                            \
+      __VA_ARGS__;                                   \
+    } else {                                         \
+      using act_t = float;                           \
+      __VA_ARGS__;                                   \
+    }                                                \
+  } while (0)
+
+extern "C" {
+
+int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
+                     int K, int P, int KcP, int actf, void* stream) {
+  const long long total = (long long)B * H * W * KcP;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL(
+                          (k_im2col<act_t>), grid, block, 0,
+                          (hipStream_t)stream, (const act_t*)x, (act_t*)cols,
+                          B, H, W, Cin, K, P, KcP));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
+                   void* C, long long M, int K, int N, int ldA, int ldC,
+                   int b_kxn, int epilogue, int actf, void* stream) {
+  const int ntiles = (N + BN - 1) / BN;
+  const long long mtiles = (M + BM - 1) / BM;
+  dim3 grid((unsigned)(mtiles * ntiles)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t>), grid, block, 0,
+                                    (hipStream_t)stream, (const act_t*)A,
+                                    Bsrc, bias, (act_t*)C, M, K, N, ldA, ldC,
+                                    b_kxn, epilogue));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
+                         long long M, int KcP, int N, int MS, int actf,
+                         void* stream) {
+  const int ntiles = (N + BN - 1) / BN;
+  const int ktiles = (KcP + BM - 1) / BM;
+  dim3 grid((unsigned)(ktiles * ntiles * MS)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_wgrad_gemm<act_t>), grid, block,
+                                          0, (hipStream_t)stream,
+                                          (const act_t*)cols,
+                                          (const act_t*)dpre, dW, M, KcP, N,
+                                          MS));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
+                     int slices, int actf, void* stream) {
+  dim3 grid((unsigned)(((N + 255) / 256) * slices)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_colsum<act_t>), grid, block, 0,
+                                          (hipStream_t)stream,
+                                          (const act_t*)dpre, db, M, N,
+                                          slices));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
+                            int B, int H, int W, int Cin, int K, int P,
+                            int KcP, int actf, void* stream) {
+  const long long total = (long long)B * H * W * Cin;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_col2im_sigbwd<act_t>), grid,
+                                          block, 0, (hipStream_t)stream,
+                                          (const act_t*)dcols,
+                                          (const act_t*)pout, (act_t*)out, B,
+                                          H, W, Cin, K, P, KcP));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_pool_fwd(const void* a, const float* pw, void* pout, int B,
+                       int H, int W, int C, int K, int actf, void* stream) {
+  const long long total = (long long)B * (H / K) * (W / K) * C;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_fwd<act_t>), grid, block, 0,
+                                          (hipStream_t)stream,
+                                          (const act_t*)a, pw, (act_t*)pout,
+                                          B, H, W, C, K));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_pool_bwd(const void* dppre, const void* a, const float* pw,
+                       void* dapre, int B, int H, int W, int C, int K,
+                       int actf, void* stream) {
+  const long long total = (long long)B * H * W * C;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_bwd<act_t>), grid, block, 0,
+                                          (hipStream_t)stream,
+                                          (const act_t*)dppre,
+                                          (const act_t*)a, pw, (act_t*)dapre,
+                                          B, H, W, C, K));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
+                         int H, int W, int C, int K, int G, int actf,
+                         void* stream) {
+  dim3 grid(G), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_wgrad<act_t>), grid, block,
+                                          0, (hipStream_t)stream,
+                                          (const act_t*)dppre,
+                                          (const act_t*)a, dpw, B, H, W, C, K,
+                                          G));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
+                     const int* labels, float* yg, float* dzg,
+                     float* loss_accum, int* correct_accum, int B, int FCIN,
+                     int NCLS, int mode, int actf, void* stream) {
+  dim3 grid(B), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_fc_fwd<act_t>), grid, block, 0,
+                                          (hipStream_t)stream,
+                                          (const act_t*)flat, fw, fb, labels,
+                                          yg, dzg, loss_accum, correct_accum,
+                                          B, FCIN, NCLS, mode));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_fc_bwd(const float* dzg, const void* flat, const float* fw,
+                     void* dflat, int B, int FCIN, int NCLS, int actf,
+                     void* stream) {
+  const long long total = (long long)B * FCIN;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_fc_bwd<act_t>), grid, block, 0,
+                                          (hipStream_t)stream, dzg,
+                                          (const act_t*)flat, fw,
+                                          (act_t*)dflat, B, FCIN, NCLS));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_fc_wgrad(const float* dzg, const void* flat, float* gfw,
+                       float* gfb, int B, int FCIN, int NCLS, int FS,
+                       int actf, void* stream) {
+  const int nw = NCLS * FCIN;
+  const int blocks_per_slice = (nw + 255 + NCLS) / 256 + 1;
+  dim3 grid(blocks_per_slice * FS), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_fc_wgrad<act_t>), grid, block, 0,
+                                          (hipStream_t)stream, dzg,
+                                          (const act_t*)flat, gfw, gfb, B,
+                                          FCIN, NCLS, FS));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_update(float* params, float* grads, long long n, float step,
+                     void* stream) {
+  dim3 grid((unsigned)((n + 255) / 256)), block(256);
+  hipLaunchKernelGGL(k_update_n, grid, block, 0, (hipStream_t)stream, params,
+                     grads, n, step);
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_mfma_selftest(const float* A, const float* Bm, float* D,
+                            void* stream) {
+  hipLaunchKernelGGL(k_mfma_selftest, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, A, Bm, D);
+  return (int)hipGetLastError();
+}
+}

@@ -52,6 +52,24 @@ def load_mnist(images_path: str, labels_path: str
     return x, y
 
 
+def synthetic_images(n: int, h: int, w: int, c: int, seed: int = 0,
+                     structured: bool = True, n_classes: int = 10
+                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """General synthetic image set: fp32 [n, h*w*c] (NHWC flat) in [0,1] +
+    labels.  structured=True adds a label-dependent bright band."""
+    g = torch.Generator().manual_seed(seed)
+    x = torch.rand(n, h, w, c, generator=g)
+    y = torch.randint(0, n_classes, (n,), generator=g)
+    if structured:
+        x *= 0.3
+        band = max(1, h // (n_classes + 2))
+        for lbl in range(n_classes):
+            rows = slice(1 + lbl * band, 1 + lbl * band + band)
+            x[y == lbl, rows, :, :] += 0.7
+        x.clamp_(0, 1)
+    return x.reshape(n, h * w * c), y
+
+
 def synthetic_mnist(n: int, seed: int = 0, structured: bool = True
                     ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Synthetic 28x28x1 images in [0,1] + labels in [0,10).

@@ -1,0 +1,224 @@
+"""Training/eval driver for the DeepCNN family (im2col + MFMA GEMM path).
+
+Backends: "hip" (gfx950 kernels, csrc/hip/conv_kernels.hip) and "torchref"
+(the fp32 oracle, also the CPU execution path for this family).
+
+Step sequence (hip): per stage {im2col, fused GEMM+bias+sigmoid, pool};
+fc fwd (+ residual loss); then backward {fc bwd/wgrad; per stage: pool
+wgrad, pool bwd (in-place over the conv activation), conv wgrad GEMM +
+colsum bias, dgrad GEMM (into the cols buffer) + col2im}, one fused DP
+all-reduce of the flat gradient bucket, SGD update.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ..config import TrainConfig
+from ..models.deepcnn import DeepCNN
+from ..ops import deep_ref, native
+from ..parallel import dist as pdist
+
+MODE_TRAIN, MODE_EVAL, MODE_INFER = 0, 1, 2
+
+
+class DeepWorkspace:
+    def __init__(self, model: DeepCNN, max_batch: int, device, act_dtype):
+        spec = model.spec
+        B = max_batch
+        self.max_batch = B
+        self.act_dtype = act_dtype
+        self.cols = []    # [M, KcP] per stage (reused as dcols in backward)
+        self.acts = []    # [B*H*W, Cout] per stage (conv act == NHWC)
+        self.pouts = []   # [B*OH*OW, Cout] per stage
+        self.dppre = []   # pool preact grads, same shape as pouts
+        for st in spec.stages:
+            M = B * st.h * st.w
+            self.cols.append(torch.empty(M, st.kcp, dtype=act_dtype,
+                                         device=device))
+            self.acts.append(torch.empty(M, st.cout, dtype=act_dtype,
+                                         device=device))
+            mo = B * st.oh * st.ow
+            self.pouts.append(torch.empty(mo, st.cout, dtype=act_dtype,
+                                          device=device))
+            self.dppre.append(torch.empty(mo, st.cout, dtype=act_dtype,
+                                          device=device))
+        self.y = torch.empty(B, spec.n_classes, dtype=torch.float32,
+                             device=device)
+        self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
+                              device=device)
+        self.loss_accum = torch.zeros(1, dtype=torch.float32, device=device)
+        self.correct_accum = torch.zeros(1, dtype=torch.int32, device=device)
+
+
+class DeepTrainer:
+    def __init__(self, cfg: TrainConfig, model: Optional[DeepCNN] = None,
+                 ctx: Optional[pdist.DistContext] = None,
+                 max_batch: Optional[int] = None):
+        self.cfg = cfg
+        self.ctx = ctx or pdist.DistContext()
+        self.device = torch.device(cfg.resolved_device())
+        backend = cfg.resolved_backend()
+        if backend == "cpu":
+            backend = "torchref"  # deep family's CPU path is the oracle
+        self.backend = backend
+        if backend == "hip":
+            self._C = native.require()
+        self.model = model or DeepCNN(self.device, seed=cfg.seed)
+        act_map = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}
+        self.act_dtype = (act_map[cfg.act_dtype] if backend == "hip"
+                          else torch.float32)
+        self.ws = DeepWorkspace(self.model, max_batch or cfg.batch_size,
+                                self.device, self.act_dtype)
+        self._loss_host = 0.0
+        self._samples_seen = 0
+        self.global_step = 0
+
+    # ------------------------------------------------------------------ util
+    def _scale(self, B: int) -> float:
+        if self.cfg.grad_reduction == "mean":
+            return 1.0 / float(B * self.ctx.world_size)
+        return 1.0
+
+    def stage_batch(self, x: torch.Tensor, labels: torch.Tensor):
+        """x: host fp32 [B, H*W*Cin] (NHWC flat)."""
+        if self.backend == "hip":
+            return (x.to(self.device, dtype=self.act_dtype,
+                         non_blocking=True),
+                    labels.to(self.device, dtype=torch.int32,
+                              non_blocking=True))
+        return x.to(torch.float32), labels.to(torch.int64)
+
+    # ------------------------------------------------------------- hip paths
+    def _hip_forward(self, x: torch.Tensor, labels: torch.Tensor, B: int,
+                     mode: int):
+        m, w, spec = self.model, self.ws, self.model.spec
+        st_h = native.current_stream_handle()
+        src = x
+        for i, st in enumerate(spec.stages):
+            self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin, st.k,
+                                st.pad, st.kcp, st_h)
+            M = B * st.h * st.w
+            self._C.deep_gemm(w.cols[i], m.view(f"conv{i}_w"),
+                              m.view(f"conv{i}_b"), w.acts[i], M, st.kcp,
+                              st.cout, st.kcp, st.cout, 1, 1, st_h)
+            self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
+                                  w.pouts[i], B, st.h, st.w, st.cout,
+                                  st.pool_k, st_h)
+            src = w.pouts[i]
+        self._C.deep_fc_fwd(w.pouts[-1], m.view("fc_w"), m.view("fc_b"),
+                            labels, w.y, w.dz, w.loss_accum,
+                            w.correct_accum, B, spec.fc_in, spec.n_classes,
+                            mode, st_h)
+
+    def _hip_backward(self, x: torch.Tensor, B: int):
+        m, w, spec = self.model, self.ws, self.model.spec
+        st_h = native.current_stream_handle()
+        nstage = len(spec.stages)
+        last = spec.stages[-1]
+        self._C.deep_fc_bwd(w.dz, w.pouts[-1], m.view("fc_w"),
+                            w.dppre[-1], B, spec.fc_in, spec.n_classes, st_h)
+        fs = max(1, min(32, B // 64))
+        self._C.deep_fc_wgrad(w.dz, w.pouts[-1], m.grad_view("fc_w"),
+                              m.grad_view("fc_b"), B, spec.fc_in,
+                              spec.n_classes, fs, st_h)
+        for i in range(nstage - 1, -1, -1):
+            st = spec.stages[i]
+            M = B * st.h * st.w
+            G = max(8, min(256, (B * st.oh * st.ow * st.cout) // (256 * 32)))
+            self._C.deep_pool_wgrad(w.dppre[i], w.acts[i],
+                                    m.grad_view(f"pool{i}_w"), B, st.h, st.w,
+                                    st.cout, st.pool_k, G, st_h)
+            # pool bwd writes the conv preact grad IN PLACE over the conv
+            # activation (elementwise same-index, safe)
+            self._C.deep_pool_bwd(w.dppre[i], w.acts[i],
+                                  m.view(f"pool{i}_w"), w.acts[i], B, st.h,
+                                  st.w, st.cout, st.pool_k, st_h)
+            dapre = w.acts[i]
+            ktiles = (st.kcp + 63) // 64
+            ntiles = (st.cout + 63) // 64
+            ms = max(1, min(64, 256 // (ktiles * ntiles)))
+            self._C.deep_wgrad_gemm(w.cols[i], dapre,
+                                    m.grad_view(f"conv{i}_w"), M, st.kcp,
+                                    st.cout, ms, st_h)
+            self._C.deep_colsum(dapre, m.grad_view(f"conv{i}_b"), M,
+                                st.cout, 64, st_h)
+            if i > 0:
+                # dgrad into the cols buffer (its forward use is done)
+                self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
+                                  torch.empty(0), w.cols[i], M, st.cout,
+                                  st.kcp, st.cout, st.kcp, 0, 0, st_h)
+                prev = spec.stages[i - 1]
+                self._C.deep_col2im_sigbwd(w.cols[i], w.pouts[i - 1],
+                                           w.dppre[i - 1], B, st.h, st.w,
+                                           st.cin, st.k, st.pad, st.kcp,
+                                           st_h)
+
+    # ------------------------------------------------------------------ step
+    def step(self, x: torch.Tensor, labels: torch.Tensor) -> None:
+        B = x.shape[0]
+        assert B <= self.ws.max_batch
+        scale = self._scale(B)
+        if self.backend == "hip":
+            self._hip_forward(x, labels, B, MODE_TRAIN)
+            self._hip_backward(x, B)
+            pdist.allreduce_grads(self.model.grads)
+            self._C.deep_update(self.model.params, self.model.grads,
+                                self.cfg.dt * scale,
+                                native.current_stream_handle())
+        else:
+            spec = self.model.spec
+            xh = x.view(B, spec.in_h, spec.in_w, spec.in_ch)
+            acts, pouts, y = deep_ref.forward(xh, self.model)
+            grads, loss = deep_ref.backward(xh, self.model, acts, pouts, y,
+                                            labels)
+            self._loss_host += loss
+            self.model.grads += grads
+            pdist.allreduce_grads(self.model.grads)
+            with torch.no_grad():
+                self.model.params += self.cfg.dt * scale * self.model.grads
+                self.model.grads.zero_()
+        self._samples_seen += B * self.ctx.world_size
+        self.global_step += 1
+
+    def consume_loss(self) -> Tuple[float, int]:
+        if self.backend == "hip":
+            local = float(self.ws.loss_accum.item())
+            self.ws.loss_accum.zero_()
+        else:
+            local = self._loss_host
+            self._loss_host = 0.0
+        dev = self.device if self.device.type == "cuda" else None
+        total = pdist.allreduce_scalar(local, device=dev)
+        n = self._samples_seen
+        self._samples_seen = 0
+        return total, n
+
+    @torch.no_grad()
+    def evaluate(self, x: torch.Tensor, labels: torch.Tensor,
+                 batch_size: Optional[int] = None) -> float:
+        bs = batch_size or self.ws.max_batch
+        n = x.shape[0]
+        correct = 0
+        if self.backend == "hip":
+            self.ws.correct_accum.zero_()
+        per = (n + self.ctx.world_size - 1) // self.ctx.world_size
+        lo, hi = self.ctx.rank * per, min(n, (self.ctx.rank + 1) * per)
+        for i in range(lo, hi, bs):
+            xb, lb = self.stage_batch(x[i:i + bs], labels[i:i + bs])
+            B = xb.shape[0]
+            if self.backend == "hip":
+                self._hip_forward(xb, lb, B, MODE_EVAL)
+            else:
+                spec = self.model.spec
+                _, _, y = deep_ref.forward(
+                    xb.view(B, spec.in_h, spec.in_w, spec.in_ch), self.model)
+                correct += int((y.argmax(1) == lb).sum().item())
+        if self.backend == "hip":
+            correct = int(self.ws.correct_accum.item())
+            self.ws.correct_accum.zero_()
+        dev = self.device if self.device.type == "cuda" else None
+        correct = int(pdist.allreduce_scalar(float(correct), device=dev))
+        return 100.0 * (1.0 - correct / float(n))

@@ -20,6 +20,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 
 HIP_SOURCES = [
     os.path.join(HERE, "csrc", "hip", "lenet_kernels.hip"),
+    os.path.join(HERE, "csrc", "hip", "conv_kernels.hip"),
 ]
 
 
