@@ -233,14 +233,19 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
   const int row_s = tid >> 3;        // source row m (32 rows, 8 thr each)
   const int cq = (tid & 7) * 8;      // 8 columns per thread
   for (long long mt = m_lo; mt < m_hi; mt += BK) {
-    // stage cols chunk [32m][64kc] -> LDS As[kc][m] (transposed)
+    // stage cols chunk [32m][64kc] -> LDS As[kc][m] (transposed).
+    // Guard kc against KcP: the last kc-tile of a 96-wide cols buffer
+    // would otherwise read past the row (and past the allocation on the
+    // final row).
     {
       const long long m = mt + row_s;
       const act_t* src = cols + m * KcP + kc0 + cq;
       const bool ok = m < m_hi && m < M;
 #pragma unroll
-      for (int u = 0; u < 8; ++u)
-        L.As[cq + u][row_s] = (__bf16)(ok ? (float)src[u] : 0.f);
+      for (int u = 0; u < 8; ++u) {
+        const bool okc = ok && (kc0 + cq + u) < KcP;
+        L.As[cq + u][row_s] = (__bf16)(okc ? (float)src[u] : 0.f);
+      }
     }
     // stage dpre chunk [32m][<=64n] -> LDS Bs[n][m] (transposed)
     {
