@@ -48,6 +48,36 @@ __device__ __forceinline__ void stf(T* p, float v) {
   *p = (T)v;
 }
 
+// 8 consecutive elements -> fp32 (16-byte-aligned bf16/fp16, 32B fp32).
+__device__ __forceinline__ void ld8v(const bf16* p, float* out) {
+  const ushort4 a = *reinterpret_cast<const ushort4*>(p);
+  const ushort4 b = *reinterpret_cast<const ushort4*>(p + 4);
+  const bf16* ea = reinterpret_cast<const bf16*>(&a);
+  const bf16* eb = reinterpret_cast<const bf16*>(&b);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    out[u] = (float)ea[u];
+    out[u + 4] = (float)eb[u];
+  }
+}
+__device__ __forceinline__ void ld8v(const fp16* p, float* out) {
+  const ushort4 a = *reinterpret_cast<const ushort4*>(p);
+  const ushort4 b = *reinterpret_cast<const ushort4*>(p + 4);
+  const fp16* ea = reinterpret_cast<const fp16*>(&a);
+  const fp16* eb = reinterpret_cast<const fp16*>(&b);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    out[u] = (float)ea[u];
+    out[u + 4] = (float)eb[u];
+  }
+}
+__device__ __forceinline__ void ld8v(const float* p, float* out) {
+  const float4 a = *reinterpret_cast<const float4*>(p);
+  const float4 b = *reinterpret_cast<const float4*>(p + 4);
+  out[0] = a.x; out[1] = a.y; out[2] = a.z; out[3] = a.w;
+  out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
+}
+
 // ---------------------------------------------------------------------------
 // im2col (NHWC, same-padding):
 //   cols[(b*OH+oh)*OW+ow][ (i*K+j)*Cin+ci ] = x[b][oh+i-P][ow+j-P][ci] or 0
@@ -58,28 +88,41 @@ template <typename act_t>
 __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
                          int B, int H, int W, int Cin, int K, int P,
                          int KcP) {
+  // 8 consecutive kc per thread (KcP % 8 == 0): one 8-element vector store
+  // per thread; x reads are L1/L2 hits (each input element is re-read by up
+  // to K*K column positions).
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long long total = (long long)B * H * W * KcP;
+  const long long total = ((long long)B * H * W * KcP) / 8;
   if (idx >= total) return;
-  const int kc = (int)(idx % KcP);
-  const long long m = idx / KcP;
+  const int kc0 = (int)((idx * 8) % KcP);
+  const long long m = (idx * 8) / KcP;
   const int Kc = K * K * Cin;
-  act_t v = (act_t)0.f;
-  if (kc < Kc) {
-    const int ci = kc % Cin;
-    const int ij = kc / Cin;
-    const int i = ij / K;
-    const int j = ij - i * K;
-    const int ow = (int)(m % W);
-    const long long bh = m / W;
-    const int oh = (int)(bh % H);
-    const int b = (int)(bh / H);
-    const int ih = oh + i - P;
-    const int iw = ow + j - P;
-    if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-      v = x[(((long long)b * H + ih) * W + iw) * Cin + ci];
+  const int ow = (int)(m % W);
+  const long long bh = m / W;
+  const int oh = (int)(bh % H);
+  const int b = (int)(bh / H);
+  float v[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int kc = kc0 + u;
+    float val = 0.f;
+    if (kc < Kc) {
+      const int ci = kc % Cin;
+      const int ij = kc / Cin;
+      const int i = ij / K;
+      const int j = ij - i * K;
+      const int ih = oh + i - P;
+      const int iw = ow + j - P;
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+        val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
+    }
+    v[u] = val;
   }
-  cols[m * KcP + kc] = v;
+  act_t out[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) out[u] = (act_t)v[u];
+  *reinterpret_cast<uint4*>(cols + m * KcP + kc0) =
+      *reinterpret_cast<const uint4*>(out);
 }
 
 // ---------------------------------------------------------------------------
@@ -137,9 +180,10 @@ __global__ __launch_bounds__(256) void k_gemm(
     {
       const long long m = m0 + row_a;
       if (m < M) {
-        const act_t* src = A + m * ldA + kt + kq;
+        float v8[8];
+        ld8v(A + m * ldA + kt + kq, v8);
 #pragma unroll
-        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)(float)src[u];
+        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)v8[u];
       } else {
 #pragma unroll
         for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)0.f;
@@ -239,24 +283,36 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     // final row).
     {
       const long long m = mt + row_s;
-      const act_t* src = cols + m * KcP + kc0 + cq;
       const bool ok = m < m_hi && m < M;
+      const bool okc_all = (kc0 + cq + 8) <= KcP;
+      float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ok && okc_all) {
+        ld8v(cols + m * KcP + kc0 + cq, v8);
+      } else if (ok) {
+        const act_t* src = cols + m * KcP + kc0 + cq;
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const bool okc = ok && (kc0 + cq + u) < KcP;
-        L.As[cq + u][row_s] = (__bf16)(okc ? (float)src[u] : 0.f);
+        for (int u = 0; u < 8; ++u)
+          if ((kc0 + cq + u) < KcP) v8[u] = (float)src[u];
       }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) L.As[cq + u][row_s] = (__bf16)v8[u];
     }
     // stage dpre chunk [32m][<=64n] -> LDS Bs[n][m] (transposed)
     {
       const long long m = mt + row_s;
-      const act_t* src = dpre + m * N + n0 + cq;
       const bool ok = m < m_hi && m < M;
+      const bool okn_all = (n0 + cq + 8) <= N;
+      float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (ok && okn_all) {
+        ld8v(dpre + m * N + n0 + cq, v8);
+      } else if (ok) {
+        const act_t* src = dpre + m * N + n0 + cq;
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const bool okn = ok && (n0 + cq + u) < N;
-        L.Bs[cq + u][row_s] = (__bf16)(okn ? (float)src[u] : 0.f);
+        for (int u = 0; u < 8; ++u)
+          if ((n0 + cq + u) < N) v8[u] = (float)src[u];
       }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) L.Bs[cq + u][row_s] = (__bf16)v8[u];
     }
     __syncthreads();
     const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
@@ -292,18 +348,28 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 }
 
 // Column-sum for the conv bias grad: db[n] += sum_m dpre[m][n].
+// 256 threads cover 256/N consecutive rows per iteration (perfectly
+// coalesced); cross-thread tree-reduce per column, then one atomic.
 template <typename act_t>
-__global__ void k_colsum(const act_t* __restrict__ dpre,
-                         float* __restrict__ db, long long M, int N,
-                         int slices) {
-  const int n = blockIdx.x % ((N + 255) / 256) * 256 + threadIdx.x;
-  const int slice = blockIdx.x / ((N + 255) / 256);
-  if (n >= N) return;
-  const long long m_lo = (M * slice) / slices;
-  const long long m_hi = (M * (slice + 1)) / slices;
+__global__ __launch_bounds__(256) void k_colsum(const act_t* __restrict__ dpre,
+                                                float* __restrict__ db,
+                                                long long M, int N, int G) {
+  const int tid = threadIdx.x;
+  const int rows_per_iter = 256 / N;  // N in {32, 64, 128, 256}
+  const int n = tid % N;
+  const int rsub = tid / N;
   float acc = 0.f;
-  for (long long m = m_lo; m < m_hi; ++m) acc += ldf(dpre + m * N + n);
-  unsafeAtomicAdd(&db[n], acc);
+  for (long long m = (long long)blockIdx.x * rows_per_iter + rsub; m < M;
+       m += (long long)G * rows_per_iter)
+    acc += ldf(dpre + m * N + n);
+  __shared__ float s[256];
+  s[tid] = acc;
+  __syncthreads();
+  for (int off = 128; off >= N; off >>= 1) {
+    if (tid < off) s[tid] += s[tid + off];
+    __syncthreads();
+  }
+  if (tid < N) unsafeAtomicAdd(&db[tid], s[tid]);
 }
 
 // ---------------------------------------------------------------------------
@@ -641,7 +707,7 @@ int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
 
 int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
                      int slices, int actf, void* stream) {
-  dim3 grid((unsigned)(((N + 255) / 256) * slices)), block(256);
+  dim3 grid((unsigned)slices), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_colsum<act_t>), grid, block, 0,
                                           (hipStream_t)stream,
                                           (const act_t*)dpre, db, M, N,

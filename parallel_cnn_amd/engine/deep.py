@@ -127,7 +127,7 @@ class DeepTrainer:
         for i in range(nstage - 1, -1, -1):
             st = spec.stages[i]
             M = B * st.h * st.w
-            G = max(8, min(256, (B * st.oh * st.ow * st.cout) // (256 * 32)))
+            G = max(8, min(512, (B * st.oh * st.ow * st.cout) // (256 * 16)))
             self._C.deep_pool_wgrad(w.dppre[i], w.acts[i],
                                     m.grad_view(f"pool{i}_w"), B, st.h, st.w,
                                     st.cout, st.pool_k, G, st_h)
@@ -139,12 +139,13 @@ class DeepTrainer:
             dapre = w.acts[i]
             ktiles = (st.kcp + 63) // 64
             ntiles = (st.cout + 63) // 64
-            ms = max(1, min(64, 256 // (ktiles * ntiles)))
+            ms = max(1, min(64, 512 // (ktiles * ntiles)))
             self._C.deep_wgrad_gemm(w.cols[i], dapre,
                                     m.grad_view(f"conv{i}_w"), M, st.kcp,
                                     st.cout, ms, st_h)
+            gsum = max(32, min(512, (M * st.cout) // (256 * 96)))
             self._C.deep_colsum(dapre, m.grad_view(f"conv{i}_b"), M,
-                                st.cout, 64, st_h)
+                                st.cout, gsum, st_h)
             if i > 0:
                 # dgrad into the cols buffer (its forward use is done)
                 self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
