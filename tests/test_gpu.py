@@ -254,6 +254,33 @@ def test_native_extension_is_loaded_on_gpu(device):
     assert t.backend == "hip"
 
 
+def test_native_cli_trainer(device, tmp_path):
+    """tools/pcnn_train: the no-Python native driver must train, report in
+    the reference's stdout shape, and write a checkpoint the Python side
+    can load (cross-implementation checkpoint compatibility)."""
+    import os
+    bin_path = os.path.join(os.path.dirname(__file__), "..", "tools",
+                            "pcnn_train")
+    if not os.path.exists(bin_path):
+        pytest.skip("native CLI not built")
+    ck = str(tmp_path / "cli.bin")
+    out = subprocess.run(
+        [bin_path, "--epochs", "2", "--train-count", "4096", "--test-count",
+         "1024", "--batch-size", "64", "--ckpt-save", ck],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "Learning" in out.stdout
+    assert "error: " in out.stdout
+    assert "Error Rate: " in out.stdout
+    errs = [float(l.split("error: ")[1].split(",")[0])
+            for l in out.stdout.splitlines() if l.startswith("error: ")]
+    assert errs[-1] < errs[0], errs  # learning on the structured bands
+    from parallel_cnn_amd.models.lenet import LeNet5
+    m = LeNet5()
+    m.load(ck)  # right size, loadable
+    assert torch.isfinite(m.params).all()
+
+
 def test_bench_contract_single_gpu(device):
     out = subprocess.run(
         [sys.executable, "bench.py", "--gpus", "1", "--steps", "30",
