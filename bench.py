@@ -39,6 +39,8 @@ def main() -> int:
     p.add_argument("--wgrad-chunk", type=int, default=0)
     p.add_argument("--model", type=str, default="lenet5",
                    choices=["lenet5", "deepcnn"])
+    p.add_argument("--use-graph", action="store_true",
+                   help="capture the step in a hipGraph and replay it")
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
@@ -58,6 +60,8 @@ def main() -> int:
 
     trainer = (DeepTrainer(cfg, ctx=ctx) if args.model == "deepcnn"
                else Trainer(cfg, ctx=ctx))
+    if args.use_graph:
+        trainer.enable_graph()
     B = args.batch_size
 
     # Device-resident synthetic epoch pool (no H2D inside the timed loop; the
@@ -128,6 +132,7 @@ def main() -> int:
             "input": "32x32x3" if args.model == "deepcnn" else "28x28x1",
             "parallelism": f"dp{n_gpus}",
             "backend": trainer.backend,
+            "hipgraph": bool(getattr(trainer, "_graph", None)),
         },
     }
     if ctx.is_main:

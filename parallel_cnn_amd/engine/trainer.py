@@ -125,6 +125,53 @@ class Trainer:
         self._samples_seen += B * self.ctx.world_size
         self.global_step += 1
 
+    # ------------------------------------------------------------ hipGraph
+    def enable_graph(self) -> None:
+        """Capture one full training step (3 kernels + the RCCL all-reduce
+        when distributed) into a hipGraph; run_steps_pooled then replays it.
+        Requires world_size==1 or the nccl/RCCL backend (gloo is host-side,
+        not capturable)."""
+        if self.backend != "hip":
+            raise RuntimeError("graph capture requires the hip backend")
+        if self.ctx.world_size > 1 and \
+                torch.distributed.get_backend() != "nccl":
+            raise RuntimeError("graph capture requires RCCL (nccl backend)")
+        B = self.ws.max_batch
+        self._gx = torch.zeros(B, S.IN_PIX, dtype=self.act_dtype,
+                               device=self.device)
+        self._gl = torch.zeros(B, dtype=torch.int32, device=self.device)
+        # warmup on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._graph_body(B)
+        torch.cuda.current_stream().wait_stream(s)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._graph_body(B)
+
+    def _graph_body(self, B: int) -> None:
+        m, w = self.model, self.ws
+        stream = native.current_stream_handle()
+        self._C.hip_fwdbwd(self._gx, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
+                           w.dz1, self._gl, w.loss_accum, w.correct_accum,
+                           B, MODE_TRAIN, stream)
+        self._C.hip_wgrad(self._gx, w.a1, w.a2, w.dz, w.dz2, w.dz1, m.grads,
+                          B, self.cfg.wgrad_chunk, stream)
+        pdist.allreduce_grads(m.grads)
+        self._C.hip_update(m.params, m.grads,
+                           self.cfg.dt * self._update_scale(B),
+                           native.current_stream_handle())
+
+    def step_graph(self, x: torch.Tensor, labels: torch.Tensor) -> None:
+        """Replay the captured step on a staged batch (one D2D copy in)."""
+        self._gx.copy_(x, non_blocking=True)
+        self._gl.copy_(labels, non_blocking=True)
+        self._graph.replay()
+        self._samples_seen += x.shape[0] * self.ctx.world_size
+        self.global_step += 1
+
     def run_steps_pooled(self, x_pool: torch.Tensor,
                          labels_pool: torch.Tensor, steps: int) -> None:
         """Run `steps` training steps over a device-resident batch pool
@@ -133,6 +180,11 @@ class Trainer:
         runs keep the per-step loop with the gradient all-reduce."""
         B = self.ws.max_batch
         P = x_pool.shape[0] // B
+        if getattr(self, "_graph", None) is not None:
+            for s in range(steps):
+                i = (s % P) * B
+                self.step_graph(x_pool[i:i + B], labels_pool[i:i + B])
+            return
         if self.backend == "hip" and self.ctx.world_size == 1:
             w = self.ws
             self._C.hip_train_steps(

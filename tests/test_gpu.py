@@ -254,6 +254,26 @@ def test_native_extension_is_loaded_on_gpu(device):
     assert t.backend == "hip"
 
 
+def test_graph_step_matches_eager(device):
+    """hipGraph-captured step replay == eager step trajectory."""
+    x, y = synthetic_mnist(64, seed=19)
+    cfg = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0)
+    t1, t2 = Trainer(cfg), Trainer(cfg)
+    t1.enable_graph()
+    xp1, yp1 = t1.stage_batch(x, y)
+    xp2, yp2 = t2.stage_batch(x, y)
+    t1.run_steps_pooled(xp1.contiguous(), yp1.contiguous(), 4)  # graph
+    t2.run_steps_pooled(xp2.contiguous(), yp2.contiguous(), 4)  # C++ loop
+    torch.cuda.synchronize()
+    diff = (t1.model.params - t2.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+    l1, n1 = t1.consume_loss()
+    l2, n2 = t2.consume_loss()
+    assert n1 == n2 == 64
+    assert abs(l1 - l2) < 1e-3 * max(1.0, l2)
+
+
 def test_native_cli_trainer(device, tmp_path):
     """tools/pcnn_train: the no-Python native driver must train, report in
     the reference's stdout shape, and write a checkpoint the Python side
