@@ -140,7 +140,10 @@ class Trainer:
         self._gx = torch.zeros(B, S.IN_PIX, dtype=self.act_dtype,
                                device=self.device)
         self._gl = torch.zeros(B, dtype=torch.int32, device=self.device)
-        # warmup on a side stream, then capture
+        # warmup on a side stream, then capture; the warmup replays REAL
+        # steps, so snapshot and restore the training state around it
+        params0 = self.model.params.clone()
+        grads0 = self.model.grads.clone()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -150,6 +153,11 @@ class Trainer:
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph):
             self._graph_body(B)
+        with torch.no_grad():
+            self.model.params.copy_(params0)
+            self.model.grads.copy_(grads0)
+            self.ws.loss_accum.zero_()
+        torch.cuda.synchronize()
 
     def _graph_body(self, B: int) -> None:
         m, w = self.model, self.ws
