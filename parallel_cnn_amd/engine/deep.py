@@ -197,6 +197,29 @@ class DeepTrainer:
         self._samples_seen = 0
         return total, n
 
+    def train_epoch(self, x: torch.Tensor, labels: torch.Tensor,
+                    log=print) -> float:
+        """One epoch over a host dataset, DP-sharded; returns the mean
+        per-sample error norm."""
+        Bl = self.ws.max_batch
+        Bg = Bl * self.ctx.world_size
+        n = (x.shape[0] // Bg) * Bg
+        total_loss, total_n = 0.0, 0
+        for s in range(0, n, Bg):
+            lo = s + self.ctx.rank * Bl
+            self.step(*self.stage_batch(x[lo:lo + Bl], labels[lo:lo + Bl]))
+            if self.cfg.log_interval and \
+                    self.global_step % self.cfg.log_interval == 0:
+                loss, cnt = self.consume_loss()
+                total_loss += loss
+                total_n += cnt
+                if self.ctx.is_main and cnt:
+                    log(f"step {self.global_step}: error {loss / cnt:e}")
+        loss, cnt = self.consume_loss()
+        total_loss += loss
+        total_n += cnt
+        return total_loss / max(1, total_n)
+
     @torch.no_grad()
     def evaluate(self, x: torch.Tensor, labels: torch.Tensor,
                  batch_size: Optional[int] = None) -> float:

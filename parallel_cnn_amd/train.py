@@ -15,7 +15,8 @@ import time
 import torch
 
 from .config import TrainConfig
-from .data.mnist import load_mnist, synthetic_mnist
+from .data.mnist import load_mnist, synthetic_images, synthetic_mnist
+from .engine.deep import DeepTrainer
 from .engine.trainer import Trainer
 from .parallel import dist as pdist
 
@@ -28,6 +29,11 @@ def load_datasets(cfg: TrainConfig):
         xte, yte = load_mnist(
             os.path.join(cfg.data_dir, "t10k-images.idx3-ubyte"),
             os.path.join(cfg.data_dir, "t10k-labels.idx1-ubyte"))
+    elif cfg.model == "deepcnn":
+        xtr, ytr = synthetic_images(cfg.train_count, 32, 32, 3,
+                                    seed=cfg.seed)
+        xte, yte = synthetic_images(cfg.test_count, 32, 32, 3,
+                                    seed=cfg.seed + 1)
     else:
         xtr, ytr = synthetic_mnist(cfg.train_count, seed=cfg.seed)
         xte, yte = synthetic_mnist(cfg.test_count, seed=cfg.seed + 1)
@@ -41,7 +47,8 @@ def main(argv=None) -> int:
     cfg = TrainConfig.from_args(args)
 
     ctx = pdist.init_from_env(cfg.resolved_device())
-    trainer = Trainer(cfg, ctx=ctx)
+    trainer = (DeepTrainer(cfg, ctx=ctx) if cfg.model == "deepcnn"
+               else Trainer(cfg, ctx=ctx))
     if cfg.ckpt_load:
         trainer.model.load(cfg.ckpt_load)
 
