@@ -415,6 +415,7 @@ __global__ void k_col2im_sigbwd(const act_t* __restrict__ dcols,
 // Trainable pool (NHWC, shared KxK kernel, stride == K, scalar bias; the
 // framework's generalization of the reference's trainable 4x4 pool).
 // ---------------------------------------------------------------------------
+// 8 consecutive channels per thread (C % 8 == 0): all loads/stores 16B.
 template <typename act_t>
 __global__ void k_pool_fwd(const act_t* __restrict__ a,
                            const float* __restrict__ pw,  // [K*K] then bias
@@ -422,20 +423,34 @@ __global__ void k_pool_fwd(const act_t* __restrict__ a,
                            int C, int K) {
   const int OH = H / K, OW = W / K;
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long long total = (long long)B * OH * OW * C;
+  const long long total = (long long)B * OH * OW * C / 8;
   if (idx >= total) return;
-  const int c = (int)(idx % C);
-  long long t = idx / C;
+  const int c0 = (int)((idx * 8) % C);
+  long long t = (idx * 8) / C;
   const int q = (int)(t % OW);
   t /= OW;
   const int p = (int)(t % OH);
   const int b = (int)(t / OH);
-  float acc = pw[K * K];
+  float acc[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) acc[u] = pw[K * K];
   for (int i = 0; i < K; ++i)
-    for (int j = 0; j < K; ++j)
-      acc += pw[i * K + j] *
-             ldf(a + (((long long)b * H + p * K + i) * W + q * K + j) * C + c);
-  stf(pout + idx, sigmoidf_dev(acc));
+    for (int j = 0; j < K; ++j) {
+      float av[8];
+      ld8v(a + (((long long)b * H + p * K + i) * W + q * K + j) * C + c0, av);
+      const float wv = pw[i * K + j];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) acc[u] += wv * av[u];
+    }
+  act_t out[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) out[u] = (act_t)sigmoidf_dev(acc[u]);
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(pout + idx * 8) =
+        *reinterpret_cast<const uint4*>(out);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) pout[idx * 8 + u] = out[u];
 }
 
 // dpre_conv[b,h,w,c] = dppre[b,h/K,w/K,c] * pw[h%K,w%K] * a*(1-a)
@@ -447,18 +462,28 @@ __global__ void k_pool_bwd(const act_t* __restrict__ dppre,
                            int C, int K) {
   const int OH = H / K, OW = W / K;
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long long total = (long long)B * H * W * C;
+  const long long total = (long long)B * H * W * C / 8;
   if (idx >= total) return;
-  const int c = (int)(idx % C);
-  long long t = idx / C;
+  const int c0 = (int)((idx * 8) % C);
+  long long t = (idx * 8) / C;
   const int w = (int)(t % W);
   t /= W;
   const int h = (int)(t % H);
   const int b = (int)(t / H);
-  const float d =
-      ldf(dppre + (((long long)b * OH + h / K) * OW + w / K) * C + c);
-  const float av = ldf(a + idx);
-  stf(dapre + idx, d * pw[(h % K) * K + (w % K)] * av * (1.0f - av));
+  float dv[8], av[8];
+  ld8v(dppre + (((long long)b * OH + h / K) * OW + w / K) * C + c0, dv);
+  ld8v(a + idx * 8, av);
+  const float wv = pw[(h % K) * K + (w % K)];
+  act_t out[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u)
+    out[u] = (act_t)(dv[u] * wv * av[u] * (1.0f - av[u]));
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(dapre + idx * 8) =
+        *reinterpret_cast<const uint4*>(out);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) dapre[idx * 8 + u] = out[u];
 }
 
 // pool wgrad: dpw[i,j] += sum dppre[b,p,q,c] * a[b,pK+i,qK+j,c];
@@ -669,7 +694,7 @@ extern "C" {
 
 int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
                      int K, int P, int KcP, int actf, void* stream) {
-  const long long total = (long long)B * H * W * KcP;
+  const long long total = (long long)B * H * W * KcP / 8;  // 8 elems/thread
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL(
                           (k_im2col<act_t>), grid, block, 0,
@@ -730,7 +755,7 @@ int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
 
 int pcnn_deep_pool_fwd(const void* a, const float* pw, void* pout, int B,
                        int H, int W, int C, int K, int actf, void* stream) {
-  const long long total = (long long)B * (H / K) * (W / K) * C;
+  const long long total = (long long)B * (H / K) * (W / K) * C / 8;
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_fwd<act_t>), grid, block, 0,
                                           (hipStream_t)stream,
@@ -742,7 +767,7 @@ int pcnn_deep_pool_fwd(const void* a, const float* pw, void* pout, int B,
 int pcnn_deep_pool_bwd(const void* dppre, const void* a, const float* pw,
                        void* dapre, int B, int H, int W, int C, int K,
                        int actf, void* stream) {
-  const long long total = (long long)B * H * W * C;
+  const long long total = (long long)B * H * W * C / 8;
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_bwd<act_t>), grid, block, 0,
                                           (hipStream_t)stream,

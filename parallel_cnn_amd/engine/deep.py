@@ -139,7 +139,7 @@ class DeepTrainer:
             dapre = w.acts[i]
             ktiles = (st.kcp + 63) // 64
             ntiles = (st.cout + 63) // 64
-            ms = max(1, min(64, 512 // (ktiles * ntiles)))
+            ms = max(1, min(128, 512 // (ktiles * ntiles)))
             self._C.deep_wgrad_gemm(w.cols[i], dapre,
                                     m.grad_view(f"conv{i}_w"), M, st.kcp,
                                     st.cout, ms, st_h)
