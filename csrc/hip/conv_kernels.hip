@@ -88,9 +88,11 @@ template <typename act_t>
 __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
                          int B, int H, int W, int Cin, int K, int P,
                          int KcP) {
-  // 8 consecutive kc per thread (KcP % 8 == 0): one 8-element vector store
-  // per thread; x reads are L1/L2 hits (each input element is re-read by up
-  // to K*K column positions).
+  // 8 consecutive kc per thread (KcP % 8 == 0): one 16B vector store.
+  // Key layout fact: within one kernel row i, consecutive kc = (i*K+j)*Cin
+  // + ci map to CONSECUTIVE NHWC x addresses (base + (iw0*Cin + ci0) + t),
+  // so when the 8-span stays inside row i and inside the image it is ONE
+  // 16B vector load; otherwise fall back to per-element gather.
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long total = ((long long)B * H * W * KcP) / 8;
   if (idx >= total) return;
@@ -101,28 +103,49 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
   const long long bh = m / W;
   const int oh = (int)(bh % H);
   const int b = (int)(bh / H);
-  float v[8];
-#pragma unroll
-  for (int u = 0; u < 8; ++u) {
-    const int kc = kc0 + u;
-    float val = 0.f;
-    if (kc < Kc) {
-      const int ci = kc % Cin;
-      const int ij = kc / Cin;
-      const int i = ij / K;
-      const int j = ij - i * K;
-      const int ih = oh + i - P;
-      const int iw = ow + j - P;
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-        val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
-    }
-    v[u] = val;
-  }
+  const int rowc = K * Cin;  // elements per kernel row i
+
   act_t out[8];
+  const int ci0 = kc0 % Cin;
+  const int ij0 = kc0 / Cin;
+  const int i0 = ij0 / K;
+  const int j0 = ij0 - i0 * K;
+  const int ih0 = oh + i0 - P;
+  const int iw0 = ow + j0 - P;
+  // last element's column position
+  const int j7 = ((kc0 + 7) / Cin) - i0 * K;
+  // the vector load needs 8-byte source alignment: holds iff Cin % 8 == 0
+  const bool same_row = (kc0 % rowc) + 8 <= rowc;
+  if ((Cin % 8) == 0 && same_row && kc0 + 8 <= Kc && ih0 >= 0 && ih0 < H &&
+      iw0 >= 0 && (ow + j7 - P) < W) {
+    float v8[8];
+    ld8v(x + (((long long)b * H + ih0) * W + iw0) * Cin + ci0, v8);
 #pragma unroll
-  for (int u = 0; u < 8; ++u) out[u] = (act_t)v[u];
-  *reinterpret_cast<uint4*>(cols + m * KcP + kc0) =
-      *reinterpret_cast<const uint4*>(out);
+    for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
+  } else {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int kc = kc0 + u;
+      float val = 0.f;
+      if (kc < Kc) {
+        const int ci = kc % Cin;
+        const int ij = kc / Cin;
+        const int i = ij / K;
+        const int j = ij - i * K;
+        const int ih = oh + i - P;
+        const int iw = ow + j - P;
+        if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+          val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
+      }
+      out[u] = (act_t)val;
+    }
+  }
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(cols + m * KcP + kc0) =
+        *reinterpret_cast<const uint4*>(out);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) cols[m * KcP + kc0 + u] = out[u];
 }
 
 // ---------------------------------------------------------------------------
@@ -133,7 +156,8 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
 //   C[16x16]: lane l, reg r (0..3):  C[row = (l>>4)*4 + r][col = l&15]
 // ---------------------------------------------------------------------------
 
-constexpr int BM = 64, BN = 64, BK = 32;
+constexpr int BM = 64, BN = 64, BK = 64;  // 64-deep K-step: 2 MFMAs per
+                                          // fragment per barrier pair
 constexpr int LDP = BK + 8;  // LDS row stride (bf16) — conflict-free b128
 
 struct GemmLds {
@@ -174,52 +198,62 @@ __global__ __launch_bounds__(256) void k_gemm(
   for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
 
   const int row_a = tid >> 2;            // 64 rows, 4 threads each
-  const int kq = (tid & 3) * 8;          // 8 k per thread
+  const int kq = (tid & 3) * 16;         // 16 k per thread (two 8-chunks)
   for (int kt = 0; kt < K; kt += BK) {
-    // stage A tile [BM][BK]
+    // stage A tile [BM][BK]; K % 32 == 0, so each 8-chunk is fully in or
+    // fully out of range (zeros otherwise)
     {
       const long long m = m0 + row_a;
-      if (m < M) {
-        float v8[8];
-        ld8v(A + m * ldA + kt + kq, v8);
 #pragma unroll
-        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)v8[u];
-      } else {
+      for (int h = 0; h < 2; ++h) {
+        const int kk = kq + h * 8;
+        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (m < M && (kt + kk) < K) ld8v(A + m * ldA + kt + kk, v8);
 #pragma unroll
-        for (int u = 0; u < 8; ++u) L.As[row_a][kq + u] = (__bf16)0.f;
+        for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)v8[u];
       }
     }
     // stage B tile into [n][k] image
     if (b_kxn) {
       // Bsrc[K][N]: read rows k (coalesced along n), write transposed
-      const int k = tid >> 3;            // 32 k rows, 8 threads each
-      const int nq = (tid & 7) * 8;      // 8 n per thread
+      const int k = tid >> 2;            // 64 k rows, 4 threads each
+      const int nq = (tid & 3) * 16;     // 16 n per thread
+      const bool kok = (kt + k) < K;
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
+      for (int u = 0; u < 16; ++u) {
         const int n = n0 + nq + u;
-        const float v = (n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
+        const float v =
+            (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
         L.Bs[nq + u][k] = (__bf16)v;
       }
     } else {
       // Bsrc[N][K]: row per n, staged directly
       const int n = tid >> 2;            // 64 n rows, 4 threads each
-      const float* src = Bsrc + (long long)(n0 + n) * K + kt + kq;
       const bool ok = (n0 + n) < N;
 #pragma unroll
-      for (int u = 0; u < 8; ++u)
-        L.Bs[n][kq + u] = (__bf16)(ok ? src[u] : 0.f);
+      for (int h = 0; h < 2; ++h) {
+        const int kk = kq + h * 8;
+        const float* src = Bsrc + (long long)(n0 + n) * K + kt + kk;
+        const bool kok = ok && (kt + kk) < K;
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          L.Bs[n][kk + u] = (__bf16)(kok ? src[u] : 0.f);
+      }
     }
     __syncthreads();
-    // wave wv owns C rows [wv*16, wv*16+16)
-    const bf16x8 a0 =
-        frag_from_lds(L.As[wv * 16 + (lane & 15)], (lane >> 4) * 8);
+    // wave wv owns C rows [wv*16, wv*16+16); two 32-deep MFMA sub-steps
 #pragma unroll
-    for (int f = 0; f < BN / 16; ++f) {
-      if (f < nf) {
-        const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
-                                        (lane >> 4) * 8);
-        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0, 0,
-                                                         0);
+    for (int kk = 0; kk < 2; ++kk) {
+      const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
+                                      kk * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int f = 0; f < BN / 16; ++f) {
+        if (f < nf) {
+          const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
+                                          kk * 32 + (lane >> 4) * 8);
+          acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0,
+                                                           0, 0);
+        }
       }
     }
     __syncthreads();
@@ -274,8 +308,8 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 #pragma unroll
   for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
 
-  const int row_s = tid >> 3;        // source row m (32 rows, 8 thr each)
-  const int cq = (tid & 7) * 8;      // 8 columns per thread
+  const int row_s = tid >> 2;        // source row m (64 rows, 4 thr each)
+  const int cq = (tid & 3) * 16;     // 16 columns per thread
   for (long long mt = m_lo; mt < m_hi; mt += BK) {
     // stage cols chunk [32m][64kc] -> LDS As[kc][m] (transposed).
     // Guard kc against KcP: the last kc-tile of a 96-wide cols buffer
@@ -284,46 +318,55 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     {
       const long long m = mt + row_s;
       const bool ok = m < m_hi && m < M;
-      const bool okc_all = (kc0 + cq + 8) <= KcP;
-      float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (ok && okc_all) {
-        ld8v(cols + m * KcP + kc0 + cq, v8);
-      } else if (ok) {
-        const act_t* src = cols + m * KcP + kc0 + cq;
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          if ((kc0 + cq + u) < KcP) v8[u] = (float)src[u];
+      for (int h = 0; h < 2; ++h) {
+        const int c = cq + h * 8;
+        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (ok && (kc0 + c + 8) <= KcP) {
+          ld8v(cols + m * KcP + kc0 + c, v8);
+        } else if (ok) {
+          const act_t* src = cols + m * KcP + kc0 + c;
+#pragma unroll
+          for (int u = 0; u < 8; ++u)
+            if ((kc0 + c + u) < KcP) v8[u] = (float)src[u];
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) L.As[c + u][row_s] = (__bf16)v8[u];
       }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) L.As[cq + u][row_s] = (__bf16)v8[u];
     }
     // stage dpre chunk [32m][<=64n] -> LDS Bs[n][m] (transposed)
     {
       const long long m = mt + row_s;
       const bool ok = m < m_hi && m < M;
-      const bool okn_all = (n0 + cq + 8) <= N;
-      float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (ok && okn_all) {
-        ld8v(dpre + m * N + n0 + cq, v8);
-      } else if (ok) {
-        const act_t* src = dpre + m * N + n0 + cq;
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          if ((n0 + cq + u) < N) v8[u] = (float)src[u];
+      for (int h = 0; h < 2; ++h) {
+        const int c = cq + h * 8;
+        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (ok && (n0 + c + 8) <= N) {
+          ld8v(dpre + m * N + n0 + c, v8);
+        } else if (ok) {
+          const act_t* src = dpre + m * N + n0 + c;
+#pragma unroll
+          for (int u = 0; u < 8; ++u)
+            if ((n0 + c + u) < N) v8[u] = (float)src[u];
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) L.Bs[c + u][row_s] = (__bf16)v8[u];
       }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) L.Bs[cq + u][row_s] = (__bf16)v8[u];
     }
     __syncthreads();
-    const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
-                                    (lane >> 4) * 8);
 #pragma unroll
-    for (int f = 0; f < BN / 16; ++f) {
-      if (f < nf) {
-        const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
-                                        (lane >> 4) * 8);
-        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0, 0,
-                                                         0);
+    for (int kk = 0; kk < 2; ++kk) {
+      const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
+                                      kk * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int f = 0; f < BN / 16; ++f) {
+        if (f < nf) {
+          const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
+                                          kk * 32 + (lane >> 4) * 8);
+          acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0,
+                                                           0, 0);
+        }
       }
     }
     __syncthreads();
@@ -384,16 +427,18 @@ __global__ void k_col2im_sigbwd(const act_t* __restrict__ dcols,
                                 const act_t* __restrict__ pout,
                                 act_t* __restrict__ out, int B, int H, int W,
                                 int Cin, int K, int P, int KcP) {
+  // 8 consecutive ci per thread (Cin % 8 == 0 for every caller): the K*K
+  // gather loads become 8-wide vector loads.
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long long total = (long long)B * H * W * Cin;
+  const long long total = (long long)B * H * W * Cin / 8;
   if (idx >= total) return;
-  const int ci = (int)(idx % Cin);
-  long long t = idx / Cin;
+  const int ci0 = (int)((idx * 8) % Cin);
+  long long t = (idx * 8) / Cin;
   const int w = (int)(t % W);
   t /= W;
   const int h = (int)(t % H);
   const int b = (int)(t / H);
-  float g = 0.f;
+  float g[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   for (int i = 0; i < K; ++i) {
     const int oh = h + P - i;
     if (oh < 0 || oh >= H) continue;
@@ -401,14 +446,29 @@ __global__ void k_col2im_sigbwd(const act_t* __restrict__ dcols,
       const int ow = w + P - j;
       if (ow < 0 || ow >= W) continue;
       const long long m = ((long long)b * H + oh) * W + ow;
-      g += ldf(dcols + m * KcP + (i * K + j) * Cin + ci);
+      float v8[8];
+      ld8v(dcols + m * KcP + (i * K + j) * Cin + ci0, v8);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) g[u] += v8[u];
     }
   }
+  act_t o8[8];
   if (pout != nullptr) {
-    const float pv = ldf(pout + idx);
-    g *= pv * (1.0f - pv);
+    float p8[8];
+    ld8v(pout + idx * 8, p8);
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      o8[u] = (act_t)(g[u] * p8[u] * (1.0f - p8[u]));
+  } else {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) o8[u] = (act_t)g[u];
   }
-  stf(out + idx, g);
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(out + idx * 8) =
+        *reinterpret_cast<const uint4*>(o8);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) out[idx * 8 + u] = o8[u];
 }
 
 // ---------------------------------------------------------------------------
@@ -743,7 +803,7 @@ int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
 int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
                             int B, int H, int W, int Cin, int K, int P,
                             int KcP, int actf, void* stream) {
-  const long long total = (long long)B * H * W * Cin;
+  const long long total = (long long)B * H * W * Cin / 8;
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_col2im_sigbwd<act_t>), grid,
                                           block, 0, (hipStream_t)stream,
