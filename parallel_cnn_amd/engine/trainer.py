@@ -68,6 +68,10 @@ class Trainer:
         self._loss_host = 0.0       # cpu/torchref backends accumulate here
         self._samples_seen = 0
         self.global_step = 0
+        # cached stream handle for the hot per-step path (the graph-capture
+        # body resolves the live stream instead)
+        self._sh = native.current_stream_handle() if self.backend == "hip" \
+            else 0
 
     # ------------------------------------------------------------------ util
     def _update_scale(self, local_batch: int) -> float:
@@ -94,7 +98,7 @@ class Trainer:
         m, w = self.model, self.ws
         scale = self._update_scale(B)
         if self.backend == "hip":
-            stream = native.current_stream_handle()
+            stream = self._sh
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream)
@@ -102,7 +106,7 @@ class Trainer:
                               self.cfg.wgrad_chunk, stream)
             pdist.allreduce_grads(m.grads)
             self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
-                               native.current_stream_handle())
+                               stream)
         elif self.backend == "cpu":
             # CPU path keeps fp32 activations in the workspace directly.
             a1 = self._cpu_view(w.a1, B)
