@@ -12,11 +12,11 @@
 
 namespace pcnn {
 void cpu_forward(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
-                 at::Tensor y);
+                 at::Tensor y, int64_t pool_mode, int64_t loss_mode);
 double cpu_backward(at::Tensor x, at::Tensor params, at::Tensor a1,
                     at::Tensor a2, at::Tensor y, at::Tensor labels,
                     at::Tensor dz, at::Tensor dz2, at::Tensor dz1,
-                    at::Tensor grads);
+                    at::Tensor grads, int64_t pool_mode, int64_t loss_mode);
 void cpu_update(at::Tensor params, at::Tensor grads, double dt, double scale);
 }  // namespace pcnn
 
@@ -25,6 +25,11 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
                        float* y, float* dz, float* dz2, float* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int mode, void* stream);
+int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
+                          void* a2, float* y, float* dz, float* dz2,
+                          float* dz1, const int* labels, float* loss_accum,
+                          int* correct, int B, int act_is_bf16, int mode,
+                          int pool_mode, int loss_mode, void* stream);
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
@@ -93,11 +98,11 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
                 at::Tensor y, at::Tensor dz, at::Tensor dz2, at::Tensor dz1,
                 at::Tensor labels, at::Tensor loss_accum,
                 at::Tensor correct_accum, int64_t B, int64_t mode,
-                int64_t stream) {
+                int64_t stream, int64_t pool_mode, int64_t loss_mode) {
   TORCH_CHECK(x.is_cuda() && params.is_cuda(), "expected device tensors");
   TORCH_CHECK(labels.scalar_type() == at::kInt, "labels must be int32");
   int f = act_flag(x);
-  check_hip(pcnn_launch_fwdbwd(
+  check_hip(pcnn_launch_fwdbwd_ex(
                 x.data_ptr(), params.data_ptr<float>(), a1.data_ptr(),
                 a2.data_ptr(), y.numel() ? y.data_ptr<float>() : nullptr,
                 dz.numel() ? dz.data_ptr<float>() : nullptr,
@@ -106,7 +111,8 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
                 labels.data_ptr<int>(),
                 loss_accum.numel() ? loss_accum.data_ptr<float>() : nullptr,
                 correct_accum.numel() ? correct_accum.data_ptr<int>() : nullptr,
-                (int)B, f, (int)mode, (void*)stream),
+                (int)B, f, (int)mode, (int)pool_mode, (int)loss_mode,
+                (void*)stream),
             "fwdbwd");
 }
 
@@ -153,7 +159,9 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
                      at::Tensor a2, at::Tensor y, at::Tensor dz,
                      at::Tensor dz2, at::Tensor dz1, at::Tensor loss_accum,
                      int64_t B, int64_t steps, int64_t chunk_imgs,
-                     double step_scale, int64_t stream) {
+                     double step_scale, int64_t stream, int64_t pool_mode,
+                     int64_t loss_mode) {
+  const int wroles = pool_mode == 1 ? 5 : 7;  // max pool: no pool wgrad
   TORCH_CHECK(x_pool.is_cuda() && x_pool.dim() == 2, "x_pool [P*B, 784]");
   TORCH_CHECK(labels_pool.scalar_type() == at::kInt, "labels must be int32");
   const int64_t pool_rows = x_pool.size(0);
@@ -170,16 +178,18 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
     const int64_t i = st % P;
     const void* xb = xp + (size_t)i * B * pcnn::IN_PIX * esz;
     const int* lb = lp + i * B;
-    check_hip(pcnn_launch_fwdbwd(xb, pp, a1.data_ptr(), a2.data_ptr(),
-                                 y.data_ptr<float>(), dz.data_ptr<float>(),
-                                 dz2.data_ptr<float>(), dz1.data_ptr<float>(),
-                                 lb, loss_accum.data_ptr<float>(), nullptr,
-                                 (int)B, f, 0, s),
+    check_hip(pcnn_launch_fwdbwd_ex(xb, pp, a1.data_ptr(), a2.data_ptr(),
+                                    y.data_ptr<float>(), dz.data_ptr<float>(),
+                                    dz2.data_ptr<float>(),
+                                    dz1.data_ptr<float>(), lb,
+                                    loss_accum.data_ptr<float>(), nullptr,
+                                    (int)B, f, 0, (int)pool_mode,
+                                    (int)loss_mode, s),
               "train_steps/fwdbwd");
-    check_hip(pcnn_launch_wgrad(xb, a1.data_ptr(), a2.data_ptr(),
-                                dz.data_ptr<float>(), dz2.data_ptr<float>(),
-                                dz1.data_ptr<float>(), gp, (int)B, f,
-                                (int)chunk_imgs, s),
+    check_hip(pcnn_launch_wgrad_ex(xb, a1.data_ptr(), a2.data_ptr(),
+                                   dz.data_ptr<float>(), dz2.data_ptr<float>(),
+                                   dz1.data_ptr<float>(), gp, (int)B, f,
+                                   (int)chunk_imgs, wroles, s),
               "train_steps/wgrad");
     check_hip(pcnn_launch_update(pp, gp, (float)step_scale, s),
               "train_steps/update");
@@ -323,14 +333,30 @@ void deep_mfma_selftest(at::Tensor A, at::Tensor Bm, at::Tensor D,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "parallel_cnn_amd native ops (CPU reference + gfx950 HIP kernels)";
-  m.def("cpu_forward", &pcnn::cpu_forward);
-  m.def("cpu_backward", &pcnn::cpu_backward);
+  m.def("cpu_forward", &pcnn::cpu_forward, py::arg("x"), py::arg("params"),
+        py::arg("a1"), py::arg("a2"), py::arg("y"), py::arg("pool_mode") = 0,
+        py::arg("loss_mode") = 0);
+  m.def("cpu_backward", &pcnn::cpu_backward, py::arg("x"), py::arg("params"),
+        py::arg("a1"), py::arg("a2"), py::arg("y"), py::arg("labels"),
+        py::arg("dz"), py::arg("dz2"), py::arg("dz1"), py::arg("grads"),
+        py::arg("pool_mode") = 0, py::arg("loss_mode") = 0);
   m.def("cpu_update", &pcnn::cpu_update);
-  m.def("hip_fwdbwd", &hip_fwdbwd);
+  m.def("hip_fwdbwd", &hip_fwdbwd, py::arg("x"), py::arg("params"),
+        py::arg("a1"), py::arg("a2"), py::arg("y"), py::arg("dz"),
+        py::arg("dz2"), py::arg("dz1"), py::arg("labels"),
+        py::arg("loss_accum"), py::arg("correct_accum"), py::arg("B"),
+        py::arg("mode"), py::arg("stream"), py::arg("pool_mode") = 0,
+        py::arg("loss_mode") = 0);
   m.def("hip_wgrad", &hip_wgrad);
   m.def("hip_wgrad_roles", &hip_wgrad_roles);
   m.def("hip_update", &hip_update);
-  m.def("hip_train_steps", &hip_train_steps);
+  m.def("hip_train_steps", &hip_train_steps, py::arg("x_pool"),
+        py::arg("labels_pool"), py::arg("params"), py::arg("grads"),
+        py::arg("a1"), py::arg("a2"), py::arg("y"), py::arg("dz"),
+        py::arg("dz2"), py::arg("dz1"), py::arg("loss_accum"), py::arg("B"),
+        py::arg("steps"), py::arg("chunk_imgs"), py::arg("step_scale"),
+        py::arg("stream"), py::arg("pool_mode") = 0,
+        py::arg("loss_mode") = 0);
   m.def("deep_im2col", &deep_im2col);
   m.def("deep_gemm", &deep_gemm);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm);

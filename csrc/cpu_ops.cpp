@@ -25,6 +25,7 @@
 #include <ATen/Parallel.h>
 #include <torch/extension.h>
 
+#include <algorithm>
 #include <cmath>
 #include <cstring>
 #include <vector>
@@ -35,9 +36,12 @@ namespace pcnn {
 
 static inline float sigmoidf(float v) { return 1.0f / (1.0f + std::exp(-v)); }
 
+// pool_mode: 0 = trainable (reference), 1 = max.  loss_mode: 0 = residual
+// (reference), 1 = softmax cross-entropy.
+
 // Forward for one image: x[784] -> a1[3456], a2[216], y[10].
 static void forward_one(const float* x, const float* p, float* a1, float* a2,
-                        float* y) {
+                        float* y, int pool_mode, int loss_mode) {
   const float* c1w = p + OFF_C1W;
   const float* c1b = p + OFF_C1B;
   const float* s1w = p + OFF_S1W;
@@ -58,25 +62,49 @@ static void forward_one(const float* x, const float* p, float* a1, float* a2,
       }
     }
   }
-  // trainable pool + sigmoid (shared 4x4 kernel, stride 4, scalar bias)
+  // pool + sigmoid: trainable shared 4x4 kernel (reference) or max
   for (int o = 0; o < C1_CH; ++o) {
     for (int pr = 0; pr < S1_H; ++pr) {
       for (int pc = 0; pc < S1_W; ++pc) {
-        float acc = s1b;
-        for (int i = 0; i < S1_K; ++i)
-          for (int j = 0; j < S1_K; ++j)
-            acc += s1w[i * S1_K + j] *
-                   a1[o * C1_PIX + (pr * S1_K + i) * C1_W + (pc * S1_K + j)];
+        float acc;
+        if (pool_mode == 1) {
+          acc = -1e30f;
+          for (int i = 0; i < S1_K; ++i)
+            for (int j = 0; j < S1_K; ++j)
+              acc = std::max(
+                  acc,
+                  a1[o * C1_PIX + (pr * S1_K + i) * C1_W + (pc * S1_K + j)]);
+        } else {
+          acc = s1b;
+          for (int i = 0; i < S1_K; ++i)
+            for (int j = 0; j < S1_K; ++j)
+              acc += s1w[i * S1_K + j] *
+                     a1[o * C1_PIX + (pr * S1_K + i) * C1_W +
+                        (pc * S1_K + j)];
+        }
         a2[o * S1_PIX + pr * S1_W + pc] = sigmoidf(acc);
       }
     }
   }
-  // fc + sigmoid
+  // fc + sigmoid (residual) or softmax (cross-entropy)
+  float z[FC_OUT];
   for (int k = 0; k < FC_OUT; ++k) {
     float acc = fb[k];
     const float* wk = fw + k * FC_IN;
     for (int m = 0; m < FC_IN; ++m) acc += wk[m] * a2[m];
-    y[k] = sigmoidf(acc);
+    z[k] = acc;
+  }
+  if (loss_mode == 1) {
+    float mx = z[0];
+    for (int k = 1; k < FC_OUT; ++k) mx = std::max(mx, z[k]);
+    float sum = 0.f;
+    for (int k = 0; k < FC_OUT; ++k) {
+      y[k] = std::exp(z[k] - mx);
+      sum += y[k];
+    }
+    for (int k = 0; k < FC_OUT; ++k) y[k] /= sum;
+  } else {
+    for (int k = 0; k < FC_OUT; ++k) y[k] = sigmoidf(z[k]);
   }
 }
 
@@ -84,16 +112,20 @@ static void forward_one(const float* x, const float* p, float* a1, float* a2,
 // parameter gradients (sum) into g[N_PARAMS]; returns ||dz||_2.
 static float backward_one(const float* x, const float* p, const float* a1,
                           const float* a2, const float* y, int64_t label,
-                          float* dz, float* dz2, float* dz1, float* g) {
+                          float* dz, float* dz2, float* dz1, float* g,
+                          int pool_mode, int loss_mode) {
   const float* s1w = p + OFF_S1W;
   const float* fw = p + OFF_FW;
 
-  // residual loss gradient (no output sigmoid')
+  // residual / softmax-CE gradient: dz = onehot - y in both conventions
   float sq = 0.f;
   for (int k = 0; k < FC_OUT; ++k) {
     dz[k] = (k == label ? 1.0f : 0.0f) - y[k];
     sq += dz[k] * dz[k];
   }
+  const float loss_v = loss_mode == 1
+                           ? -std::log(std::max(y[label], 1e-30f))
+                           : std::sqrt(sq);
 
   // fc wgrad / bgrad
   for (int k = 0; k < FC_OUT; ++k) {
@@ -111,29 +143,58 @@ static float backward_one(const float* x, const float* p, const float* a1,
     dz2[m] = da * v * (1.0f - v);
     s1b_acc += dz2[m];
   }
-  g[OFF_S1B] += s1b_acc / (float)S1_OUT;
-
-  // pool wgrad
-  for (int i = 0; i < S1_K; ++i) {
-    for (int j = 0; j < S1_K; ++j) {
-      float acc = 0.f;
-      for (int o = 0; o < C1_CH; ++o)
-        for (int pr = 0; pr < S1_H; ++pr)
-          for (int pc = 0; pc < S1_W; ++pc)
-            acc += dz2[o * S1_PIX + pr * S1_W + pc] *
-                   a1[o * C1_PIX + (pr * S1_K + i) * C1_W + (pc * S1_K + j)];
-      g[OFF_S1W + i * S1_K + j] += acc;
+  if (pool_mode == 0) {
+    g[OFF_S1B] += s1b_acc / (float)S1_OUT;
+    // pool wgrad (trainable only; max pool has no parameters)
+    for (int i = 0; i < S1_K; ++i) {
+      for (int j = 0; j < S1_K; ++j) {
+        float acc = 0.f;
+        for (int o = 0; o < C1_CH; ++o)
+          for (int pr = 0; pr < S1_H; ++pr)
+            for (int pc = 0; pc < S1_W; ++pc)
+              acc +=
+                  dz2[o * S1_PIX + pr * S1_W + pc] *
+                  a1[o * C1_PIX + (pr * S1_K + i) * C1_W + (pc * S1_K + j)];
+        g[OFF_S1W + i * S1_K + j] += acc;
+      }
     }
   }
 
-  // conv1 output grad (gather: stride==kernel, non-overlapping) -> preact grad
-  for (int o = 0; o < C1_CH; ++o) {
-    for (int r = 0; r < C1_H; ++r) {
-      for (int c = 0; c < C1_W; ++c) {
-        float da = dz2[o * S1_PIX + (r / S1_K) * S1_W + (c / S1_K)] *
-                   s1w[(r % S1_K) * S1_K + (c % S1_K)];
-        float v = a1[o * C1_PIX + r * C1_W + c];
-        dz1[o * C1_PIX + r * C1_W + c] = da * v * (1.0f - v);
+  // conv1 output grad -> preact grad.  Trainable pool: every position gets
+  // dz2 * kernel weight; max pool: only the window argmax gets dz2.
+  if (pool_mode == 1) {
+    for (int o = 0; o < C1_CH; ++o) {
+      for (int pr = 0; pr < S1_H; ++pr) {
+        for (int pc = 0; pc < S1_W; ++pc) {
+          int best = 0;
+          float bv = -1e30f;
+          for (int t = 0; t < S1_K * S1_K; ++t) {
+            const int r = pr * S1_K + t / S1_K, c = pc * S1_K + t % S1_K;
+            const float v = a1[o * C1_PIX + r * C1_W + c];
+            if (v > bv) {
+              bv = v;
+              best = t;
+            }
+          }
+          for (int t = 0; t < S1_K * S1_K; ++t) {
+            const int r = pr * S1_K + t / S1_K, c = pc * S1_K + t % S1_K;
+            const float v = a1[o * C1_PIX + r * C1_W + c];
+            const float da =
+                (t == best) ? dz2[o * S1_PIX + pr * S1_W + pc] : 0.f;
+            dz1[o * C1_PIX + r * C1_W + c] = da * v * (1.0f - v);
+          }
+        }
+      }
+    }
+  } else {
+    for (int o = 0; o < C1_CH; ++o) {
+      for (int r = 0; r < C1_H; ++r) {
+        for (int c = 0; c < C1_W; ++c) {
+          float da = dz2[o * S1_PIX + (r / S1_K) * S1_W + (c / S1_K)] *
+                     s1w[(r % S1_K) * S1_K + (c % S1_K)];
+          float v = a1[o * C1_PIX + r * C1_W + c];
+          dz1[o * C1_PIX + r * C1_W + c] = da * v * (1.0f - v);
+        }
       }
     }
   }
@@ -155,7 +216,7 @@ static float backward_one(const float* x, const float* p, const float* a1,
       }
     }
   }
-  return std::sqrt(sq);
+  return loss_v;
 }
 
 static void check_cpu_f32(const at::Tensor& t, const char* name, int64_t numel) {
@@ -167,7 +228,7 @@ static void check_cpu_f32(const at::Tensor& t, const char* name, int64_t numel) 
 }
 
 void cpu_forward(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
-                 at::Tensor y) {
+                 at::Tensor y, int64_t pool_mode, int64_t loss_mode) {
   int64_t B = x.size(0);
   check_cpu_f32(x, "x", B * IN_PIX);
   check_cpu_f32(params, "params", N_PARAMS);
@@ -182,7 +243,7 @@ void cpu_forward(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
   at::parallel_for(0, B, 1, [&](int64_t b0, int64_t b1) {
     for (int64_t b = b0; b < b1; ++b)
       forward_one(xp + b * IN_PIX, pp, a1p + b * C1_OUT, a2p + b * S1_OUT,
-                  yp + b * FC_OUT);
+                  yp + b * FC_OUT, (int)pool_mode, (int)loss_mode);
   });
 }
 
@@ -191,7 +252,7 @@ void cpu_forward(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
 double cpu_backward(at::Tensor x, at::Tensor params, at::Tensor a1,
                     at::Tensor a2, at::Tensor y, at::Tensor labels,
                     at::Tensor dz, at::Tensor dz2, at::Tensor dz1,
-                    at::Tensor grads) {
+                    at::Tensor grads, int64_t pool_mode, int64_t loss_mode) {
   int64_t B = x.size(0);
   check_cpu_f32(x, "x", B * IN_PIX);
   check_cpu_f32(params, "params", N_PARAMS);
@@ -226,7 +287,8 @@ double cpu_backward(at::Tensor x, at::Tensor params, at::Tensor a1,
       losses[tid] += backward_one(xp + b * IN_PIX, pp, a1p + b * C1_OUT,
                                   a2p + b * S1_OUT, yp + b * FC_OUT, lp[b],
                                   dzp + b * FC_OUT, dz2p + b * S1_OUT,
-                                  dz1p + b * C1_OUT, g.data());
+                                  dz1p + b * C1_OUT, g.data(),
+                                  (int)pool_mode, (int)loss_mode);
     }
   });
   double loss = 0.0;

@@ -131,7 +131,8 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     act_t* __restrict__ a1g, act_t* __restrict__ a2g, float* __restrict__ yg,
     float* __restrict__ dzg, float* __restrict__ dz2g,
     float* __restrict__ dz1g, const int* __restrict__ labels,
-    float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B) {
+    float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B,
+    int pool_mode, int loss_mode) {
   __shared__ FwdLds L;
   const int b = blockIdx.x;
   if (b >= B) return;
@@ -161,7 +162,8 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
              &xw[u][v4 * 4]);
     const float* w = &L.ps[OFF_C1W + o * C1_K * C1_K];
     const float cb = L.ps[OFF_C1B + o];
-    float pacc = L.ps[OFF_S1B];
+    // pool preact: trainable weighted sum (reference) or max
+    float pacc = pool_mode == 1 ? -1e30f : L.ps[OFF_S1B];
 #pragma unroll
     for (int i = 0; i < S1_K; ++i) {
 #pragma unroll
@@ -174,7 +176,10 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
             acc += w[u * C1_K + v] * xw[i + u][j + v];
         const float av = sigmoidf_dev(acc);
         a1v[i * S1_K + j] = av;
-        pacc += L.ps[OFF_S1W + i * S1_K + j] * av;
+        if (pool_mode == 1)
+          pacc = fmaxf(pacc, av);
+        else
+          pacc += L.ps[OFF_S1W + i * S1_K + j] * av;
       }
     }
     if (MODE == MODE_TRAIN) {
@@ -207,18 +212,52 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 #pragma unroll
     for (int off = 8; off > 0; off >>= 1) p += __shfl_down(p, off, 16);
     if (l == 0) {
-      const float v = sigmoidf_dev(p + params[OFF_FB + k]);
-      L.ys[k] = v;
-      if (yg != nullptr) yg[(size_t)b * FC_OUT + k] = v;
-      if (MODE == MODE_TRAIN) {
-        const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
-        L.dzs[k] = d;
-        L.sq[k] = d * d;
-        dzg[(size_t)b * FC_OUT + k] = d;
+      const float z = p + params[OFF_FB + k];
+      if (loss_mode == 1) {
+        L.ys[k] = z;  // raw logit; softmax finished below
+      } else {
+        const float v = sigmoidf_dev(z);
+        L.ys[k] = v;
+        if (yg != nullptr) yg[(size_t)b * FC_OUT + k] = v;
+        if (MODE == MODE_TRAIN) {
+          const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
+          L.dzs[k] = d;
+          L.sq[k] = d * d;
+          dzg[(size_t)b * FC_OUT + k] = d;
+        }
       }
     }
   }
   __syncthreads();
+  if (loss_mode == 1) {
+    // softmax over the 10 logits + CE residual (dz = onehot - softmax)
+    if (tid == 0) {
+      float mx = L.ys[0];
+#pragma unroll
+      for (int k = 1; k < FC_OUT; ++k) mx = fmaxf(mx, L.ys[k]);
+      float sum = 0.f;
+      float e[FC_OUT];
+#pragma unroll
+      for (int k = 0; k < FC_OUT; ++k) {
+        e[k] = __expf(L.ys[k] - mx);
+        sum += e[k];
+      }
+#pragma unroll
+      for (int k = 0; k < FC_OUT; ++k) {
+        const float v = e[k] / sum;
+        L.ys[k] = v;
+        if (yg != nullptr) yg[(size_t)b * FC_OUT + k] = v;
+        if (MODE == MODE_TRAIN) {
+          const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
+          L.dzs[k] = d;
+          dzg[(size_t)b * FC_OUT + k] = d;
+        }
+      }
+      if (MODE == MODE_TRAIN)
+        L.sq[0] = -__logf(fmaxf(L.ys[labels[b]], 1e-30f));
+    }
+    __syncthreads();
+  }
 
   if (MODE == MODE_EVAL) {
     // argmax + correct-count (replaces the reference's per-image D2H copy +
@@ -241,24 +280,43 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
       da += params[OFF_FW + k * FC_IN + tid] * L.dzs[k];
     const float d2 = da * a2v * (1.0f - a2v);
     dz2g[(size_t)b * S1_OUT + tid] = d2;
-    // pool backward for this thread's own 16 conv positions (registers)
+    // pool backward for this thread's own 16 conv positions (registers):
+    // trainable -> d2 * kernel weight everywhere; max -> d2 at the argmax
+    int best = 0;
+    if (pool_mode == 1) {
+      float bv = a1v[0];
+#pragma unroll
+      for (int t = 1; t < S1_K * S1_K; ++t)
+        if (a1v[t] > bv) {
+          bv = a1v[t];
+          best = t;
+        }
+    }
 #pragma unroll
     for (int i = 0; i < S1_K; ++i) {
       float row[S1_K];
 #pragma unroll
       for (int j = 0; j < S1_K; ++j) {
         const float av = a1v[i * S1_K + j];
-        row[j] = d2 * L.ps[OFF_S1W + i * S1_K + j] * av * (1.0f - av);
+        const float dd =
+            pool_mode == 1
+                ? (i * S1_K + j == best ? d2 : 0.f)
+                : d2 * L.ps[OFF_S1W + i * S1_K + j];
+        row[j] = dd * av * (1.0f - av);
       }
       *reinterpret_cast<float4*>(
           dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
           pc * S1_K) = *reinterpret_cast<const float4*>(row);
     }
   } else if (tid == S1_OUT && loss_accum != nullptr) {
-    float ssum = 0.f;
+    if (loss_mode == 1) {
+      unsafeAtomicAdd(loss_accum, L.sq[0]);
+    } else {
+      float ssum = 0.f;
 #pragma unroll
-    for (int k = 0; k < FC_OUT; ++k) ssum += L.sq[k];
-    unsafeAtomicAdd(loss_accum, sqrtf(ssum));
+      for (int k = 0; k < FC_OUT; ++k) ssum += L.sq[k];
+      unsafeAtomicAdd(loss_accum, sqrtf(ssum));
+    }
   }
 }
 
@@ -509,20 +567,24 @@ template <int MODE>
 int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
                        float* y, float* dz, float* dz2, float* dz1,
                        const int* labels, float* loss_accum, int* correct,
-                       int B, int act_is_bf16, hipStream_t stream) {
+                       int B, int act_is_bf16, int pool_mode, int loss_mode,
+                       hipStream_t stream) {
   dim3 grid(B), block(256);
   if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_fwdbwd<bf16, MODE>), grid, block, 0, stream,
                        (const bf16*)x, params, (bf16*)a1, (bf16*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B);
+                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
+                       loss_mode);
   } else if (act_is_bf16 == 2) {
     hipLaunchKernelGGL((k_fwdbwd<fp16, MODE>), grid, block, 0, stream,
                        (const fp16*)x, params, (fp16*)a1, (fp16*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B);
+                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
+                       loss_mode);
   } else {
     hipLaunchKernelGGL((k_fwdbwd<float, MODE>), grid, block, 0, stream,
                        (const float*)x, params, (float*)a1, (float*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B);
+                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
+                       loss_mode);
   }
   return (int)hipGetLastError();
 }
@@ -530,25 +592,38 @@ int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
 
 extern "C" {
 
+int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
+                          void* a2, float* y, float* dz, float* dz2,
+                          float* dz1, const int* labels, float* loss_accum,
+                          int* correct, int B, int act_is_bf16, int mode,
+                          int pool_mode, int loss_mode, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  switch (mode) {
+    case MODE_TRAIN:
+      return launch_fwdbwd_mode<MODE_TRAIN>(x, params, a1, a2, y, dz, dz2,
+                                            dz1, labels, loss_accum, correct,
+                                            B, act_is_bf16, pool_mode,
+                                            loss_mode, s);
+    case MODE_EVAL:
+      return launch_fwdbwd_mode<MODE_EVAL>(x, params, a1, a2, y, dz, dz2, dz1,
+                                           labels, loss_accum, correct, B,
+                                           act_is_bf16, pool_mode, loss_mode,
+                                           s);
+    default:
+      return launch_fwdbwd_mode<MODE_INFER>(x, params, a1, a2, y, dz, dz2,
+                                            dz1, labels, loss_accum, correct,
+                                            B, act_is_bf16, pool_mode,
+                                            loss_mode, s);
+  }
+}
+
 int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
                        float* y, float* dz, float* dz2, float* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int mode, void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  switch (mode) {
-    case MODE_TRAIN:
-      return launch_fwdbwd_mode<MODE_TRAIN>(x, params, a1, a2, y, dz, dz2, dz1,
-                                            labels, loss_accum, correct, B,
-                                            act_is_bf16, s);
-    case MODE_EVAL:
-      return launch_fwdbwd_mode<MODE_EVAL>(x, params, a1, a2, y, dz, dz2, dz1,
-                                           labels, loss_accum, correct, B,
-                                           act_is_bf16, s);
-    default:
-      return launch_fwdbwd_mode<MODE_INFER>(x, params, a1, a2, y, dz, dz2, dz1,
-                                            labels, loss_accum, correct, B,
-                                            act_is_bf16, s);
-  }
+  return pcnn_launch_fwdbwd_ex(x, params, a1, a2, y, dz, dz2, dz1, labels,
+                               loss_accum, correct, B, act_is_bf16, mode, 0,
+                               0, stream);
 }
 
 // chunk_imgs is the conv1 slices-per-channel knob (GC); <=0 -> default.

@@ -18,6 +18,8 @@ class TrainConfig:
     # model / numerics
     model: str = "lenet5"
     act_dtype: str = "bf16"          # activation storage on GPU: bf16 | fp16 | fp32
+    pool: str = "trainable"          # trainable (reference) | max
+    loss: str = "residual"           # residual (reference) | softmax_ce
     seed: int = 0
 
     # optimizer (reference semantics: p += dt * grad)

@@ -72,6 +72,10 @@ class Trainer:
         # body resolves the live stream instead)
         self._sh = native.current_stream_handle() if self.backend == "hip" \
             else 0
+        self._pool_mode = 1 if cfg.pool == "max" else 0
+        self._loss_mode = 1 if cfg.loss == "softmax_ce" else 0
+        # max pool has no parameters: skip the pool wgrad role
+        self._wroles = 5 if self._pool_mode == 1 else 7
 
     # ------------------------------------------------------------------ util
     def _update_scale(self, local_batch: int) -> float:
@@ -101,9 +105,11 @@ class Trainer:
             stream = self._sh
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
-                               B, MODE_TRAIN, stream)
-            self._C.hip_wgrad(x, w.a1, w.a2, w.dz, w.dz2, w.dz1, m.grads, B,
-                              self.cfg.wgrad_chunk, stream)
+                               B, MODE_TRAIN, stream, self._pool_mode,
+                               self._loss_mode)
+            self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                    m.grads, B, self.cfg.wgrad_chunk,
+                                    self._wroles, stream)
             pdist.allreduce_grads(m.grads)
             self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
                                stream)
@@ -111,17 +117,20 @@ class Trainer:
             # CPU path keeps fp32 activations in the workspace directly.
             a1 = self._cpu_view(w.a1, B)
             a2 = self._cpu_view(w.a2, B)
-            self._C.cpu_forward(x, m.params, a1, a2, w.y[:B])
+            self._C.cpu_forward(x, m.params, a1, a2, w.y[:B],
+                                self._pool_mode, self._loss_mode)
             loss = self._C.cpu_backward(x, m.params, a1, a2, w.y[:B], labels,
                                         w.dz[:B], w.dz2[:B], w.dz1[:B],
-                                        m.grads)
+                                        m.grads, self._pool_mode,
+                                        self._loss_mode)
             self._loss_host += loss
             pdist.allreduce_grads(m.grads)
             self._C.cpu_update(m.params, m.grads, self.cfg.dt, scale)
         else:  # torchref
-            a1, a2, y = torch_ref.forward(x, m.params)
+            a1, a2, y = torch_ref.forward(x, m.params, self.cfg.pool,
+                                          self.cfg.loss)
             dz, dz2, dz1, grads, loss = torch_ref.backward(
-                x, m.params, a1, a2, y, labels)
+                x, m.params, a1, a2, y, labels, self.cfg.pool, self.cfg.loss)
             self._loss_host += loss
             m.grads += grads
             pdist.allreduce_grads(m.grads)
@@ -168,9 +177,11 @@ class Trainer:
         stream = native.current_stream_handle()
         self._C.hip_fwdbwd(self._gx, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                            w.dz1, self._gl, w.loss_accum, w.correct_accum,
-                           B, MODE_TRAIN, stream)
-        self._C.hip_wgrad(self._gx, w.a1, w.a2, w.dz, w.dz2, w.dz1, m.grads,
-                          B, self.cfg.wgrad_chunk, stream)
+                           B, MODE_TRAIN, stream, self._pool_mode,
+                           self._loss_mode)
+        self._C.hip_wgrad_roles(self._gx, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                m.grads, B, self.cfg.wgrad_chunk,
+                                self._wroles, stream)
         pdist.allreduce_grads(m.grads)
         self._C.hip_update(m.params, m.grads,
                            self.cfg.dt * self._update_scale(B),
@@ -203,7 +214,8 @@ class Trainer:
                 x_pool, labels_pool, self.model.params, self.model.grads,
                 w.a1, w.a2, w.y, w.dz, w.dz2, w.dz1, w.loss_accum, B, steps,
                 self.cfg.wgrad_chunk, self.cfg.dt * self._update_scale(B),
-                native.current_stream_handle())
+                native.current_stream_handle(), self._pool_mode,
+                self._loss_mode)
             self._samples_seen += B * steps
             self.global_step += steps
         else:
@@ -252,7 +264,8 @@ class Trainer:
                 self._C.hip_fwdbwd(xb, self.model.params, w.a1, w.a2, w.y,
                                    w.dz, w.dz2, w.dz1, lb, w.loss_accum,
                                    w.correct_accum, B, MODE_EVAL,
-                                   native.current_stream_handle())
+                                   native.current_stream_handle(),
+                                   self._pool_mode, self._loss_mode)
             else:
                 preds = self._forward_preds(xb, B)
                 correct += int((preds == lb).sum().item())
@@ -267,10 +280,12 @@ class Trainer:
         w = self.ws
         if self.backend == "cpu":
             a1, a2 = self._cpu_view(w.a1, B), self._cpu_view(w.a2, B)
-            self._C.cpu_forward(xb, self.model.params, a1, a2, w.y[:B])
+            self._C.cpu_forward(xb, self.model.params, a1, a2, w.y[:B],
+                                self._pool_mode, self._loss_mode)
             y = w.y[:B]
         else:
-            _, _, y = torch_ref.forward(xb, self.model.params)
+            _, _, y = torch_ref.forward(xb, self.model.params, self.cfg.pool,
+                                        self.cfg.loss)
         return y.argmax(dim=1)
 
     @torch.no_grad()
@@ -289,7 +304,8 @@ class Trainer:
                 self._C.hip_fwdbwd(xd, self.model.params, w.a1, w.a2, w.y,
                                    w.dz, w.dz2, w.dz1, dummy, w.loss_accum,
                                    w.correct_accum, B, MODE_INFER,
-                                   native.current_stream_handle())
+                                   native.current_stream_handle(),
+                                   self._pool_mode, self._loss_mode)
                 preds.append(w.y[:B].argmax(dim=1).cpu())
             else:
                 xd = xb.to(torch.float32)

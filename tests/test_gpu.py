@@ -254,6 +254,38 @@ def test_native_extension_is_loaded_on_gpu(device):
     assert t.backend == "hip"
 
 
+@pytest.mark.parametrize("pool,loss", [("max", "residual"),
+                                       ("trainable", "softmax_ce"),
+                                       ("max", "softmax_ce")])
+def test_gpu_modes_match_oracle(pool, loss, device):
+    """Max-pool and softmax-CE kernel modes vs the fp32 torch oracle:
+    full step trajectory (fwdbwd + wgrad + update)."""
+    B = 16
+    cfg_g = TrainConfig(batch_size=B, device="cuda", backend="hip",
+                        act_dtype="fp32", log_interval=0, pool=pool,
+                        loss=loss)
+    cfg_c = TrainConfig(batch_size=B, device="cpu", backend="cpu",
+                        log_interval=0, pool=pool, loss=loss)
+    tg, tc = Trainer(cfg_g), Trainer(cfg_c)
+    x, y = synthetic_mnist(B * 3, seed=23)
+    for s in range(3):
+        xb, yb = x[s * B:(s + 1) * B], y[s * B:(s + 1) * B]
+        tg.step(*tg.stage_batch(xb, yb))
+        tc.step(*tc.stage_batch(xb, yb))
+    torch.cuda.synchronize()
+    diff = (tg.model.params.cpu() - tc.model.params).abs().max().item()
+    assert diff < 1e-3, f"{pool}/{loss}: {diff}"
+    lg, ng = tg.consume_loss()
+    lc, nc = tc.consume_loss()
+    assert ng == nc
+    assert abs(lg - lc) < 1e-2 * max(1.0, lc), (lg, lc)
+    if pool == "max":
+        from parallel_cnn_amd.ops import shapes as SS
+        s1grad = tg.model.params[SS.OFF_S1W:SS.OFF_FW].cpu()
+        s1init = tc.model.params[SS.OFF_S1W:SS.OFF_FW]
+        assert torch.equal(s1grad, s1init)  # pool params untouched
+
+
 def test_graph_step_matches_eager(device):
     """hipGraph-captured step replay == eager step trajectory."""
     x, y = synthetic_mnist(64, seed=19)
