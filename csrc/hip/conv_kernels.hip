@@ -199,36 +199,36 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   const int row_a = tid >> 2;            // 64 rows, 4 threads each
   const int kq = (tid & 3) * 16;         // 16 k per thread (two 8-chunks)
-  for (int kt = 0; kt < K; kt += BK) {
-    // stage A tile [BM][BK]; K % 32 == 0, so each 8-chunk is fully in or
-    // fully out of range (zeros otherwise)
-    {
-      const long long m = m0 + row_a;
+  // T14-style software pipeline: next tile's global loads are issued into
+  // registers BEFORE the MFMA block (HBM/L2 latency hides under compute);
+  // the LDS write happens after the read barrier.
+  float ra[2][8];    // A prefetch
+  float rb[2][8];    // B prefetch
+  const long long m_a = m0 + row_a;
+
+  auto load_regs = [&](int kt) {
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const int kk = kq + h * 8;
-        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (m < M && (kt + kk) < K) ld8v(A + m * ldA + kt + kk, v8);
+    for (int h = 0; h < 2; ++h) {
+      const int kk = kq + h * 8;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)v8[u];
-      }
+      for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
+      if (m_a < M && (kt + kk) < K) ld8v(A + m_a * ldA + kt + kk, ra[h]);
     }
-    // stage B tile into [n][k] image
     if (b_kxn) {
-      // Bsrc[K][N]: read rows k (coalesced along n), write transposed
-      const int k = tid >> 2;            // 64 k rows, 4 threads each
-      const int nq = (tid & 3) * 16;     // 16 n per thread
+      // Bsrc[K][N]: thread owns row k = tid>>2, 16 n (transposed write)
+      const int k = tid >> 2;
+      const int nq = (tid & 3) * 16;
       const bool kok = (kt + k) < K;
 #pragma unroll
-      for (int u = 0; u < 16; ++u) {
-        const int n = n0 + nq + u;
-        const float v =
-            (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
-        L.Bs[nq + u][k] = (__bf16)v;
-      }
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int n = n0 + nq + h * 8 + u;
+          rb[h][u] = (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
+        }
     } else {
-      // Bsrc[N][K]: row per n, staged directly
-      const int n = tid >> 2;            // 64 n rows, 4 threads each
+      // Bsrc[N][K]: row per n
+      const int n = tid >> 2;
       const bool ok = (n0 + n) < N;
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
@@ -236,11 +236,43 @@ __global__ __launch_bounds__(256) void k_gemm(
         const float* src = Bsrc + (long long)(n0 + n) * K + kt + kk;
         const bool kok = ok && (kt + kk) < K;
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          L.Bs[n][kk + u] = (__bf16)(kok ? src[u] : 0.f);
+        for (int u = 0; u < 8; ++u) rb[h][u] = kok ? src[u] : 0.f;
       }
     }
-    __syncthreads();
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int kk = kq + h * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)ra[h][u];
+    }
+    if (b_kxn) {
+      const int k = tid >> 2;
+      const int nq = (tid & 3) * 16;
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          L.Bs[nq + h * 8 + u][k] = (__bf16)rb[h][u];
+    } else {
+      const int n = tid >> 2;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int kk = kq + h * 8;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) L.Bs[n][kk + u] = (__bf16)rb[h][u];
+      }
+    }
+  };
+
+  load_regs(0);
+  write_lds();
+  for (int kt = 0; kt < K; kt += BK) {
+    __syncthreads();  // tile kt visible in LDS
+    const bool more = kt + BK < K;
+    if (more) load_regs(kt + BK);
     // wave wv owns C rows [wv*16, wv*16+16); two 32-deep MFMA sub-steps
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -256,7 +288,8 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
       }
     }
-    __syncthreads();
+    __syncthreads();  // all reads of tile kt done
+    if (more) write_lds();
   }
 
   // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
@@ -310,51 +343,59 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 
   const int row_s = tid >> 2;        // source row m (64 rows, 4 thr each)
   const int cq = (tid & 3) * 16;     // 16 columns per thread
+  // T14-style pipeline (see k_gemm): prefetch the next M-chunk into
+  // registers under the MFMAs, write to LDS after the read barrier.
+  float rc[2][8];  // cols prefetch
+  float rd[2][8];  // dpre prefetch
+
+  auto load_regs = [&](long long mt) {
+    const long long m = mt + row_s;
+    const bool ok = m < m_hi && m < M;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = cq + h * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        rc[h][u] = 0.f;
+        rd[h][u] = 0.f;
+      }
+      if (ok && (kc0 + c + 8) <= KcP) {
+        ld8v(cols + m * KcP + kc0 + c, rc[h]);
+      } else if (ok) {
+        const act_t* src = cols + m * KcP + kc0 + c;
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          if ((kc0 + c + u) < KcP) rc[h][u] = (float)src[u];
+      }
+      if (ok && (n0 + c + 8) <= N) {
+        ld8v(dpre + m * N + n0 + c, rd[h]);
+      } else if (ok) {
+        const act_t* src = dpre + m * N + n0 + c;
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          if ((n0 + c + u) < N) rd[h][u] = (float)src[u];
+      }
+    }
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = cq + h * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        L.As[c + u][row_s] = (__bf16)rc[h][u];
+        L.Bs[c + u][row_s] = (__bf16)rd[h][u];
+      }
+    }
+  };
+
+  load_regs(m_lo);
+  write_lds();
   for (long long mt = m_lo; mt < m_hi; mt += BK) {
-    // stage cols chunk [32m][64kc] -> LDS As[kc][m] (transposed).
-    // Guard kc against KcP: the last kc-tile of a 96-wide cols buffer
-    // would otherwise read past the row (and past the allocation on the
-    // final row).
-    {
-      const long long m = mt + row_s;
-      const bool ok = m < m_hi && m < M;
-#pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const int c = cq + h * 8;
-        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (ok && (kc0 + c + 8) <= KcP) {
-          ld8v(cols + m * KcP + kc0 + c, v8);
-        } else if (ok) {
-          const act_t* src = cols + m * KcP + kc0 + c;
-#pragma unroll
-          for (int u = 0; u < 8; ++u)
-            if ((kc0 + c + u) < KcP) v8[u] = (float)src[u];
-        }
-#pragma unroll
-        for (int u = 0; u < 8; ++u) L.As[c + u][row_s] = (__bf16)v8[u];
-      }
-    }
-    // stage dpre chunk [32m][<=64n] -> LDS Bs[n][m] (transposed)
-    {
-      const long long m = mt + row_s;
-      const bool ok = m < m_hi && m < M;
-#pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const int c = cq + h * 8;
-        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (ok && (n0 + c + 8) <= N) {
-          ld8v(dpre + m * N + n0 + c, v8);
-        } else if (ok) {
-          const act_t* src = dpre + m * N + n0 + c;
-#pragma unroll
-          for (int u = 0; u < 8; ++u)
-            if ((n0 + c + u) < N) v8[u] = (float)src[u];
-        }
-#pragma unroll
-        for (int u = 0; u < 8; ++u) L.Bs[c + u][row_s] = (__bf16)v8[u];
-      }
-    }
     __syncthreads();
+    const bool more = mt + BK < m_hi;
+    if (more) load_regs(mt + BK);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
@@ -370,6 +411,7 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
       }
     }
     __syncthreads();
+    if (more) write_lds();
   }
 
   const int crow = wv * 16 + (lane >> 4) * 4;  // kc within tile
