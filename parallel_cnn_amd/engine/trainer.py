@@ -22,6 +22,7 @@ from ..models.lenet import LeNet5
 from ..ops import native, torch_ref
 from ..ops import shapes as S
 from ..parallel import dist as pdist
+from ..utils.timers import PhaseTimers
 
 MODE_TRAIN, MODE_EVAL, MODE_INFER = 0, 1, 2
 
@@ -72,6 +73,7 @@ class Trainer:
         # body resolves the live stream instead)
         self._sh = native.current_stream_handle() if self.backend == "hip" \
             else 0
+        self.timers: Optional[PhaseTimers] = None  # set by enable_profiling
         self._pool_mode = 1 if cfg.pool == "max" else 0
         self._loss_mode = 1 if cfg.loss == "softmax_ce" else 0
         # max pool has no parameters: skip the pool wgrad role
@@ -94,9 +96,17 @@ class Trainer:
             ld = labels.to(torch.int64)
         return xd, ld
 
+    def enable_profiling(self) -> None:
+        """Per-phase sync-correct timers (the framework analog of the
+        reference's per-layer clock() accumulators, but device-synced).
+        Adds a device sync per phase — profiling mode only."""
+        self.timers = PhaseTimers()
+
     # ------------------------------------------------------------------ step
     def step(self, x: torch.Tensor, labels: torch.Tensor) -> None:
         """One training step on an already-staged batch."""
+        if self.timers is not None:
+            return self._step_profiled(x, labels)
         B = x.shape[0]
         assert B <= self.ws.max_batch
         m, w = self.model, self.ws
@@ -135,6 +145,36 @@ class Trainer:
             m.grads += grads
             pdist.allreduce_grads(m.grads)
             torch_ref.update(m.params, m.grads, self.cfg.dt, scale)
+        self._samples_seen += B * self.ctx.world_size
+        self.global_step += 1
+
+    def _step_profiled(self, x: torch.Tensor, labels: torch.Tensor) -> None:
+        B = x.shape[0]
+        m, w, t = self.model, self.ws, self.timers
+        scale = self._update_scale(B)
+        if self.backend != "hip":
+            with t.phase("step"):
+                self.timers = None
+                try:
+                    self.step(x, labels)
+                finally:
+                    self.timers = t
+            return
+        stream = self._sh
+        with t.phase("fwd+bwd-data"):
+            self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
+                               w.dz1, labels, w.loss_accum, w.correct_accum,
+                               B, MODE_TRAIN, stream, self._pool_mode,
+                               self._loss_mode)
+        with t.phase("weight-grad"):
+            self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                    m.grads, B, self.cfg.wgrad_chunk,
+                                    self._wroles, stream)
+        with t.phase("all-reduce"):
+            pdist.allreduce_grads(m.grads)
+        with t.phase("update"):
+            self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
+                               stream)
         self._samples_seen += B * self.ctx.world_size
         self.global_step += 1
 

@@ -43,12 +43,18 @@ def load_datasets(cfg: TrainConfig):
 def main(argv=None) -> int:
     p = argparse.ArgumentParser(description=__doc__)
     TrainConfig.add_cli_args(p)
+    p.add_argument("--profile", action="store_true",
+                   help="per-phase device-synced timing report")
     args = p.parse_args(argv)
     cfg = TrainConfig.from_args(args)
 
+    import sys
+    profile = "--profile" in (argv or sys.argv)
     ctx = pdist.init_from_env(cfg.resolved_device())
     trainer = (DeepTrainer(cfg, ctx=ctx) if cfg.model == "deepcnn"
                else Trainer(cfg, ctx=ctx))
+    if profile and hasattr(trainer, "enable_profiling"):
+        trainer.enable_profiling()
     if cfg.ckpt_load:
         trainer.model.load(cfg.ckpt_load)
 
@@ -77,6 +83,9 @@ def main(argv=None) -> int:
     err_rate = trainer.evaluate(xte, yte)
     if ctx.is_main:
         print(f"Error Rate: {err_rate:.2f}%", flush=True)
+        if profile and getattr(trainer, "timers", None) is not None:
+            print("-- per-phase timers (device-synced) --")
+            print(trainer.timers.report(), flush=True)
     pdist.barrier()
     return 0
 
