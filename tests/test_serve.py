@@ -17,7 +17,7 @@ def client(tmp_path_factory):
     cfg = TrainConfig(device="cpu", backend="cpu", batch_size=32,
                       log_interval=0)
     t = Trainer(cfg)
-    x, y = synthetic_mnist(512, seed=1)
+    x, y = synthetic_mnist(2048, seed=1)
     for _ in range(3):
         t.train_epoch(x, y, log=lambda *a: None)
     ck = str(tmp_path_factory.mktemp("serve") / "w.bin")
@@ -46,7 +46,7 @@ def test_predict(client):
     assert len(body["probs"]) == 16 and len(body["probs"][0]) == 10
     # the trained model should beat chance on the structured bands
     acc = sum(int(a == b) for a, b in zip(body["labels"], y.tolist())) / 16
-    assert acc > 0.3
+    assert acc > 0.5, acc
 
 
 def test_predict_validation(client):
