@@ -361,10 +361,9 @@ class Trainer:
         Bg = Bl * self.ctx.world_size
         n = (x.shape[0] // Bg) * Bg  # drop ragged tail like a fixed-shape bench
         total_loss, total_n = 0.0, 0
-        for s in range(0, n, Bg):
-            lo = s + self.ctx.rank * Bl
-            xb, lb = self.stage_batch(x[lo:lo + Bl], labels[lo:lo + Bl])
-            self.step(xb, lb)
+
+        def maybe_log():
+            nonlocal total_loss, total_n
             if self.cfg.log_interval and \
                     self.global_step % self.cfg.log_interval == 0:
                 loss, cnt = self.consume_loss()
@@ -372,6 +371,21 @@ class Trainer:
                 total_n += cnt
                 if self.ctx.is_main and cnt:
                     log(f"step {self.global_step}: error {loss / cnt:e}")
+
+        if self.backend == "hip":
+            # double-buffered async H2D staging on a copy stream
+            from ..data.pipeline import DevicePrefetcher
+            pf = DevicePrefetcher(x, labels, Bl, self.device, self.act_dtype,
+                                  lo=self.ctx.rank * Bl, hi=n, stride=Bg)
+            for xb, lb in pf:
+                self.step(xb, lb)
+                maybe_log()
+        else:
+            for s in range(0, n, Bg):
+                lo = s + self.ctx.rank * Bl
+                xb, lb = self.stage_batch(x[lo:lo + Bl], labels[lo:lo + Bl])
+                self.step(xb, lb)
+                maybe_log()
         loss, cnt = self.consume_loss()
         total_loss += loss
         total_n += cnt

@@ -306,6 +306,21 @@ def test_graph_step_matches_eager(device):
     assert abs(l1 - l2) < 1e-3 * max(1.0, l2)
 
 
+def test_prefetcher_epoch_matches_sync_staging(device):
+    """train_epoch's double-buffered copy-stream pipeline must produce the
+    same trajectory as synchronous staging."""
+    x, y = synthetic_mnist(256, seed=31)
+    cfg = TrainConfig(batch_size=32, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0)
+    t1, t2 = Trainer(cfg), Trainer(cfg)
+    t1.train_epoch(x, y, log=lambda *a: None)      # prefetcher path
+    for s in range(0, 256, 32):                    # sync path
+        t2.step(*t2.stage_batch(x[s:s + 32], y[s:s + 32]))
+    torch.cuda.synchronize()
+    diff = (t1.model.params - t2.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+
+
 def test_native_cli_trainer(device, tmp_path):
     """tools/pcnn_train: the no-Python native driver must train, report in
     the reference's stdout shape, and write a checkpoint the Python side
