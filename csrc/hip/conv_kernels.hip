@@ -84,54 +84,58 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
 // One thread per cols element; kc is the fast axis (coalesced stores, and
 // coalesced loads since ci is the fast axis of NHWC x).
 // ---------------------------------------------------------------------------
+// Row-piece formulation: one thread owns an 8-element span of a kernel row
+// (kc = i*K*Cin + t, t in [0, K*Cin)).  Within one kernel row the source
+// addresses are CONTIGUOUS in NHWC x (x row ih, starting at column ow-P),
+// so the whole span is one 16B load + one 16B store, with scalar clipping
+// only at image borders / row boundaries.  Index decode happens once per
+// thread (the per-element version was 83% VALUBusy on index math).
 template <typename act_t>
 __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
                          int B, int H, int W, int Cin, int K, int P,
                          int KcP) {
-  // 8 consecutive kc per thread (KcP % 8 == 0): one 16B vector store.
-  // Key layout fact: within one kernel row i, consecutive kc = (i*K+j)*Cin
-  // + ci map to CONSECUTIVE NHWC x addresses (base + (iw0*Cin + ci0) + t),
-  // so when the 8-span stays inside row i and inside the image it is ONE
-  // 16B vector load; otherwise fall back to per-element gather.
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long total = ((long long)B * H * W * KcP) / 8;
   if (idx >= total) return;
   const int kc0 = (int)((idx * 8) % KcP);
   const long long m = (idx * 8) / KcP;
   const int Kc = K * K * Cin;
+  const int rowc = K * Cin;  // elements per kernel row
   const int ow = (int)(m % W);
   const long long bh = m / W;
   const int oh = (int)(bh % H);
   const int b = (int)(bh / H);
-  const int rowc = K * Cin;  // elements per kernel row i
 
   act_t out[8];
-  const int ci0 = kc0 % Cin;
-  const int ij0 = kc0 / Cin;
-  const int i0 = ij0 / K;
-  const int j0 = ij0 - i0 * K;
-  const int ih0 = oh + i0 - P;
-  const int iw0 = ow + j0 - P;
-  // last element's column position
-  const int j7 = ((kc0 + 7) / Cin) - i0 * K;
-  // the vector load needs 8-byte source alignment: holds iff Cin % 8 == 0
-  const bool same_row = (kc0 % rowc) + 8 <= rowc;
-  if ((Cin % 8) == 0 && same_row && kc0 + 8 <= Kc && ih0 >= 0 && ih0 < H &&
-      iw0 >= 0 && (ow + j7 - P) < W) {
-    float v8[8];
-    ld8v(x + (((long long)b * H + ih0) * W + iw0) * Cin + ci0, v8);
+  bool fast = false;
+  if (kc0 + 8 <= Kc && (kc0 % rowc) + 8 <= rowc && (Cin % 8) == 0) {
+    const int i = kc0 / rowc;
+    const int ih = oh + i - P;
+    const int t0 = kc0 - i * rowc;       // offset within the kernel row
+    // span covers x row ih, columns iw = ow - P + t/Cin, elements
+    // contiguous from ((b,ih,ow-P))*Cin + t0
+    const int j0 = t0 / Cin;
+    const int j7 = (t0 + 7) / Cin;
+    const int iw0 = ow + j0 - P;
+    const int iw7 = ow + j7 - P;
+    if (ih >= 0 && ih < H && iw0 >= 0 && iw7 < W) {
+      float v8[8];
+      ld8v(x + (((long long)b * H + ih) * W + (ow - P)) * Cin + t0, v8);
 #pragma unroll
-    for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
-  } else {
+      for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
+      fast = true;
+    }
+  }
+  if (!fast) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       const int kc = kc0 + u;
       float val = 0.f;
       if (kc < Kc) {
-        const int ci = kc % Cin;
-        const int ij = kc / Cin;
-        const int i = ij / K;
-        const int j = ij - i * K;
+        const int i = kc / rowc;
+        const int t = kc - i * rowc;
+        const int j = t / Cin;
+        const int ci = t - j * Cin;
         const int ih = oh + i - P;
         const int iw = ow + j - P;
         if (ih >= 0 && ih < H && iw >= 0 && iw < W)

@@ -31,3 +31,29 @@ def test_cli_deepcnn_end_to_end():
                    "--epochs", "1", "--log-interval", "0"])
     assert out.returncode == 0, out.stdout + out.stderr
     assert "Error Rate: " in out.stdout
+
+
+def test_cli_mnist_idx_files(tmp_path):
+    """End-to-end: real IDX files on disk -> --data mnist -> train+test."""
+    import struct
+    import numpy as np
+    rng = np.random.default_rng(0)
+    d = tmp_path / "data"
+    d.mkdir()
+    for name, n, lname in [("train-images.idx3-ubyte", 128,
+                            "train-labels.idx1-ubyte"),
+                           ("t10k-images.idx3-ubyte", 64,
+                            "t10k-labels.idx1-ubyte")]:
+        imgs = rng.integers(0, 256, size=(n, 28, 28), dtype=np.uint8)
+        lbls = rng.integers(0, 10, size=n, dtype=np.uint8)
+        with open(d / name, "wb") as f:
+            f.write(struct.pack(">iiii", 2051, n, 28, 28))
+            f.write(imgs.tobytes())
+        with open(d / lname, "wb") as f:
+            f.write(struct.pack(">ii", 2049, n))
+            f.write(lbls.tobytes())
+    out = run_cli(["--device", "cpu", "--data", "mnist", "--data-dir",
+                   str(d), "--batch-size", "32", "--epochs", "1",
+                   "--log-interval", "0"])
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "Error Rate: " in out.stdout
