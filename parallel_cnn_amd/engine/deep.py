@@ -117,7 +117,6 @@ class DeepTrainer:
         m, w, spec = self.model, self.ws, self.model.spec
         st_h = native.current_stream_handle()
         nstage = len(spec.stages)
-        last = spec.stages[-1]
         self._C.deep_fc_bwd(w.dz, w.pouts[-1], m.view("fc_w"),
                             w.dppre[-1], B, spec.fc_in, spec.n_classes, st_h)
         fs = max(1, min(32, B // 64))

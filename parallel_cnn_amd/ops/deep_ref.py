@@ -8,8 +8,6 @@ batch; update p += dt*scale*g.  No LeNet normalization quirks.
 """
 from __future__ import annotations
 
-from typing import List, Tuple
-
 import torch
 import torch.nn.functional as F
 
