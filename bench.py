@@ -41,6 +41,9 @@ def main() -> int:
                    choices=["lenet5", "deepcnn"])
     p.add_argument("--use-graph", action="store_true",
                    help="capture the step in a hipGraph and replay it")
+    p.add_argument("--overlap-comm", action="store_true",
+                   help="two-bucket DP: overlap the fc/pool grad all-reduce "
+                        "with the conv wgrad")
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
@@ -51,7 +54,8 @@ def main() -> int:
 
     cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
                       device=args.device, log_interval=0, data="synthetic",
-                      wgrad_chunk=args.wgrad_chunk, model=args.model)
+                      wgrad_chunk=args.wgrad_chunk, model=args.model,
+                      overlap_comm=args.overlap_comm)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus

@@ -117,10 +117,28 @@ class Trainer:
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
                                self._loss_mode)
-            self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
-                                    m.grads, B, self.cfg.wgrad_chunk,
-                                    self._wroles, stream)
-            pdist.allreduce_grads(m.grads)
+            if self.cfg.overlap_comm and self.ctx.world_size > 1:
+                # two-bucket overlap (SURVEY §5.8 / north star): the pool+fc
+                # gradient tail all-reduces on the RCCL stream while the
+                # conv wgrad role still computes on the compute stream
+                tail_roles = self._wroles & ~1
+                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                        m.grads, B, self.cfg.wgrad_chunk,
+                                        tail_roles, stream)
+                wk_tail = pdist.allreduce_grads_async(m.grads[S.OFF_S1W:])
+                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                        m.grads, B, self.cfg.wgrad_chunk, 1,
+                                        stream)
+                wk_head = pdist.allreduce_grads_async(m.grads[:S.OFF_S1W])
+                if wk_tail is not None:
+                    wk_tail.wait()
+                if wk_head is not None:
+                    wk_head.wait()
+            else:
+                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                        m.grads, B, self.cfg.wgrad_chunk,
+                                        self._wroles, stream)
+                pdist.allreduce_grads(m.grads)
             self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
                                stream)
         elif self.backend == "cpu":

@@ -87,6 +87,21 @@ def allreduce_scalar(value: float, device=None) -> float:
     return float(t.item())
 
 
+def allreduce_grads_async(grads: torch.Tensor):
+    """Start an async SUM all-reduce (RCCL stream overlaps with compute
+    enqueued afterwards on the current stream); returns a Work handle or
+    None.  gloo with device tensors falls back to a synchronous host
+    bounce (correctness path)."""
+    if not is_distributed():
+        return None
+    if grads.is_cuda and dist.get_backend() == "gloo":
+        host = grads.cpu()
+        dist.all_reduce(host, op=dist.ReduceOp.SUM)
+        grads.copy_(host)
+        return None
+    return dist.all_reduce(grads, op=dist.ReduceOp.SUM, async_op=True)
+
+
 def allreduce_max_scalar(value: float, device=None) -> float:
     if not is_distributed():
         return value

@@ -223,17 +223,19 @@ def test_fwdbwd_wgrad_fp16_acts_close(device):
     assert gdiff < 1e-2 * max(1.0, want[6].abs().max().item()), gdiff
 
 
-def test_dp2_gloo_single_gpu(device):
+@pytest.mark.parametrize("overlap", [False, True])
+def test_dp2_gloo_single_gpu(overlap, device):
     """Two ranks sharing one GPU over gloo: exercises the full distributed
-    engine path (sharding, fused-bucket all-reduce, identical update) on
-    device tensors.  Transport is gloo; the 8-GPU RCCL run uses the same
-    code with backend nccl."""
+    engine path (sharding, fused-bucket all-reduce — single or two-bucket
+    overlapped — identical update) on device tensors.  Transport is gloo;
+    the 8-GPU RCCL run uses the same code with backend nccl."""
     import subprocess
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
          "--master-port", "29617", "bench.py", "--gpus", "2", "--steps",
-         "20", "--warmup", "5", "--batch-size", "16"],
+         "20", "--warmup", "5", "--batch-size", "16"]
+        + (["--overlap-comm"] if overlap else []),
         capture_output=True, text=True, timeout=600,
         env={**__import__("os").environ, "PCNN_DIST_BACKEND": "gloo"})
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
