@@ -30,6 +30,12 @@ int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
                           float* dz1, const int* labels, float* loss_accum,
                           int* correct, int B, int act_is_bf16, int mode,
                           int pool_mode, int loss_mode, void* stream);
+int pcnn_launch_fwdbwd_ex2(const void* x, const float* params, void* a1,
+                           void* a2, float* y, float* dz, float* dz2,
+                           float* dz1, const int* labels, float* loss_accum,
+                           int* correct, int B, int act_is_bf16, int mode,
+                           int pool_mode, int loss_mode, float* grads,
+                           int wgrad_fuse, void* stream);
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
                       const float* dz, const float* dz2, const float* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
@@ -98,11 +104,12 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
                 at::Tensor y, at::Tensor dz, at::Tensor dz2, at::Tensor dz1,
                 at::Tensor labels, at::Tensor loss_accum,
                 at::Tensor correct_accum, int64_t B, int64_t mode,
-                int64_t stream, int64_t pool_mode, int64_t loss_mode) {
+                int64_t stream, int64_t pool_mode, int64_t loss_mode,
+                at::Tensor grads, int64_t wgrad_fuse) {
   TORCH_CHECK(x.is_cuda() && params.is_cuda(), "expected device tensors");
   TORCH_CHECK(labels.scalar_type() == at::kInt, "labels must be int32");
   int f = act_flag(x);
-  check_hip(pcnn_launch_fwdbwd_ex(
+  check_hip(pcnn_launch_fwdbwd_ex2(
                 x.data_ptr(), params.data_ptr<float>(), a1.data_ptr(),
                 a2.data_ptr(), y.numel() ? y.data_ptr<float>() : nullptr,
                 dz.numel() ? dz.data_ptr<float>() : nullptr,
@@ -112,7 +119,8 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
                 loss_accum.numel() ? loss_accum.data_ptr<float>() : nullptr,
                 correct_accum.numel() ? correct_accum.data_ptr<int>() : nullptr,
                 (int)B, f, (int)mode, (int)pool_mode, (int)loss_mode,
-                (void*)stream),
+                grads.numel() ? grads.data_ptr<float>() : nullptr,
+                (int)wgrad_fuse, (void*)stream),
             "fwdbwd");
 }
 
@@ -161,7 +169,8 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
                      int64_t B, int64_t steps, int64_t chunk_imgs,
                      double step_scale, int64_t stream, int64_t pool_mode,
                      int64_t loss_mode) {
-  const int wroles = pool_mode == 1 ? 5 : 7;  // max pool: no pool wgrad
+  // fused conv/pool wgrad inside fwdbwd; kernel B covers the fc role only
+  const int wroles = 4;
   TORCH_CHECK(x_pool.is_cuda() && x_pool.dim() == 2, "x_pool [P*B, 784]");
   TORCH_CHECK(labels_pool.scalar_type() == at::kInt, "labels must be int32");
   const int64_t pool_rows = x_pool.size(0);
@@ -178,13 +187,14 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
     const int64_t i = st % P;
     const void* xb = xp + (size_t)i * B * pcnn::IN_PIX * esz;
     const int* lb = lp + i * B;
-    check_hip(pcnn_launch_fwdbwd_ex(xb, pp, a1.data_ptr(), a2.data_ptr(),
-                                    y.data_ptr<float>(), dz.data_ptr<float>(),
-                                    dz2.data_ptr<float>(),
-                                    dz1.data_ptr<float>(), lb,
-                                    loss_accum.data_ptr<float>(), nullptr,
-                                    (int)B, f, 0, (int)pool_mode,
-                                    (int)loss_mode, s),
+    check_hip(pcnn_launch_fwdbwd_ex2(xb, pp, a1.data_ptr(), a2.data_ptr(),
+                                     y.data_ptr<float>(),
+                                     dz.data_ptr<float>(),
+                                     dz2.data_ptr<float>(),
+                                     dz1.data_ptr<float>(), lb,
+                                     loss_accum.data_ptr<float>(), nullptr,
+                                     (int)B, f, 0, (int)pool_mode,
+                                     (int)loss_mode, gp, 1, s),
               "train_steps/fwdbwd");
     check_hip(pcnn_launch_wgrad_ex(xb, a1.data_ptr(), a2.data_ptr(),
                                    dz.data_ptr<float>(), dz2.data_ptr<float>(),
@@ -346,7 +356,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dz2"), py::arg("dz1"), py::arg("labels"),
         py::arg("loss_accum"), py::arg("correct_accum"), py::arg("B"),
         py::arg("mode"), py::arg("stream"), py::arg("pool_mode") = 0,
-        py::arg("loss_mode") = 0);
+        py::arg("loss_mode") = 0, py::arg("grads") = at::Tensor(),
+        py::arg("wgrad_fuse") = 0);
   m.def("hip_wgrad", &hip_wgrad);
   m.def("hip_wgrad_roles", &hip_wgrad_roles);
   m.def("hip_update", &hip_update);

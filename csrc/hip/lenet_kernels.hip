@@ -114,6 +114,10 @@ struct FwdLds {
   float ys[FC_OUT];     // logits (post-sigmoid)
   float dzs[FC_OUT];    // residual gradient
   float sq[FC_OUT];     // per-class squared error
+  // fused weight-grad accumulators (wgrad_fuse mode): conv1 per-channel
+  // 5x5 + bias, pool 4x4 + bias
+  float gw[C1_CH][C1_K * C1_K + 1];
+  float gs1[S1_WSZ + 1];
 };
 
 // Fused forward + backward-data, one 256-thread workgroup per image.
@@ -132,7 +136,8 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     float* __restrict__ dzg, float* __restrict__ dz2g,
     float* __restrict__ dz1g, const int* __restrict__ labels,
     float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B,
-    int pool_mode, int loss_mode) {
+    int pool_mode, int loss_mode, float* __restrict__ grads,
+    int wgrad_fuse) {
   __shared__ FwdLds L;
   const int b = blockIdx.x;
   if (b >= B) return;
@@ -140,6 +145,12 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 
   // ---- phase 0: stage conv/pool parameters and the input image ----
   if (tid < OFF_FW) L.ps[tid] = params[tid];
+  if (MODE == MODE_TRAIN && wgrad_fuse) {
+    if (tid < C1_CH * (C1_K * C1_K + 1))
+      L.gw[tid / (C1_K * C1_K + 1)][tid % (C1_K * C1_K + 1)] = 0.f;
+    else if (tid < C1_CH * (C1_K * C1_K + 1) + S1_WSZ + 1)
+      L.gs1[tid - C1_CH * (C1_K * C1_K + 1)] = 0.f;
+  }
   const act_t* xb = x + (size_t)b * IN_PIX;
   for (int i = tid; i < IN_PIX; i += 256) L.xs[i] = ldf(xb + i);
   __syncthreads();
@@ -292,9 +303,9 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
           best = t;
         }
     }
+    float dz1v[S1_K * S1_K];
 #pragma unroll
     for (int i = 0; i < S1_K; ++i) {
-      float row[S1_K];
 #pragma unroll
       for (int j = 0; j < S1_K; ++j) {
         const float av = a1v[i * S1_K + j];
@@ -302,11 +313,43 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
             pool_mode == 1
                 ? (i * S1_K + j == best ? d2 : 0.f)
                 : d2 * L.ps[OFF_S1W + i * S1_K + j];
-        row[j] = dd * av * (1.0f - av);
+        dz1v[i * S1_K + j] = dd * av * (1.0f - av);
       }
       *reinterpret_cast<float4*>(
           dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
-          pc * S1_K) = *reinterpret_cast<const float4*>(row);
+          pc * S1_K) = *reinterpret_cast<const float4*>(&dz1v[i * S1_K]);
+    }
+    if (wgrad_fuse) {
+      // conv1 wgrad: this cell's 16 dz1 x its 8x8 input window (from LDS);
+      // per-thread register accumulation, LDS-atomic combine per channel,
+      // one global atomic per weight per block (in the final phase below).
+      float cacc[C1_K * C1_K];
+#pragma unroll
+      for (int w2 = 0; w2 < C1_K * C1_K; ++w2) cacc[w2] = 0.f;
+      float csum = 0.f;
+#pragma unroll
+      for (int t = 0; t < S1_K * S1_K; ++t) {
+        const float d = dz1v[t];
+        csum += d;
+        const int r = pr * S1_K + t / S1_K;
+        const int c = pc * S1_K + (t & 3);
+#pragma unroll
+        for (int u = 0; u < C1_K; ++u)
+#pragma unroll
+          for (int v = 0; v < C1_K; ++v)
+            cacc[u * C1_K + v] += d * L.xs[(r + u) * IN_W + (c + v)];
+      }
+#pragma unroll
+      for (int w2 = 0; w2 < C1_K * C1_K; ++w2)
+        atomicAdd(&L.gw[o][w2], cacc[w2]);
+      atomicAdd(&L.gw[o][C1_K * C1_K], csum);
+      if (pool_mode == 0) {
+        // pool wgrad: dz2(own cell) x own a1 activations
+#pragma unroll
+        for (int t = 0; t < S1_K * S1_K; ++t)
+          atomicAdd(&L.gs1[t], d2 * a1v[t]);
+        atomicAdd(&L.gs1[S1_WSZ], d2);
+      }
     }
   } else if (tid == S1_OUT && loss_accum != nullptr) {
     if (loss_mode == 1) {
@@ -316,6 +359,27 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
 #pragma unroll
       for (int k = 0; k < FC_OUT; ++k) ssum += L.sq[k];
       unsafeAtomicAdd(loss_accum, sqrtf(ssum));
+    }
+  }
+  if (MODE == MODE_TRAIN && wgrad_fuse) {
+    __syncthreads();
+    // conv grads carry the reference 1/(24*24) factor; pool bias /216
+    constexpr float inv_pix = 1.0f / (float)C1_PIX;
+    if (tid < C1_CH * (C1_K * C1_K + 1)) {
+      const int o2 = tid / (C1_K * C1_K + 1);
+      const int w2 = tid % (C1_K * C1_K + 1);
+      const float v = L.gw[o2][w2] * inv_pix;
+      if (w2 < C1_K * C1_K)
+        unsafeAtomicAdd(&grads[OFF_C1W + o2 * C1_K * C1_K + w2], v);
+      else
+        unsafeAtomicAdd(&grads[OFF_C1B + o2], v);
+    } else if (pool_mode == 0 &&
+               tid < C1_CH * (C1_K * C1_K + 1) + S1_WSZ + 1) {
+      const int w2 = tid - C1_CH * (C1_K * C1_K + 1);
+      if (w2 < S1_WSZ)
+        unsafeAtomicAdd(&grads[OFF_S1W + w2], L.gs1[w2]);
+      else
+        unsafeAtomicAdd(&grads[OFF_S1B], L.gs1[S1_WSZ] / (float)S1_OUT);
     }
   }
 }
@@ -568,23 +632,23 @@ int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
                        float* y, float* dz, float* dz2, float* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int pool_mode, int loss_mode,
-                       hipStream_t stream) {
+                       float* grads, int wgrad_fuse, hipStream_t stream) {
   dim3 grid(B), block(256);
   if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_fwdbwd<bf16, MODE>), grid, block, 0, stream,
                        (const bf16*)x, params, (bf16*)a1, (bf16*)a2, y, dz,
                        dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode);
+                       loss_mode, grads, wgrad_fuse);
   } else if (act_is_bf16 == 2) {
     hipLaunchKernelGGL((k_fwdbwd<fp16, MODE>), grid, block, 0, stream,
                        (const fp16*)x, params, (fp16*)a1, (fp16*)a2, y, dz,
                        dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode);
+                       loss_mode, grads, wgrad_fuse);
   } else {
     hipLaunchKernelGGL((k_fwdbwd<float, MODE>), grid, block, 0, stream,
                        (const float*)x, params, (float*)a1, (float*)a2, y, dz,
                        dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode);
+                       loss_mode, grads, wgrad_fuse);
   }
   return (int)hipGetLastError();
 }
@@ -592,29 +656,40 @@ int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
 
 extern "C" {
 
-int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
-                          void* a2, float* y, float* dz, float* dz2,
-                          float* dz1, const int* labels, float* loss_accum,
-                          int* correct, int B, int act_is_bf16, int mode,
-                          int pool_mode, int loss_mode, void* stream) {
+int pcnn_launch_fwdbwd_ex2(const void* x, const float* params, void* a1,
+                           void* a2, float* y, float* dz, float* dz2,
+                           float* dz1, const int* labels, float* loss_accum,
+                           int* correct, int B, int act_is_bf16, int mode,
+                           int pool_mode, int loss_mode, float* grads,
+                           int wgrad_fuse, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   switch (mode) {
     case MODE_TRAIN:
       return launch_fwdbwd_mode<MODE_TRAIN>(x, params, a1, a2, y, dz, dz2,
                                             dz1, labels, loss_accum, correct,
                                             B, act_is_bf16, pool_mode,
-                                            loss_mode, s);
+                                            loss_mode, grads, wgrad_fuse, s);
     case MODE_EVAL:
       return launch_fwdbwd_mode<MODE_EVAL>(x, params, a1, a2, y, dz, dz2, dz1,
                                            labels, loss_accum, correct, B,
                                            act_is_bf16, pool_mode, loss_mode,
-                                           s);
+                                           grads, wgrad_fuse, s);
     default:
       return launch_fwdbwd_mode<MODE_INFER>(x, params, a1, a2, y, dz, dz2,
                                             dz1, labels, loss_accum, correct,
                                             B, act_is_bf16, pool_mode,
-                                            loss_mode, s);
+                                            loss_mode, grads, wgrad_fuse, s);
   }
+}
+
+int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
+                          void* a2, float* y, float* dz, float* dz2,
+                          float* dz1, const int* labels, float* loss_accum,
+                          int* correct, int B, int act_is_bf16, int mode,
+                          int pool_mode, int loss_mode, void* stream) {
+  return pcnn_launch_fwdbwd_ex2(x, params, a1, a2, y, dz, dz2, dz1, labels,
+                                loss_accum, correct, B, act_is_bf16, mode,
+                                pool_mode, loss_mode, nullptr, 0, stream);
 }
 
 int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
@@ -636,9 +711,11 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   int GC = chunk_imgs > 0 ? chunk_imgs : (int)(4.0f * __builtin_cbrtf((float)B) + 0.5f);
   if (GC < 2) GC = 2;
   if (GC > 256) GC = 256;
+  if (roles == 4) GC = 0;  // fc-only launch (fused-wgrad mode)
   int GS = (B * S1_OUT + 256 * 48 - 1) / (256 * 48);
   if (GS < 2) GS = 2;
   if (GS > 96) GS = 96;
+  if (roles == 4) GS = 0;
   int FS = B / 64;
   if (FS < 1) FS = 1;
   if (FS > 32) FS = 32;

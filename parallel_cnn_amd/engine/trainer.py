@@ -76,8 +76,9 @@ class Trainer:
         self.timers: Optional[PhaseTimers] = None  # set by enable_profiling
         self._pool_mode = 1 if cfg.pool == "max" else 0
         self._loss_mode = 1 if cfg.loss == "softmax_ce" else 0
-        # max pool has no parameters: skip the pool wgrad role
-        self._wroles = 5 if self._pool_mode == 1 else 7
+        # conv/pool weight-grads are fused into the fwdbwd kernel; the
+        # wgrad kernel covers the fc role only
+        self._wroles = 4
 
     # ------------------------------------------------------------------ util
     def _update_scale(self, local_batch: int) -> float:
@@ -116,24 +117,21 @@ class Trainer:
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
-                               self._loss_mode)
+                               self._loss_mode, m.grads, 1)
             if self.cfg.overlap_comm and self.ctx.world_size > 1:
-                # two-bucket overlap (SURVEY §5.8 / north star): the pool+fc
-                # gradient tail all-reduces on the RCCL stream while the
-                # conv wgrad role still computes on the compute stream
-                tail_roles = self._wroles & ~1
+                # two-bucket overlap (SURVEY §5.8 / north star): the
+                # conv/pool gradient head (ready — fused into fwdbwd)
+                # all-reduces on the RCCL stream while the fc wgrad still
+                # computes on the compute stream
+                wk_head = pdist.allreduce_grads_async(m.grads[:S.OFF_FW])
                 self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                         m.grads, B, self.cfg.wgrad_chunk,
-                                        tail_roles, stream)
-                wk_tail = pdist.allreduce_grads_async(m.grads[S.OFF_S1W:])
-                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
-                                        m.grads, B, self.cfg.wgrad_chunk, 1,
-                                        stream)
-                wk_head = pdist.allreduce_grads_async(m.grads[:S.OFF_S1W])
-                if wk_tail is not None:
-                    wk_tail.wait()
+                                        4, stream)
+                wk_tail = pdist.allreduce_grads_async(m.grads[S.OFF_FW:])
                 if wk_head is not None:
                     wk_head.wait()
+                if wk_tail is not None:
+                    wk_tail.wait()
             else:
                 self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                         m.grads, B, self.cfg.wgrad_chunk,
@@ -179,11 +177,11 @@ class Trainer:
                     self.timers = t
             return
         stream = self._sh
-        with t.phase("fwd+bwd-data"):
+        with t.phase("fwd+bwd-data+conv/pool-wgrad"):
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
-                               self._loss_mode)
+                               self._loss_mode, m.grads, 1)
         with t.phase("weight-grad"):
             self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                     m.grads, B, self.cfg.wgrad_chunk,
@@ -236,7 +234,7 @@ class Trainer:
         self._C.hip_fwdbwd(self._gx, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                            w.dz1, self._gl, w.loss_accum, w.correct_accum,
                            B, MODE_TRAIN, stream, self._pool_mode,
-                           self._loss_mode)
+                           self._loss_mode, m.grads, 1)
         self._C.hip_wgrad_roles(self._gx, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                 m.grads, B, self.cfg.wgrad_chunk,
                                 self._wroles, stream)

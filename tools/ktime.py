@@ -48,9 +48,16 @@ def main():
 
         fwd = lambda: _C.hip_fwdbwd(x, params, a1, a2, y, dz, dz2, dz1,
                                     labels, loss, corr, B, 0, st)
+        fwd_fused = lambda: _C.hip_fwdbwd(x, params, a1, a2, y, dz, dz2,
+                                          dz1, labels, loss, corr, B, 0, st,
+                                          0, 0, grads, 1)
         fwd()
         print(f"== B={B}")
         print(f"  fwdbwd          : {time_loop(fwd):8.2f} us")
+        print(f"  fwdbwd+cs-wgrad : {time_loop(fwd_fused):8.2f} us")
+        t = time_loop(lambda: _C.hip_wgrad_roles(
+            x, a1, a2, dz, dz2, dz1, grads, B, 0, 4, st))
+        print(f"  wgrad fc-only   : {t:8.2f} us")
         for roles, name in [(7, "wgrad all"), (1, "wgrad c1"),
                             (2, "wgrad s1"), (4, "wgrad fc")]:
             t = time_loop(lambda: _C.hip_wgrad_roles(
