@@ -168,9 +168,11 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
                      at::Tensor dz2, at::Tensor dz1, at::Tensor loss_accum,
                      int64_t B, int64_t steps, int64_t chunk_imgs,
                      double step_scale, int64_t stream, int64_t pool_mode,
-                     int64_t loss_mode) {
-  // fused conv/pool wgrad inside fwdbwd; kernel B covers the fc role only
-  const int wroles = 4;
+                     int64_t loss_mode, int64_t wgrad_fuse) {
+  // wgrad_fuse=1: conv/pool grads accumulate inside the fwdbwd kernel and
+  // kernel B covers only the fc role.  Measured slower at bs=64 (the LDS
+  // atomic combine serializes) — default off.
+  const int wroles = wgrad_fuse ? 4 : (pool_mode == 1 ? 5 : 7);
   TORCH_CHECK(x_pool.is_cuda() && x_pool.dim() == 2, "x_pool [P*B, 784]");
   TORCH_CHECK(labels_pool.scalar_type() == at::kInt, "labels must be int32");
   const int64_t pool_rows = x_pool.size(0);
@@ -194,7 +196,7 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
                                      dz1.data_ptr<float>(), lb,
                                      loss_accum.data_ptr<float>(), nullptr,
                                      (int)B, f, 0, (int)pool_mode,
-                                     (int)loss_mode, gp, 1, s),
+                                     (int)loss_mode, gp, (int)wgrad_fuse, s),
               "train_steps/fwdbwd");
     check_hip(pcnn_launch_wgrad_ex(xb, a1.data_ptr(), a2.data_ptr(),
                                    dz.data_ptr<float>(), dz2.data_ptr<float>(),
@@ -356,7 +358,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dz2"), py::arg("dz1"), py::arg("labels"),
         py::arg("loss_accum"), py::arg("correct_accum"), py::arg("B"),
         py::arg("mode"), py::arg("stream"), py::arg("pool_mode") = 0,
-        py::arg("loss_mode") = 0, py::arg("grads") = at::Tensor(),
+        py::arg("loss_mode") = 0, py::arg("grads") = at::empty({0}),
         py::arg("wgrad_fuse") = 0);
   m.def("hip_wgrad", &hip_wgrad);
   m.def("hip_wgrad_roles", &hip_wgrad_roles);
@@ -367,7 +369,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dz2"), py::arg("dz1"), py::arg("loss_accum"), py::arg("B"),
         py::arg("steps"), py::arg("chunk_imgs"), py::arg("step_scale"),
         py::arg("stream"), py::arg("pool_mode") = 0,
-        py::arg("loss_mode") = 0);
+        py::arg("loss_mode") = 0, py::arg("wgrad_fuse") = 0);
   m.def("deep_im2col", &deep_im2col);
   m.def("deep_gemm", &deep_gemm);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm);

@@ -41,8 +41,10 @@ class TrainConfig:
     backend: str = "auto"            # auto | hip | cpu | torchref
     device: str = "auto"             # auto | cuda | cpu
     wgrad_chunk: int = 0             # conv1 wgrad slices/channel (0 = auto)
-    overlap_comm: bool = False       # two-bucket DP: all-reduce the fc/pool
-                                     # grads while the conv wgrad still runs
+    overlap_comm: bool = False       # two-bucket DP: all-reduce the ready
+                                     # bucket while the other wgrad runs
+    fuse_wgrad: bool = False         # experimental: conv/pool wgrad inside
+                                     # the fwdbwd kernel (measured slower)
 
     # io / observability
     log_interval: int = 100          # steps between loss readouts

@@ -76,9 +76,13 @@ class Trainer:
         self.timers: Optional[PhaseTimers] = None  # set by enable_profiling
         self._pool_mode = 1 if cfg.pool == "max" else 0
         self._loss_mode = 1 if cfg.loss == "softmax_ce" else 0
-        # conv/pool weight-grads are fused into the fwdbwd kernel; the
-        # wgrad kernel covers the fc role only
-        self._wroles = 4
+        # fuse_wgrad: conv/pool grads accumulate inside fwdbwd and the
+        # wgrad kernel covers only the fc role (experimental — measured
+        # slower at bs=64: the LDS-atomic combine serializes); default:
+        # full wgrad kernel, minus the pool role under max pooling
+        self._fuse = 1 if cfg.fuse_wgrad else 0
+        self._wroles = 4 if self._fuse else (5 if self._pool_mode == 1
+                                             else 7)
 
     # ------------------------------------------------------------------ util
     def _update_scale(self, local_batch: int) -> float:
@@ -117,17 +121,32 @@ class Trainer:
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
-                               self._loss_mode, m.grads, 1)
+                               self._loss_mode, m.grads, self._fuse)
             if self.cfg.overlap_comm and self.ctx.world_size > 1:
-                # two-bucket overlap (SURVEY §5.8 / north star): the
-                # conv/pool gradient head (ready — fused into fwdbwd)
-                # all-reduces on the RCCL stream while the fc wgrad still
-                # computes on the compute stream
-                wk_head = pdist.allreduce_grads_async(m.grads[:S.OFF_FW])
-                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
-                                        m.grads, B, self.cfg.wgrad_chunk,
-                                        4, stream)
-                wk_tail = pdist.allreduce_grads_async(m.grads[S.OFF_FW:])
+                # two-bucket overlap (SURVEY §5.8 / north star): the first
+                # bucket all-reduces on the RCCL stream while the rest of
+                # the wgrad still computes on the compute stream
+                if self._fuse:
+                    wk_head = pdist.allreduce_grads_async(
+                        m.grads[:S.OFF_FW])
+                    self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2,
+                                            w.dz1, m.grads, B,
+                                            self.cfg.wgrad_chunk, 4, stream)
+                    wk_tail = pdist.allreduce_grads_async(
+                        m.grads[S.OFF_FW:])
+                else:
+                    tail_roles = self._wroles & ~1
+                    self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2,
+                                            w.dz1, m.grads, B,
+                                            self.cfg.wgrad_chunk,
+                                            tail_roles, stream)
+                    wk_tail = pdist.allreduce_grads_async(
+                        m.grads[S.OFF_S1W:])
+                    self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2,
+                                            w.dz1, m.grads, B,
+                                            self.cfg.wgrad_chunk, 1, stream)
+                    wk_head = pdist.allreduce_grads_async(
+                        m.grads[:S.OFF_S1W])
                 if wk_head is not None:
                     wk_head.wait()
                 if wk_tail is not None:
@@ -177,11 +196,11 @@ class Trainer:
                     self.timers = t
             return
         stream = self._sh
-        with t.phase("fwd+bwd-data+conv/pool-wgrad"):
+        with t.phase("fwd+bwd-data"):
             self._C.hip_fwdbwd(x, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
-                               self._loss_mode, m.grads, 1)
+                               self._loss_mode, m.grads, self._fuse)
         with t.phase("weight-grad"):
             self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                     m.grads, B, self.cfg.wgrad_chunk,
@@ -234,7 +253,7 @@ class Trainer:
         self._C.hip_fwdbwd(self._gx, m.params, w.a1, w.a2, w.y, w.dz, w.dz2,
                            w.dz1, self._gl, w.loss_accum, w.correct_accum,
                            B, MODE_TRAIN, stream, self._pool_mode,
-                           self._loss_mode, m.grads, 1)
+                           self._loss_mode, m.grads, self._fuse)
         self._C.hip_wgrad_roles(self._gx, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                 m.grads, B, self.cfg.wgrad_chunk,
                                 self._wroles, stream)
@@ -271,7 +290,7 @@ class Trainer:
                 w.a1, w.a2, w.y, w.dz, w.dz2, w.dz1, w.loss_accum, B, steps,
                 self.cfg.wgrad_chunk, self.cfg.dt * self._update_scale(B),
                 native.current_stream_handle(), self._pool_mode,
-                self._loss_mode)
+                self._loss_mode, self._fuse)
             self._samples_seen += B * steps
             self.global_step += steps
         else:
