@@ -56,7 +56,8 @@ def main(argv=None) -> int:
     if profile and hasattr(trainer, "enable_profiling"):
         trainer.enable_profiling()
     if cfg.ckpt_load:
-        trainer.model.load(cfg.ckpt_load)
+        from .utils.checkpoint import load_checkpoint
+        load_checkpoint(trainer, cfg.ckpt_load)
 
     xtr, ytr, xte, yte = load_datasets(cfg)
 
@@ -77,7 +78,8 @@ def main(argv=None) -> int:
         print(f"\n Time - {total * 1e3:f} ms", flush=True)
 
     if cfg.ckpt_save and ctx.is_main:
-        trainer.model.save(cfg.ckpt_save)
+        from .utils.checkpoint import save_checkpoint
+        save_checkpoint(trainer, cfg.ckpt_save)
         print(f"saved checkpoint: {cfg.ckpt_save}", flush=True)
 
     err_rate = trainer.evaluate(xte, yte)

@@ -47,3 +47,44 @@ def test_checkpoint_wrong_size(tmp_path):
     np.zeros(10, dtype="<f4").tofile(path)
     with pytest.raises(ValueError, match="expected"):
         LeNet5().load(path)
+
+
+def test_checkpoint_sidecar_roundtrip(tmp_path):
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    from parallel_cnn_amd.data.mnist import synthetic_mnist
+    cfg = TrainConfig(device="cpu", backend="cpu", batch_size=8,
+                      log_interval=0)
+    t = Trainer(cfg)
+    x, y = synthetic_mnist(16, seed=1)
+    t.step(*t.stage_batch(x[:8], y[:8]))
+    t.step(*t.stage_batch(x[8:], y[8:]))
+    path = str(tmp_path / "ck.bin")
+    save_checkpoint(t, path)
+    t2 = Trainer(cfg)
+    meta = load_checkpoint(t2, path)
+    assert torch.equal(t.model.params, t2.model.params)
+    assert t2.global_step == 2
+    assert meta["model"] == "lenet5"
+
+
+def test_checkpoint_sidecar_wrong_model(tmp_path):
+    import json
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    cfg = TrainConfig(device="cpu", backend="cpu", batch_size=8,
+                      log_interval=0)
+    t = Trainer(cfg)
+    path = str(tmp_path / "ck.bin")
+    save_checkpoint(t, path)
+    with open(path + ".meta.json") as f:
+        meta = json.load(f)
+    meta["n_params"] = 999
+    with open(path + ".meta.json", "w") as f:
+        json.dump(meta, f)
+    with pytest.raises(ValueError, match="999"):
+        load_checkpoint(Trainer(cfg), path)
