@@ -52,6 +52,12 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    void* C, long long M, int K, int N, int ldA, int ldC,
                    int b_kxn, int epilogue, int actf, void* stream);
+int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
+                      const float* bias, void* C, long long M, int K, int N,
+                      int ldA, int ldC, int b_kxn, int epilogue, int actf,
+                      void* stream);
+int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
+                      void* stream);
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                          long long M, int KcP, int N, int MS, int actf,
                          void* stream);
@@ -222,13 +228,23 @@ void deep_im2col(at::Tensor x, at::Tensor cols, int64_t B, int64_t H,
 
 void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                int64_t M, int64_t K, int64_t N, int64_t ldA, int64_t ldC,
-               int64_t b_kxn, int64_t epilogue, int64_t stream) {
-  check_hip(pcnn_deep_gemm(A.data_ptr(), Bsrc.data_ptr<float>(),
-                           bias.numel() ? bias.data_ptr<float>() : nullptr,
-                           C.data_ptr(), M, (int)K, (int)N, (int)ldA,
-                           (int)ldC, (int)b_kxn, (int)epilogue, act_flag(A),
-                           (void*)stream),
+               int64_t b_kxn, int64_t epilogue, int64_t stream,
+               at::Tensor Bpre) {
+  check_hip(pcnn_deep_gemm_ex(
+                A.data_ptr(), Bsrc.data_ptr<float>(),
+                Bpre.numel() ? Bpre.data_ptr() : nullptr,
+                bias.numel() ? bias.data_ptr<float>() : nullptr,
+                C.data_ptr(), M, (int)K, (int)N, (int)ldA, (int)ldC,
+                (int)b_kxn, (int)epilogue, act_flag(A), (void*)stream),
             "deep_gemm");
+}
+
+void deep_cast_wt(at::Tensor W, at::Tensor out, at::Tensor outT, int64_t R,
+                  int64_t C, int64_t stream) {
+  check_hip(pcnn_deep_cast_wt(W.data_ptr<float>(), out.data_ptr(),
+                              outT.data_ptr(), (int)R, (int)C,
+                              (void*)stream),
+            "deep_cast_wt");
 }
 
 void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
@@ -371,7 +387,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stream"), py::arg("pool_mode") = 0,
         py::arg("loss_mode") = 0, py::arg("wgrad_fuse") = 0);
   m.def("deep_im2col", &deep_im2col);
-  m.def("deep_gemm", &deep_gemm);
+  m.def("deep_gemm", &deep_gemm, py::arg("A"), py::arg("Bsrc"),
+        py::arg("bias"), py::arg("C"), py::arg("M"), py::arg("K"),
+        py::arg("N"), py::arg("ldA"), py::arg("ldC"), py::arg("b_kxn"),
+        py::arg("epilogue"), py::arg("stream"),
+        py::arg("Bpre") = at::empty({0}));
+  m.def("deep_cast_wt", &deep_cast_wt);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm);
   m.def("deep_colsum", &deep_colsum);
   m.def("deep_col2im_sigbwd", &deep_col2im_sigbwd);

@@ -181,11 +181,15 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 //                 false: Bsrc is [N][K] (staged directly).
 //   epilogue: 0 = plain store, 1 = bias + sigmoid.
 //   C: act_t [M][ldC].
+// Bpre (optional): pre-cast bf16 B in [N][K] row-per-output-column layout;
+// when non-null it replaces Bsrc/b_kxn and stages with two 16B copies per
+// thread per tile.
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_gemm(
     const act_t* __restrict__ A, const float* __restrict__ Bsrc,
-    const float* __restrict__ bias, act_t* __restrict__ C, long long M,
-    int K, int N, int ldA, int ldC, int b_kxn, int epilogue) {
+    const __bf16* __restrict__ Bpre, const float* __restrict__ bias,
+    act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
+    int b_kxn, int epilogue) {
   __shared__ GemmLds L;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -218,7 +222,21 @@ __global__ __launch_bounds__(256) void k_gemm(
       for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
       if (m_a < M && (kt + kk) < K) ld8v(A + m_a * ldA + kt + kk, ra[h]);
     }
-    if (b_kxn) {
+    if (Bpre != nullptr) {
+      // bf16 [N][K] rows: two vector loads per thread
+      const int n = tid >> 2;
+      const bool ok = (n0 + n) < N;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int kk = kq + h * 8;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) rb[h][u] = 0.f;
+        if (ok && (kt + kk) < K)
+          ld8v(reinterpret_cast<const bf16*>(
+                   Bpre + (long long)(n0 + n) * K + kt + kk),
+               rb[h]);
+      }
+    } else if (b_kxn) {
       // Bsrc[K][N]: thread owns row k = tid>>2, 16 n (transposed write)
       const int k = tid >> 2;
       const int nq = (tid & 3) * 16;
@@ -252,7 +270,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 #pragma unroll
       for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)ra[h][u];
     }
-    if (b_kxn) {
+    if (Bpre == nullptr && b_kxn) {
       const int k = tid >> 2;
       const int nq = (tid & 3) * 16;
 #pragma unroll
@@ -746,6 +764,21 @@ __global__ __launch_bounds__(256) void k_fc_wgrad(
   }
 }
 
+// Weight pre-cast: fp32 W[R][C] -> bf16 copy [R][C] and bf16 transpose
+// [C][R], refreshed once per step so GEMM B-staging is plain 16B bf16 row
+// copies instead of per-tile fp32 gather+convert(+transpose).
+__global__ void k_cast_wt(const float* __restrict__ W,
+                          __bf16* __restrict__ out,
+                          __bf16* __restrict__ outT, int R, int C) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= (long long)R * C) return;
+  const int r = (int)(i / C);
+  const int c = (int)(i - (long long)r * C);
+  const __bf16 v = (__bf16)W[i];
+  out[i] = v;
+  outT[(long long)c * R + r] = v;
+}
+
 // Generic SGD apply + zero:  p += step*g; g = 0  over n params.
 __global__ void k_update_n(float* __restrict__ params,
                            float* __restrict__ grads, long long n,
@@ -809,16 +842,34 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
-                   void* C, long long M, int K, int N, int ldA, int ldC,
-                   int b_kxn, int epilogue, int actf, void* stream) {
+int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
+                      const float* bias, void* C, long long M, int K, int N,
+                      int ldA, int ldC, int b_kxn, int epilogue, int actf,
+                      void* stream) {
   const int ntiles = (N + BN - 1) / BN;
   const long long mtiles = (M + BM - 1) / BM;
   dim3 grid((unsigned)(mtiles * ntiles)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t>), grid, block, 0,
                                     (hipStream_t)stream, (const act_t*)A,
-                                    Bsrc, bias, (act_t*)C, M, K, N, ldA, ldC,
+                                    Bsrc, (const __bf16*)Bpre, bias,
+                                    (act_t*)C, M, K, N, ldA, ldC,
                                     b_kxn, epilogue));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
+                   void* C, long long M, int K, int N, int ldA, int ldC,
+                   int b_kxn, int epilogue, int actf, void* stream) {
+  return pcnn_deep_gemm_ex(A, Bsrc, nullptr, bias, C, M, K, N, ldA, ldC,
+                           b_kxn, epilogue, actf, stream);
+}
+
+int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
+                      void* stream) {
+  const long long total = (long long)R * C;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  hipLaunchKernelGGL(k_cast_wt, grid, block, 0, (hipStream_t)stream, W,
+                     (__bf16*)out, (__bf16*)outT, R, C);
   return (int)hipGetLastError();
 }
 

@@ -44,6 +44,12 @@ class DeepWorkspace:
                                           device=device))
             self.dppre.append(torch.empty(mo, st.cout, dtype=act_dtype,
                                           device=device))
+        # pre-cast bf16 weight copies ([KcP][Cout] and transposed), refreshed
+        # after each update so GEMM B-staging is plain bf16 row copies
+        self.wbf = [torch.empty(st.kcp * st.cout, dtype=torch.bfloat16,
+                                device=device) for st in spec.stages]
+        self.wbfT = [torch.empty(st.cout * st.kcp, dtype=torch.bfloat16,
+                                 device=device) for st in spec.stages]
         self.y = torch.empty(B, spec.n_classes, dtype=torch.float32,
                              device=device)
         self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
@@ -92,10 +98,18 @@ class DeepTrainer:
         return x.to(torch.float32), labels.to(torch.int64)
 
     # ------------------------------------------------------------- hip paths
+    def _hip_cast_weights(self):
+        st_h = native.current_stream_handle()
+        for i, st in enumerate(self.model.spec.stages):
+            self._C.deep_cast_wt(self.model.view(f"conv{i}_w"),
+                                 self.ws.wbf[i], self.ws.wbfT[i], st.kcp,
+                                 st.cout, st_h)
+
     def _hip_forward(self, x: torch.Tensor, labels: torch.Tensor, B: int,
                      mode: int):
         m, w, spec = self.model, self.ws, self.model.spec
         st_h = native.current_stream_handle()
+        self._hip_cast_weights()
         src = x
         for i, st in enumerate(spec.stages):
             self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin, st.k,
@@ -103,7 +117,8 @@ class DeepTrainer:
             M = B * st.h * st.w
             self._C.deep_gemm(w.cols[i], m.view(f"conv{i}_w"),
                               m.view(f"conv{i}_b"), w.acts[i], M, st.kcp,
-                              st.cout, st.kcp, st.cout, 1, 1, st_h)
+                              st.cout, st.kcp, st.cout, 1, 1, st_h,
+                              w.wbfT[i])
             self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
                                   w.pouts[i], B, st.h, st.w, st.cout,
                                   st.pool_k, st_h)
@@ -149,7 +164,8 @@ class DeepTrainer:
                 # dgrad into the cols buffer (its forward use is done)
                 self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
                                   torch.empty(0), w.cols[i], M, st.cout,
-                                  st.kcp, st.cout, st.kcp, 0, 0, st_h)
+                                  st.kcp, st.cout, st.kcp, 0, 0, st_h,
+                                  w.wbf[i])
                 prev = spec.stages[i - 1]
                 self._C.deep_col2im_sigbwd(w.cols[i], w.pouts[i - 1],
                                            w.dppre[i - 1], B, st.h, st.w,
