@@ -455,28 +455,43 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 }
 
 // Column-sum for the conv bias grad: db[n] += sum_m dpre[m][n].
-// 256 threads cover 256/N consecutive rows per iteration (perfectly
-// coalesced); cross-thread tree-reduce per column, then one atomic.
+// Each thread owns 8 consecutive elements of the flat [M*N] stream (one
+// 16B load per iteration); since the grid stride is a multiple of N, the
+// 8 columns a thread sees are FIXED, so it keeps 8 private partials and
+// the block tree-reduces per column at the end.
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_colsum(const act_t* __restrict__ dpre,
                                                 float* __restrict__ db,
                                                 long long M, int N, int G) {
   const int tid = threadIdx.x;
-  const int rows_per_iter = 256 / N;  // N in {32, 64, 128, 256}
-  const int n = tid % N;
-  const int rsub = tid / N;
-  float acc = 0.f;
-  for (long long m = (long long)blockIdx.x * rows_per_iter + rsub; m < M;
-       m += (long long)G * rows_per_iter)
-    acc += ldf(dpre + m * N + n);
-  __shared__ float s[256];
-  s[tid] = acc;
+  const long long total = M * N;
+  const long long stride = (long long)G * 256 * 8;  // multiple of N (N%8==0,
+                                                    // 2048%N==0 for N<=256)
+  long long flat = ((long long)blockIdx.x * 256 + tid) * 8;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (; flat + 8 <= total; flat += stride) {
+    float v8[8];
+    ld8v(dpre + flat, v8);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc[u] += v8[u];
+  }
+  __shared__ float sred[256][8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) sred[tid][u] = acc[u];
   __syncthreads();
-  for (int off = 128; off >= N; off >>= 1) {
-    if (tid < off) s[tid] += s[tid + off];
+  // threads with the same n0 sit N/8 apart; fold the thread axis
+  for (int off = 128; off >= N / 8; off >>= 1) {
+    if (tid < off) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) sred[tid][u] += sred[tid + off][u];
+    }
     __syncthreads();
   }
-  if (tid < N) unsafeAtomicAdd(&db[tid], s[tid]);
+  if (tid < N / 8) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      unsafeAtomicAdd(&db[tid * 8 + u], sred[tid][u]);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -616,6 +631,9 @@ template <typename act_t>
 __global__ __launch_bounds__(256) void k_pool_wgrad(
     const act_t* __restrict__ dppre, const act_t* __restrict__ a,
     float* __restrict__ dpw, int B, int H, int W, int C, int K, int G) {
+  // Two items in flight per iteration: the second item's loads issue
+  // before the first item's FMAs (the K*K+1 loads per item are otherwise
+  // a serial latency chain).
   const int OH = H / K, OW = W / K;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -623,21 +641,34 @@ __global__ __launch_bounds__(256) void k_pool_wgrad(
   for (int w = 0; w < K * K; ++w) acc[w] = 0.f;
   float bacc = 0.f;
   const long long N = (long long)B * OH * OW * C;
-  for (long long it = (long long)blockIdx.x * 256 + tid; it < N;
-       it += (long long)G * 256) {
+  const long long stride = (long long)G * 256;
+
+  auto decode_load = [&](long long it, float* d, float* av) {
     const int c = (int)(it % C);
     long long t = it / C;
     const int q = (int)(t % OW);
     t /= OW;
     const int p = (int)(t % OH);
     const int b = (int)(t / OH);
-    const float d = ldf(dppre + it);
-    bacc += d;
+    *d = ldf(dppre + it);
     for (int i = 0; i < K; ++i)
       for (int j = 0; j < K; ++j)
-        acc[i * K + j] +=
-            d * ldf(a + (((long long)b * H + p * K + i) * W + q * K + j) * C +
-                    c);
+        av[i * K + j] =
+            ldf(a + (((long long)b * H + p * K + i) * W + q * K + j) * C + c);
+  };
+
+  for (long long it = (long long)blockIdx.x * 256 + tid; it < N;
+       it += 2 * stride) {
+    float d1, av1[16], d2 = 0.f, av2[16];
+    decode_load(it, &d1, av1);
+    const long long it2 = it + stride;
+    if (it2 < N) decode_load(it2, &d2, av2);
+    bacc += d1;
+    for (int w = 0; w < K * K; ++w) acc[w] += d1 * av1[w];
+    if (it2 < N) {
+      bacc += d2;
+      for (int w = 0; w < K * K; ++w) acc[w] += d2 * av2[w];
+    }
   }
   __shared__ float red[4][17];
   const int wv = tid >> 6;
