@@ -54,8 +54,13 @@ int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    int b_kxn, int epilogue, int actf, void* stream);
 int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       const float* bias, void* C, long long M, int K, int N,
-                      int ldA, int ldC, int b_kxn, int epilogue, int actf,
-                      void* stream);
+                      int ldA, int ldC, int b_kxn, int epilogue,
+                      const void* imx, int XH, int XW, int XC, int XK,
+                      int XP, int actf, void* stream);
+int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
+                            long long M, int KcP, int N, int MS,
+                            const void* imx, int XH, int XW, int XC, int XK,
+                            int XP, int actf, void* stream);
 int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
                       void* stream);
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
@@ -229,13 +234,16 @@ void deep_im2col(at::Tensor x, at::Tensor cols, int64_t B, int64_t H,
 void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                int64_t M, int64_t K, int64_t N, int64_t ldA, int64_t ldC,
                int64_t b_kxn, int64_t epilogue, int64_t stream,
-               at::Tensor Bpre) {
+               at::Tensor Bpre, at::Tensor imx, int64_t XH, int64_t XW,
+               int64_t XC, int64_t XK, int64_t XP) {
   check_hip(pcnn_deep_gemm_ex(
                 A.data_ptr(), Bsrc.data_ptr<float>(),
                 Bpre.numel() ? Bpre.data_ptr() : nullptr,
                 bias.numel() ? bias.data_ptr<float>() : nullptr,
                 C.data_ptr(), M, (int)K, (int)N, (int)ldA, (int)ldC,
-                (int)b_kxn, (int)epilogue, act_flag(A), (void*)stream),
+                (int)b_kxn, (int)epilogue,
+                imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
+                (int)XC, (int)XK, (int)XP, act_flag(A), (void*)stream),
             "deep_gemm");
 }
 
@@ -249,10 +257,13 @@ void deep_cast_wt(at::Tensor W, at::Tensor out, at::Tensor outT, int64_t R,
 
 void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
                      int64_t M, int64_t KcP, int64_t N, int64_t MS,
-                     int64_t stream) {
-  check_hip(pcnn_deep_wgrad_gemm(cols.data_ptr(), dpre.data_ptr(),
-                                 dW.data_ptr<float>(), M, (int)KcP, (int)N,
-                                 (int)MS, act_flag(cols), (void*)stream),
+                     int64_t stream, at::Tensor imx, int64_t XH, int64_t XW,
+                     int64_t XC, int64_t XK, int64_t XP) {
+  check_hip(pcnn_deep_wgrad_gemm_ex(
+                cols.data_ptr(), dpre.data_ptr(), dW.data_ptr<float>(), M,
+                (int)KcP, (int)N, (int)MS,
+                imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
+                (int)XC, (int)XK, (int)XP, act_flag(cols), (void*)stream),
             "deep_wgrad_gemm");
 }
 
@@ -391,9 +402,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias"), py::arg("C"), py::arg("M"), py::arg("K"),
         py::arg("N"), py::arg("ldA"), py::arg("ldC"), py::arg("b_kxn"),
         py::arg("epilogue"), py::arg("stream"),
-        py::arg("Bpre") = at::empty({0}));
+        py::arg("Bpre") = at::empty({0}),
+        py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
+        py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
+        py::arg("XP") = 0);
   m.def("deep_cast_wt", &deep_cast_wt);
-  m.def("deep_wgrad_gemm", &deep_wgrad_gemm);
+  m.def("deep_wgrad_gemm", &deep_wgrad_gemm, py::arg("cols"),
+        py::arg("dpre"), py::arg("dW"), py::arg("M"), py::arg("KcP"),
+        py::arg("N"), py::arg("MS"), py::arg("stream"),
+        py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
+        py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
+        py::arg("XP") = 0);
   m.def("deep_colsum", &deep_colsum);
   m.def("deep_col2im_sigbwd", &deep_col2im_sigbwd);
   m.def("deep_pool_fwd", &deep_pool_fwd);

@@ -152,6 +152,44 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
     for (int u = 0; u < 8; ++u) cols[m * KcP + kc0 + u] = out[u];
 }
 
+// 8-wide im2col gather for implicit-GEMM staging: returns cols[m][kc0..+8)
+// computed directly from NHWC x (fast path: one 16B load when the span
+// stays in one kernel row and in-image and Cin % 8 == 0).
+template <typename act_t>
+__device__ __forceinline__ void im2col8(const act_t* __restrict__ x, int H,
+                                        int W, int Cin, int K, int P, int b,
+                                        int oh, int ow, int kc0, int Kc,
+                                        float* v8) {
+  const int rowc = K * Cin;
+  if ((Cin % 8) == 0 && kc0 + 8 <= Kc && (kc0 % rowc) + 8 <= rowc) {
+    const int i = kc0 / rowc;
+    const int ih = oh + i - P;
+    const int t0 = kc0 - i * rowc;
+    const int j0 = t0 / Cin;
+    const int j7 = (t0 + 7) / Cin;
+    if (ih >= 0 && ih < H && (ow + j0 - P) >= 0 && (ow + j7 - P) < W) {
+      ld8v(x + (((long long)b * H + ih) * W + (ow - P)) * Cin + t0, v8);
+      return;
+    }
+  }
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int kc = kc0 + u;
+    float val = 0.f;
+    if (kc < Kc) {
+      const int i = kc / rowc;
+      const int t = kc - i * rowc;
+      const int j = t / Cin;
+      const int ci = t - j * Cin;
+      const int ih = oh + i - P;
+      const int iw = ow + j - P;
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+        val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
+    }
+    v8[u] = val;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // MFMA GEMM fragment maps for mfma_f32_16x16x32_bf16 (validated on-device
 // by k_mfma_selftest / tests):
@@ -184,12 +222,16 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 // Bpre (optional): pre-cast bf16 B in [N][K] row-per-output-column layout;
 // when non-null it replaces Bsrc/b_kxn and stages with two 16B copies per
 // thread per tile.
+// imx != null: the A operand is the im2col view of NHWC imx (implicit
+// GEMM — no materialized cols buffer); A/ldA are ignored, K = KcP, and
+// the imx geometry is (XH, XW, XC, XK, XP).
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_gemm(
     const act_t* __restrict__ A, const float* __restrict__ Bsrc,
     const __bf16* __restrict__ Bpre, const float* __restrict__ bias,
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
-    int b_kxn, int epilogue) {
+    int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
+    int XC, int XK, int XP) {
   __shared__ GemmLds L;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -213,6 +255,15 @@ __global__ __launch_bounds__(256) void k_gemm(
   float ra[2][8];    // A prefetch
   float rb[2][8];    // B prefetch
   const long long m_a = m0 + row_a;
+  // implicit-A: decode this thread's im2col row position once
+  int ib = 0, ioh = 0, iow = 0, iKc = 0;
+  if (imx != nullptr) {
+    iow = (int)(m_a % XW);
+    const long long bh = m_a / XW;
+    ioh = (int)(bh % XH);
+    ib = (int)(bh / XH);
+    iKc = XK * XK * XC;
+  }
 
   auto load_regs = [&](int kt) {
 #pragma unroll
@@ -220,7 +271,13 @@ __global__ __launch_bounds__(256) void k_gemm(
       const int kk = kq + h * 8;
 #pragma unroll
       for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
-      if (m_a < M && (kt + kk) < K) ld8v(A + m_a * ldA + kt + kk, ra[h]);
+      if (m_a < M && (kt + kk) < K) {
+        if (imx != nullptr)
+          im2col8(imx, XH, XW, XC, XK, XP, ib, ioh, iow, kt + kk, iKc,
+                  ra[h]);
+        else
+          ld8v(A + m_a * ldA + kt + kk, ra[h]);
+      }
     }
     if (Bpre != nullptr) {
       // bf16 [N][K] rows: two vector loads per thread
@@ -338,10 +395,12 @@ __global__ __launch_bounds__(256) void k_gemm(
 // Both operands are transpose-staged into the [row][k=m] LDS image; the
 // M dimension is the MFMA K axis.  Grid: (kc-tiles) x (n-tiles) x MS
 // M-slices; fp32 hardware atomics combine slices.
+// imx != null: the cols operand is the im2col view of NHWC imx.
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad_gemm(
     const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
-    float* __restrict__ dW, long long M, int KcP, int N, int MS) {
+    float* __restrict__ dW, long long M, int KcP, int N, int MS,
+    const act_t* __restrict__ imx, int XH, int XW, int XC, int XK, int XP) {
   __shared__ GemmLds L;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -370,6 +429,8 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
   float rc[2][8];  // cols prefetch
   float rd[2][8];  // dpre prefetch
 
+  const int iKc = XK * XK * XC;
+
   auto load_regs = [&](long long mt) {
     const long long m = mt + row_s;
     const bool ok = m < m_hi && m < M;
@@ -381,7 +442,14 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
         rc[h][u] = 0.f;
         rd[h][u] = 0.f;
       }
-      if (ok && (kc0 + c + 8) <= KcP) {
+      if (ok && imx != nullptr) {
+        if (kc0 + c < KcP) {
+          const int iow = (int)(m % XW);
+          const long long bh = m / XW;
+          im2col8(imx, XH, XW, XC, XK, XP, (int)(bh / XH),
+                  (int)(bh % XH), iow, kc0 + c, iKc, rc[h]);
+        }
+      } else if (ok && (kc0 + c + 8) <= KcP) {
         ld8v(cols + m * KcP + kc0 + c, rc[h]);
       } else if (ok) {
         const act_t* src = cols + m * KcP + kc0 + c;
@@ -875,8 +943,9 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
 
 int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       const float* bias, void* C, long long M, int K, int N,
-                      int ldA, int ldC, int b_kxn, int epilogue, int actf,
-                      void* stream) {
+                      int ldA, int ldC, int b_kxn, int epilogue,
+                      const void* imx, int XH, int XW, int XC, int XK,
+                      int XP, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
   const long long mtiles = (M + BM - 1) / BM;
   dim3 grid((unsigned)(mtiles * ntiles)), block(256);
@@ -884,7 +953,8 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                                     (hipStream_t)stream, (const act_t*)A,
                                     Bsrc, (const __bf16*)Bpre, bias,
                                     (act_t*)C, M, K, N, ldA, ldC,
-                                    b_kxn, epilogue));
+                                    b_kxn, epilogue, (const act_t*)imx, XH,
+                                    XW, XC, XK, XP));
   return (int)hipGetLastError();
 }
 
@@ -892,7 +962,8 @@ int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    void* C, long long M, int K, int N, int ldA, int ldC,
                    int b_kxn, int epilogue, int actf, void* stream) {
   return pcnn_deep_gemm_ex(A, Bsrc, nullptr, bias, C, M, K, N, ldA, ldC,
-                           b_kxn, epilogue, actf, stream);
+                           b_kxn, epilogue, nullptr, 0, 0, 0, 0, 0, actf,
+                           stream);
 }
 
 int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
@@ -904,9 +975,10 @@ int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
-                         long long M, int KcP, int N, int MS, int actf,
-                         void* stream) {
+int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
+                            long long M, int KcP, int N, int MS,
+                            const void* imx, int XH, int XW, int XC, int XK,
+                            int XP, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
   const int ktiles = (KcP + BM - 1) / BM;
   dim3 grid((unsigned)(ktiles * ntiles * MS)), block(256);
@@ -914,8 +986,16 @@ int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                                           0, (hipStream_t)stream,
                                           (const act_t*)cols,
                                           (const act_t*)dpre, dW, M, KcP, N,
-                                          MS));
+                                          MS, (const act_t*)imx, XH, XW, XC,
+                                          XK, XP));
   return (int)hipGetLastError();
+}
+
+int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
+                         long long M, int KcP, int N, int MS, int actf,
+                         void* stream) {
+  return pcnn_deep_wgrad_gemm_ex(cols, dpre, dW, M, KcP, N, MS, nullptr, 0,
+                                 0, 0, 0, 0, actf, stream);
 }
 
 int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,

@@ -112,13 +112,19 @@ class DeepTrainer:
         self._hip_cast_weights()
         src = x
         for i, st in enumerate(spec.stages):
-            self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin, st.k,
-                                st.pad, st.kcp, st_h)
+            # implicit-im2col GEMM when the input channel count is 16B-
+            # aligned (stages 2+); stage 1 (Cin=3) materializes cols
+            implicit = st.cin % 8 == 0
+            if not implicit:
+                self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin,
+                                    st.k, st.pad, st.kcp, st_h)
             M = B * st.h * st.w
             self._C.deep_gemm(w.cols[i], m.view(f"conv{i}_w"),
                               m.view(f"conv{i}_b"), w.acts[i], M, st.kcp,
                               st.cout, st.kcp, st.cout, 1, 1, st_h,
-                              w.wbfT[i])
+                              w.wbfT[i],
+                              src if implicit else torch.empty(0),
+                              st.h, st.w, st.cin, st.k, st.pad)
             self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
                                   w.pouts[i], B, st.h, st.w, st.cout,
                                   st.pool_k, st_h)
@@ -154,9 +160,13 @@ class DeepTrainer:
             ktiles = (st.kcp + 63) // 64
             ntiles = (st.cout + 63) // 64
             ms = max(1, min(128, 512 // (ktiles * ntiles)))
+            x_in = x if i == 0 else w.pouts[i - 1]
+            implicit = st.cin % 8 == 0
             self._C.deep_wgrad_gemm(w.cols[i], dapre,
                                     m.grad_view(f"conv{i}_w"), M, st.kcp,
-                                    st.cout, ms, st_h)
+                                    st.cout, ms, st_h,
+                                    x_in if implicit else torch.empty(0),
+                                    st.h, st.w, st.cin, st.k, st.pad)
             gsum = max(32, min(512, (M * st.cout) // (256 * 96)))
             self._C.deep_colsum(dapre, m.grad_view(f"conv{i}_b"), M,
                                 st.cout, gsum, st_h)
