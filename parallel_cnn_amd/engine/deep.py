@@ -112,9 +112,11 @@ class DeepTrainer:
         self._hip_cast_weights()
         src = x
         for i, st in enumerate(spec.stages):
-            # implicit-im2col GEMM when the input channel count is 16B-
-            # aligned (stages 2+); stage 1 (Cin=3) materializes cols
-            implicit = st.cin % 8 == 0
+            # implicit-im2col staging measured SLOWER (256k vs 299k img/s
+            # at bs=256): the conditional gather inside the K-loop breaks
+            # the load pipeline (hipcc branches around the loads).  The
+            # kernels keep the capability; the engine materializes cols.
+            implicit = False
             if not implicit:
                 self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin,
                                     st.k, st.pad, st.kcp, st_h)
@@ -161,7 +163,7 @@ class DeepTrainer:
             ntiles = (st.cout + 63) // 64
             ms = max(1, min(128, 512 // (ktiles * ntiles)))
             x_in = x if i == 0 else w.pouts[i - 1]
-            implicit = st.cin % 8 == 0
+            implicit = False
             self._C.deep_wgrad_gemm(w.cols[i], dapre,
                                     m.grad_view(f"conv{i}_w"), M, st.kcp,
                                     st.cout, ms, st_h,
