@@ -202,11 +202,13 @@ constexpr int BM = 64, BN = 64, BK = 64;  // 64-deep K-step: 2 MFMAs per
                                           // fragment per barrier pair
 constexpr int LDP = BK + 8;  // LDS row stride (bf16) — conflict-free b128
 
-struct GemmLds {
-  __bf16 As[BM][LDP];  // A tile, [m][k]
-  __bf16 Bs[BN][LDP];  // B tile, [n][k]  (transposed image: frag reads are
-                       // contiguous along k for both operands)
+template <int TBM>
+struct GemmLdsT {
+  __bf16 As[TBM][LDP];  // A tile, [m][k]
+  __bf16 Bs[BN][LDP];   // B tile, [n][k]  (transposed image: frag reads
+                        // are contiguous along k for both operands)
 };
+using GemmLds = GemmLdsT<BM>;
 
 // Load an 8-element bf16 fragment from an LDS row.
 __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
@@ -225,35 +227,45 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 // imx != null: the A operand is the im2col view of NHWC imx (implicit
 // GEMM — no materialized cols buffer); A/ldA are ignored, K = KcP, and
 // the imx geometry is (XH, XW, XC, XK, XP).
-template <typename act_t>
+// TBM: M-tile (64 or 128).  At 128 each wave owns two 16-row fragments
+// (16 MFMAs per K-step — double the compute per barrier pair) at +9 KB
+// LDS; the launcher picks it for large-M calls.
+template <typename act_t, int TBM>
 __global__ __launch_bounds__(256) void k_gemm(
     const act_t* __restrict__ A, const float* __restrict__ Bsrc,
     const __bf16* __restrict__ Bpre, const float* __restrict__ bias,
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
     int XC, int XK, int XP) {
-  __shared__ GemmLds L;
+  __shared__ GemmLdsT<TBM> L;
+  constexpr int RF = TBM / 64;        // row fragments per wave
+  constexpr int TPR = 256 / TBM;      // staging threads per A row
+  constexpr int SPAN = 64 / TPR;      // k-span per staging thread
+  constexpr int NCH = SPAN / 8;       // 8-chunks per staging thread
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
   const int ntiles = (N + BN - 1) / BN;
   const long long mtile = blockIdx.x / ntiles;
   const int ntile = (int)(blockIdx.x % ntiles);
-  const long long m0 = mtile * BM;
+  const long long m0 = mtile * TBM;
   const int n0 = ntile * BN;
   const int nf = min(BN, N - n0) / 16;  // fragments along N (N % 16 == 0)
 
-  f32x4 acc[BN / 16];
+  f32x4 acc[RF][BN / 16];
 #pragma unroll
-  for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+  for (int rf = 0; rf < RF; ++rf)
+#pragma unroll
+    for (int f = 0; f < BN / 16; ++f) acc[rf][f] = {0.f, 0.f, 0.f, 0.f};
 
-  const int row_a = tid >> 2;            // 64 rows, 4 threads each
-  const int kq = (tid & 3) * 16;         // 16 k per thread (two 8-chunks)
+  const int row_a = tid / TPR;
+  const int kq = (tid % TPR) * SPAN;
+  const int bkq = (tid & 3) * 16;     // B staging: 4 threads per row
   // T14-style software pipeline: next tile's global loads are issued into
-  // registers BEFORE the MFMA block (HBM/L2 latency hides under compute);
-  // the LDS write happens after the read barrier.
-  float ra[2][8];    // A prefetch
-  float rb[2][8];    // B prefetch
+  // registers BEFORE the MFMA block; the LDS write happens after the read
+  // barrier.
+  float ra[NCH][8];
+  float rb[2][8];
   const long long m_a = m0 + row_a;
   // implicit-A: decode this thread's im2col row position once
   int ib = 0, ioh = 0, iow = 0, iKc = 0;
@@ -267,7 +279,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   auto load_regs = [&](int kt) {
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int h = 0; h < NCH; ++h) {
       const int kk = kq + h * 8;
 #pragma unroll
       for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
@@ -285,7 +297,7 @@ __global__ __launch_bounds__(256) void k_gemm(
       const bool ok = (n0 + n) < N;
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
-        const int kk = kq + h * 8;
+        const int kk = bkq + h * 8;
 #pragma unroll
         for (int u = 0; u < 8; ++u) rb[h][u] = 0.f;
         if (ok && (kt + kk) < K)
@@ -294,7 +306,7 @@ __global__ __launch_bounds__(256) void k_gemm(
                rb[h]);
       }
     } else if (b_kxn) {
-      // Bsrc[K][N]: thread owns row k = tid>>2, 16 n (transposed write)
+      // Bsrc[K][N]: read rows k (coalesced along n), write transposed
       const int k = tid >> 2;
       const int nq = (tid & 3) * 16;
       const bool kok = (kt + k) < K;
@@ -311,7 +323,7 @@ __global__ __launch_bounds__(256) void k_gemm(
       const bool ok = (n0 + n) < N;
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
-        const int kk = kq + h * 8;
+        const int kk = bkq + h * 8;
         const float* src = Bsrc + (long long)(n0 + n) * K + kt + kk;
         const bool kok = ok && (kt + kk) < K;
 #pragma unroll
@@ -322,7 +334,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   auto write_lds = [&]() {
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int h = 0; h < NCH; ++h) {
       const int kk = kq + h * 8;
 #pragma unroll
       for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)ra[h][u];
@@ -339,7 +351,7 @@ __global__ __launch_bounds__(256) void k_gemm(
       const int n = tid >> 2;
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
-        const int kk = kq + h * 8;
+        const int kk = bkq + h * 8;
 #pragma unroll
         for (int u = 0; u < 8; ++u) L.Bs[n][kk + u] = (__bf16)rb[h][u];
       }
@@ -352,18 +364,22 @@ __global__ __launch_bounds__(256) void k_gemm(
     __syncthreads();  // tile kt visible in LDS
     const bool more = kt + BK < K;
     if (more) load_regs(kt + BK);
-    // wave wv owns C rows [wv*16, wv*16+16); two 32-deep MFMA sub-steps
+    // wave wv owns C rows [wv*16*RF, +16*RF); two 32-deep MFMA sub-steps
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
-                                      kk * 32 + (lane >> 4) * 8);
 #pragma unroll
-      for (int f = 0; f < BN / 16; ++f) {
-        if (f < nf) {
-          const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
-                                          kk * 32 + (lane >> 4) * 8);
-          acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0,
-                                                           0, 0);
+      for (int rf = 0; rf < RF; ++rf) {
+        const bf16x8 a0 =
+            frag_from_lds(L.As[(wv * RF + rf) * 16 + (lane & 15)],
+                          kk * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int f = 0; f < BN / 16; ++f) {
+          if (f < nf) {
+            const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
+                                            kk * 32 + (lane >> 4) * 8);
+            acc[rf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a0, bf, acc[rf][f], 0, 0, 0);
+          }
         }
       }
     }
@@ -372,19 +388,22 @@ __global__ __launch_bounds__(256) void k_gemm(
   }
 
   // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
-  const int crow = wv * 16 + (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
-  for (int f = 0; f < BN / 16; ++f) {
-    if (f < nf) {
-      const int n = n0 + f * 16 + ccol;
+  for (int rf = 0; rf < RF; ++rf) {
+    const int crow = (wv * RF + rf) * 16 + (lane >> 4) * 4;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const long long m = m0 + crow + r;
-        if (m < M && n < N) {
-          float v = acc[f][r];
-          if (epilogue == 1) v = sigmoidf_dev(v + bias[n]);
-          stf(C + m * ldC + n, v);
+    for (int f = 0; f < BN / 16; ++f) {
+      if (f < nf) {
+        const int n = n0 + f * 16 + ccol;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long long m = m0 + crow + r;
+          if (m < M && n < N) {
+            float v = acc[rf][f][r];
+            if (epilogue == 1) v = sigmoidf_dev(v + bias[n]);
+            stf(C + m * ldC + n, v);
+          }
         }
       }
     }
@@ -947,14 +966,28 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       const void* imx, int XH, int XW, int XC, int XK,
                       int XP, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
-  const long long mtiles = (M + BM - 1) / BM;
+  // 128-row tiles once the grid still fills the chip at the bigger tile
+  const bool big = M >= 128 * 512;
+  const int tbm = big ? 128 : BM;
+  const long long mtiles = (M + tbm - 1) / tbm;
   dim3 grid((unsigned)(mtiles * ntiles)), block(256);
-  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t>), grid, block, 0,
-                                    (hipStream_t)stream, (const act_t*)A,
-                                    Bsrc, (const __bf16*)Bpre, bias,
-                                    (act_t*)C, M, K, N, ldA, ldC,
-                                    b_kxn, epilogue, (const act_t*)imx, XH,
-                                    XW, XC, XK, XP));
+  if (big) {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, 128>), grid, block,
+                                      0, (hipStream_t)stream,
+                                      (const act_t*)A, Bsrc,
+                                      (const __bf16*)Bpre, bias, (act_t*)C,
+                                      M, K, N, ldA, ldC, b_kxn, epilogue,
+                                      (const act_t*)imx, XH, XW, XC, XK,
+                                      XP));
+  } else {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, BM>), grid, block,
+                                      0, (hipStream_t)stream,
+                                      (const act_t*)A, Bsrc,
+                                      (const __bf16*)Bpre, bias, (act_t*)C,
+                                      M, K, N, ldA, ldC, b_kxn, epilogue,
+                                      (const act_t*)imx, XH, XW, XC, XK,
+                                      XP));
+  }
   return (int)hipGetLastError();
 }
 
