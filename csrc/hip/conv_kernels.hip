@@ -966,8 +966,11 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       const void* imx, int XH, int XW, int XC, int XK,
                       int XP, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
-  // 128-row tiles once the grid still fills the chip at the bigger tile
-  const bool big = M >= 128 * 512;
+  // 128-row tiles measured neutral-to-negative at this family's shapes
+  // (the 2-barrier loop is staging/latency bound, not MFMA bound — see
+  // profiles/: MfmaUtil ~3%); the instantiation stays available but the
+  // selector is off.
+  const bool big = false && M >= 128 * 512;
   const int tbm = big ? 128 : BM;
   const long long mtiles = (M + tbm - 1) / tbm;
   dim3 grid((unsigned)(mtiles * ntiles)), block(256);
