@@ -41,6 +41,10 @@ def main() -> int:
                    choices=["lenet5", "deepcnn"])
     p.add_argument("--use-graph", action="store_true",
                    help="capture the step in a hipGraph and replay it")
+    p.add_argument("--pool", default="trainable",
+                   choices=["trainable", "max"])
+    p.add_argument("--loss", default="residual",
+                   choices=["residual", "softmax_ce"])
     p.add_argument("--overlap-comm", action="store_true",
                    help="two-bucket DP: overlap the fc/pool grad all-reduce "
                         "with the conv wgrad")
@@ -55,7 +59,8 @@ def main() -> int:
     cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
                       device=args.device, log_interval=0, data="synthetic",
                       wgrad_chunk=args.wgrad_chunk, model=args.model,
-                      overlap_comm=args.overlap_comm)
+                      overlap_comm=args.overlap_comm, pool=args.pool,
+                      loss=args.loss)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
@@ -139,6 +144,8 @@ def main() -> int:
             "per_gpu_batch": B,
             "input": "32x32x3" if args.model == "deepcnn" else "28x28x1",
             "parallelism": f"dp{n_gpus}",
+            "pool": args.pool,
+            "loss": args.loss,
             "backend": trainer.backend,
             "hipgraph": bool(getattr(trainer, "_graph", None)),
         },

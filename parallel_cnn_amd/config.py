@@ -17,6 +17,7 @@ from .ops.shapes import REF_DT, REF_THRESHOLD
 class TrainConfig:
     # model / numerics
     model: str = "lenet5"
+    deep_channels: str = "32,64,64"  # DeepCNN stage widths (x16 each)
     act_dtype: str = "bf16"          # activation storage on GPU: bf16 | fp16 | fp32
     pool: str = "trainable"          # trainable (reference) | max
     loss: str = "residual"           # residual (reference) | softmax_ce

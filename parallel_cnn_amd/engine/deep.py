@@ -71,7 +71,13 @@ class DeepTrainer:
         self.backend = backend
         if backend == "hip":
             self._C = native.require()
-        self.model = model or DeepCNN(self.device, seed=cfg.seed)
+        if model is None:
+            from ..models.deepcnn import DeepCNNSpec
+            channels = tuple(
+                int(c) for c in str(cfg.deep_channels).split(",") if c)
+            spec = DeepCNNSpec(channels=channels)
+            model = DeepCNN(self.device, seed=cfg.seed, spec=spec)
+        self.model = model
         act_map = {"bf16": torch.bfloat16, "fp16": torch.float16,
                    "fp32": torch.float32}
         self.act_dtype = (act_map[cfg.act_dtype] if backend == "hip"

@@ -69,6 +69,11 @@ class DeepCNNSpec:
     stages: List[ConvStage] = field(default_factory=list)
 
     def __post_init__(self):
+        for c in self.channels:
+            if c % 16 != 0:
+                raise ValueError(
+                    f"DeepCNN channel widths must be multiples of 16 "
+                    f"(MFMA fragment width), got {self.channels}")
         if not self.stages:
             cin, h, w = self.in_ch, self.in_h, self.in_w
             for cout in self.channels:

@@ -132,3 +132,22 @@ def test_im2col_ref_matches_manual():
                 if 0 <= ih < st.h and 0 <= iw < st.w else 0.0)
         assert abs(cols[m, kc].item() - want) < 1e-6
     assert cols[:, st.kc:].abs().sum() == 0
+
+
+def test_custom_channel_widths():
+    from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
+    spec = DeepCNNSpec(channels=(16, 32, 48))
+    assert [s.cout for s in spec.stages] == [16, 32, 48]
+    assert spec.fc_in == 4 * 4 * 48
+    cfg = TrainConfig(backend="torchref", device="cpu", batch_size=4,
+                      log_interval=0, deep_channels="16,32,48")
+    t = DeepTrainer(cfg)
+    assert t.model.spec.n_params == spec.n_params
+    x, y = synthetic_images(4, 32, 32, 3, seed=1)
+    t.step(*t.stage_batch(x, y))
+
+
+def test_channel_width_validation():
+    from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
+    with pytest.raises(ValueError, match="multiples of 16"):
+        DeepCNNSpec(channels=(20, 64, 64))
