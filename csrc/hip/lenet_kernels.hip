@@ -725,7 +725,9 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   if (GC < 2) GC = 2;
   if (GC > 256) GC = 256;
   if (roles == 4) GC = 0;  // fc-only launch (fused-wgrad mode)
-  int GS = (B * S1_OUT + 256 * 48 - 1) / (256 * 48);
+  // pool-role slices: ~10 (image,cell) items per thread (2 blocks at B=64
+  // left the pool role as the kernel's longest pole)
+  int GS = (B * S1_OUT + 256 * 10 - 1) / (256 * 10);
   if (GS < 2) GS = 2;
   if (GS > 96) GS = 96;
   if (roles == 4) GS = 0;
