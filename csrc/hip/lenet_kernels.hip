@@ -727,7 +727,7 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   if (roles == 4) GC = 0;  // fc-only launch (fused-wgrad mode)
   // pool-role slices: ~10 (image,cell) items per thread (2 blocks at B=64
   // left the pool role as the kernel's longest pole)
-  int GS = (B * S1_OUT + 256 * 10 - 1) / (256 * 10);
+  int GS = (B * S1_OUT + 256 * 5 - 1) / (256 * 5);
   if (GS < 2) GS = 2;
   if (GS > 96) GS = 96;
   if (roles == 4) GS = 0;
