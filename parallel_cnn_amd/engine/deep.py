@@ -87,6 +87,11 @@ class DeepTrainer:
         self._loss_host = 0.0
         self._samples_seen = 0
         self.global_step = 0
+        self.timers = None  # set by enable_profiling
+
+    def enable_profiling(self) -> None:
+        from ..utils.timers import PhaseTimers
+        self.timers = PhaseTimers()
 
     # ------------------------------------------------------------------ util
     def _scale(self, B: int) -> float:
@@ -196,6 +201,20 @@ class DeepTrainer:
         assert B <= self.ws.max_batch
         scale = self._scale(B)
         if self.backend == "hip":
+            if self.timers is not None:
+                with self.timers.phase("forward"):
+                    self._hip_forward(x, labels, B, MODE_TRAIN)
+                with self.timers.phase("backward"):
+                    self._hip_backward(x, B)
+                with self.timers.phase("all-reduce"):
+                    pdist.allreduce_grads(self.model.grads)
+                with self.timers.phase("update"):
+                    self._C.deep_update(self.model.params, self.model.grads,
+                                        self.cfg.dt * scale,
+                                        native.current_stream_handle())
+                self._samples_seen += B * self.ctx.world_size
+                self.global_step += 1
+                return
             self._hip_forward(x, labels, B, MODE_TRAIN)
             self._hip_backward(x, B)
             pdist.allreduce_grads(self.model.grads)
