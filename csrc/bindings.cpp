@@ -22,26 +22,26 @@ void cpu_update(at::Tensor params, at::Tensor grads, double dt, double scale);
 
 extern "C" {
 int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
-                       float* y, float* dz, float* dz2, float* dz1,
+                       float* y, float* dz, float* dz2, void* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int mode, void* stream);
 int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
                           void* a2, float* y, float* dz, float* dz2,
-                          float* dz1, const int* labels, float* loss_accum,
+                          void* dz1, const int* labels, float* loss_accum,
                           int* correct, int B, int act_is_bf16, int mode,
                           int pool_mode, int loss_mode, void* stream);
 int pcnn_launch_fwdbwd_ex2(const void* x, const float* params, void* a1,
                            void* a2, float* y, float* dz, float* dz2,
-                           float* dz1, const int* labels, float* loss_accum,
+                           void* dz1, const int* labels, float* loss_accum,
                            int* correct, int B, int act_is_bf16, int mode,
                            int pool_mode, int loss_mode, float* grads,
                            int wgrad_fuse, void* stream);
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
-                      const float* dz, const float* dz2, const float* dz1,
+                      const float* dz, const float* dz2, const void* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
                       void* stream);
 int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
-                         const float* dz, const float* dz2, const float* dz1,
+                         const float* dz, const float* dz2, const void* dz1,
                          float* grads, int B, int act_is_bf16, int chunk_imgs,
                          int roles, void* stream);
 int pcnn_launch_update(float* params, float* grads, float step, void* stream);
@@ -119,13 +119,15 @@ void hip_fwdbwd(at::Tensor x, at::Tensor params, at::Tensor a1, at::Tensor a2,
                 at::Tensor grads, int64_t wgrad_fuse) {
   TORCH_CHECK(x.is_cuda() && params.is_cuda(), "expected device tensors");
   TORCH_CHECK(labels.scalar_type() == at::kInt, "labels must be int32");
+  TORCH_CHECK(!dz1.numel() || dz1.scalar_type() == x.scalar_type(),
+              "dz1 must match the activation dtype");
   int f = act_flag(x);
   check_hip(pcnn_launch_fwdbwd_ex2(
                 x.data_ptr(), params.data_ptr<float>(), a1.data_ptr(),
                 a2.data_ptr(), y.numel() ? y.data_ptr<float>() : nullptr,
                 dz.numel() ? dz.data_ptr<float>() : nullptr,
                 dz2.numel() ? dz2.data_ptr<float>() : nullptr,
-                dz1.numel() ? dz1.data_ptr<float>() : nullptr,
+                dz1.numel() ? dz1.data_ptr() : nullptr,
                 labels.data_ptr<int>(),
                 loss_accum.numel() ? loss_accum.data_ptr<float>() : nullptr,
                 correct_accum.numel() ? correct_accum.data_ptr<int>() : nullptr,
@@ -139,9 +141,11 @@ void hip_wgrad(at::Tensor x, at::Tensor a1, at::Tensor a2, at::Tensor dz,
                at::Tensor dz2, at::Tensor dz1, at::Tensor grads, int64_t B,
                int64_t chunk_imgs, int64_t stream) {
   int f = act_flag(x);
+  TORCH_CHECK(dz1.scalar_type() == x.scalar_type(),
+              "dz1 must match the activation dtype");
   check_hip(pcnn_launch_wgrad(x.data_ptr(), a1.data_ptr(), a2.data_ptr(),
                               dz.data_ptr<float>(), dz2.data_ptr<float>(),
-                              dz1.data_ptr<float>(), grads.data_ptr<float>(),
+                              dz1.data_ptr(), grads.data_ptr<float>(),
                               (int)B, f, (int)chunk_imgs, (void*)stream),
             "wgrad");
 }
@@ -154,7 +158,7 @@ void hip_wgrad_roles(at::Tensor x, at::Tensor a1, at::Tensor a2,
   int f = act_flag(x);
   check_hip(pcnn_launch_wgrad_ex(x.data_ptr(), a1.data_ptr(), a2.data_ptr(),
                                  dz.data_ptr<float>(), dz2.data_ptr<float>(),
-                                 dz1.data_ptr<float>(), grads.data_ptr<float>(),
+                                 dz1.data_ptr(), grads.data_ptr<float>(),
                                  (int)B, f, (int)chunk_imgs, (int)roles,
                                  (void*)stream),
             "wgrad_roles");
@@ -204,14 +208,14 @@ void hip_train_steps(at::Tensor x_pool, at::Tensor labels_pool,
                                      y.data_ptr<float>(),
                                      dz.data_ptr<float>(),
                                      dz2.data_ptr<float>(),
-                                     dz1.data_ptr<float>(), lb,
+                                     dz1.data_ptr(), lb,
                                      loss_accum.data_ptr<float>(), nullptr,
                                      (int)B, f, 0, (int)pool_mode,
                                      (int)loss_mode, gp, (int)wgrad_fuse, s),
               "train_steps/fwdbwd");
     check_hip(pcnn_launch_wgrad_ex(xb, a1.data_ptr(), a2.data_ptr(),
                                    dz.data_ptr<float>(), dz2.data_ptr<float>(),
-                                   dz1.data_ptr<float>(), gp, (int)B, f,
+                                   dz1.data_ptr(), gp, (int)B, f,
                                    (int)chunk_imgs, wroles, s),
               "train_steps/wgrad");
     check_hip(pcnn_launch_update(pp, gp, (float)step_scale, s),

@@ -134,7 +134,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
     const act_t* __restrict__ x, const float* __restrict__ params,
     act_t* __restrict__ a1g, act_t* __restrict__ a2g, float* __restrict__ yg,
     float* __restrict__ dzg, float* __restrict__ dz2g,
-    float* __restrict__ dz1g, const int* __restrict__ labels,
+    act_t* __restrict__ dz1g, const int* __restrict__ labels,
     float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B,
     int pool_mode, int loss_mode, float* __restrict__ grads,
     int wgrad_fuse) {
@@ -315,9 +315,9 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
                 : d2 * L.ps[OFF_S1W + i * S1_K + j];
         dz1v[i * S1_K + j] = dd * av * (1.0f - av);
       }
-      *reinterpret_cast<float4*>(
-          dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
-          pc * S1_K) = *reinterpret_cast<const float4*>(&dz1v[i * S1_K]);
+      st4f(dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
+               pc * S1_K,
+           &dz1v[i * S1_K]);
     }
     if (wgrad_fuse) {
       // conv1 wgrad: this cell's 16 dz1 x its 8x8 input window (from LDS);
@@ -408,7 +408,7 @@ template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad(
     const act_t* __restrict__ x, const act_t* __restrict__ a1g,
     const act_t* __restrict__ a2g, const float* __restrict__ dzg,
-    const float* __restrict__ dz2g, const float* __restrict__ dz1g,
+    const float* __restrict__ dz2g, const act_t* __restrict__ dz1g,
     float* __restrict__ grads, int B, int GC, int GS, int FS, int roles) {
   const int tid = threadIdx.x;
   const int blk = blockIdx.x;
@@ -642,7 +642,7 @@ using namespace pcnn;
 namespace {
 template <int MODE>
 int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
-                       float* y, float* dz, float* dz2, float* dz1,
+                       float* y, float* dz, float* dz2, void* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int pool_mode, int loss_mode,
                        float* grads, int wgrad_fuse, hipStream_t stream) {
@@ -650,18 +650,18 @@ int launch_fwdbwd_mode(const void* x, const float* params, void* a1, void* a2,
   if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_fwdbwd<bf16, MODE>), grid, block, 0, stream,
                        (const bf16*)x, params, (bf16*)a1, (bf16*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode, grads, wgrad_fuse);
+                       dz2, (bf16*)dz1, labels, loss_accum, correct, B,
+                       pool_mode, loss_mode, grads, wgrad_fuse);
   } else if (act_is_bf16 == 2) {
     hipLaunchKernelGGL((k_fwdbwd<fp16, MODE>), grid, block, 0, stream,
                        (const fp16*)x, params, (fp16*)a1, (fp16*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode, grads, wgrad_fuse);
+                       dz2, (fp16*)dz1, labels, loss_accum, correct, B,
+                       pool_mode, loss_mode, grads, wgrad_fuse);
   } else {
     hipLaunchKernelGGL((k_fwdbwd<float, MODE>), grid, block, 0, stream,
                        (const float*)x, params, (float*)a1, (float*)a2, y, dz,
-                       dz2, dz1, labels, loss_accum, correct, B, pool_mode,
-                       loss_mode, grads, wgrad_fuse);
+                       dz2, (float*)dz1, labels, loss_accum, correct, B,
+                       pool_mode, loss_mode, grads, wgrad_fuse);
   }
   return (int)hipGetLastError();
 }
@@ -671,7 +671,7 @@ extern "C" {
 
 int pcnn_launch_fwdbwd_ex2(const void* x, const float* params, void* a1,
                            void* a2, float* y, float* dz, float* dz2,
-                           float* dz1, const int* labels, float* loss_accum,
+                           void* dz1, const int* labels, float* loss_accum,
                            int* correct, int B, int act_is_bf16, int mode,
                            int pool_mode, int loss_mode, float* grads,
                            int wgrad_fuse, void* stream) {
@@ -697,7 +697,7 @@ int pcnn_launch_fwdbwd_ex2(const void* x, const float* params, void* a1,
 
 int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
                           void* a2, float* y, float* dz, float* dz2,
-                          float* dz1, const int* labels, float* loss_accum,
+                          void* dz1, const int* labels, float* loss_accum,
                           int* correct, int B, int act_is_bf16, int mode,
                           int pool_mode, int loss_mode, void* stream) {
   return pcnn_launch_fwdbwd_ex2(x, params, a1, a2, y, dz, dz2, dz1, labels,
@@ -706,7 +706,7 @@ int pcnn_launch_fwdbwd_ex(const void* x, const float* params, void* a1,
 }
 
 int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
-                       float* y, float* dz, float* dz2, float* dz1,
+                       float* y, float* dz, float* dz2, void* dz1,
                        const int* labels, float* loss_accum, int* correct,
                        int B, int act_is_bf16, int mode, void* stream) {
   return pcnn_launch_fwdbwd_ex(x, params, a1, a2, y, dz, dz2, dz1, labels,
@@ -716,7 +716,7 @@ int pcnn_launch_fwdbwd(const void* x, const float* params, void* a1, void* a2,
 
 // chunk_imgs is the conv1 slices-per-channel knob (GC); <=0 -> default.
 int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
-                         const float* dz, const float* dz2, const float* dz1,
+                         const float* dz, const float* dz2, const void* dz1,
                          float* grads, int B, int act_is_bf16, int chunk_imgs,
                          int roles, void* stream) {
   // Batch-adaptive defaults: ~36 (image,position) items per conv thread,
@@ -738,22 +738,22 @@ int pcnn_launch_wgrad_ex(const void* x, const void* a1, const void* a2,
   hipStream_t s = (hipStream_t)stream;
   if (act_is_bf16 == 1) {
     hipLaunchKernelGGL((k_wgrad<bf16>), grid, block, 0, s, (const bf16*)x,
-                       (const bf16*)a1, (const bf16*)a2, dz, dz2, dz1, grads,
-                       B, GC, GS, FS, roles);
+                       (const bf16*)a1, (const bf16*)a2, dz, dz2,
+                       (const bf16*)dz1, grads, B, GC, GS, FS, roles);
   } else if (act_is_bf16 == 2) {
     hipLaunchKernelGGL((k_wgrad<fp16>), grid, block, 0, s, (const fp16*)x,
-                       (const fp16*)a1, (const fp16*)a2, dz, dz2, dz1, grads,
-                       B, GC, GS, FS, roles);
+                       (const fp16*)a1, (const fp16*)a2, dz, dz2,
+                       (const fp16*)dz1, grads, B, GC, GS, FS, roles);
   } else {
     hipLaunchKernelGGL((k_wgrad<float>), grid, block, 0, s, (const float*)x,
-                       (const float*)a1, (const float*)a2, dz, dz2, dz1, grads,
-                       B, GC, GS, FS, roles);
+                       (const float*)a1, (const float*)a2, dz, dz2,
+                       (const float*)dz1, grads, B, GC, GS, FS, roles);
   }
   return (int)hipGetLastError();
 }
 
 int pcnn_launch_wgrad(const void* x, const void* a1, const void* a2,
-                      const float* dz, const float* dz2, const float* dz1,
+                      const float* dz, const float* dz2, const void* dz1,
                       float* grads, int B, int act_is_bf16, int chunk_imgs,
                       void* stream) {
   return pcnn_launch_wgrad_ex(x, a1, a2, dz, dz2, dz1, grads, B, act_is_bf16,

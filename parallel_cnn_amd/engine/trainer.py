@@ -41,7 +41,9 @@ class Workspace:
         self.y = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=dev)
         self.dz = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=dev)
         self.dz2 = torch.empty(B, S.S1_OUT, dtype=torch.float32, device=dev)
-        self.dz1 = torch.empty(B, S.C1_OUT, dtype=torch.float32, device=dev)
+        # dz1 is the one large backward tensor: stored in the activation
+        # dtype (fp32 math stays inside the kernels)
+        self.dz1 = torch.empty(B, S.C1_OUT, dtype=act_dtype, device=dev)
         self.loss_accum = torch.zeros(1, dtype=torch.float32, device=dev)
         self.correct_accum = torch.zeros(1, dtype=torch.int32, device=dev)
 

@@ -40,7 +40,7 @@ def run_gpu_step_pieces(x, labels, params, act_dtype, device):
     y = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=device)
     dz = torch.empty(B, S.FC_OUT, dtype=torch.float32, device=device)
     dz2 = torch.empty(B, S.S1_OUT, dtype=torch.float32, device=device)
-    dz1 = torch.empty(B, S.C1_OUT, dtype=torch.float32, device=device)
+    dz1 = torch.empty(B, S.C1_OUT, dtype=ad, device=device)
     loss = torch.zeros(1, dtype=torch.float32, device=device)
     corr = torch.zeros(1, dtype=torch.int32, device=device)
     grads = torch.zeros(S.N_PARAMS, dtype=torch.float32, device=device)
@@ -50,7 +50,7 @@ def run_gpu_step_pieces(x, labels, params, act_dtype, device):
     _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, 8, stream)
     torch.cuda.synchronize()
     return (a1.float().cpu(), a2.float().cpu(), y.cpu(), dz.cpu(), dz2.cpu(),
-            dz1.cpu(), grads.cpu(), float(loss.item()))
+            dz1.float().cpu(), grads.cpu(), float(loss.item()))
 
 
 def ref_step_pieces(x, labels, params):
@@ -200,7 +200,7 @@ def test_wgrad_chunk_sizes_agree(device):
         grads = torch.zeros(S.N_PARAMS, device=device)
         st = native.current_stream_handle()
         _C.hip_fwdbwd(xd, pd, a1, a2, yv, dz, dz2, dz1, ld, loss, corr, B, 0,
-                      st)
+                      st)  # fp32 acts: dz1 fp32 too
         _C.hip_wgrad(xd, a1, a2, dz, dz2, dz1, grads, B, chunk, st)
         torch.cuda.synchronize()
         g = grads.cpu()

@@ -40,7 +40,7 @@ def main():
         y = torch.empty(B, S.FC_OUT, device=DEV)
         dz = torch.empty(B, S.FC_OUT, device=DEV)
         dz2 = torch.empty(B, S.S1_OUT, device=DEV)
-        dz1 = torch.empty(B, S.C1_OUT, device=DEV)
+        dz1 = torch.empty(B, S.C1_OUT, dtype=ad, device=DEV)
         loss = torch.zeros(1, device=DEV)
         corr = torch.zeros(1, dtype=torch.int32, device=DEV)
         grads = torch.zeros(S.N_PARAMS, device=DEV)
