@@ -63,7 +63,7 @@ def main():
             t = time_loop(lambda: _C.hip_wgrad_roles(
                 x, a1, a2, dz, dz2, dz1, grads, B, 0, roles, st))
             print(f"  {name:16s}: {t:8.2f} us")
-        for gc in (2, 4, 8, 16, 32, 64):
+        for gc in (8, 12, 16, 20, 24, 32):
             t = time_loop(lambda: _C.hip_wgrad_roles(
                 x, a1, a2, dz, dz2, dz1, grads, B, gc, 1, st))
             print(f"  wgrad c1 GC={gc:3d}: {t:8.2f} us")
