@@ -8,9 +8,9 @@
 // one training step of a whole batch is THREE kernels:
 //
 //   1. k_fwdbwd  — fused forward + backward-data.  One 256-thread workgroup
-//      per image; activations live in LDS; 5 barrier phases
-//      ({stage x+params} {conv1} {pool} {fc+loss-residual} {fc-bwd}
-//      {pool-bwd}); the fc dot products are 16-lane shuffle reductions so no
+//      per image; 3 barriers total (thread t owns pool cell t end-to-end:
+//      its 16 conv outputs stay in registers from forward into the pool
+//      backward); the fc dot products are 16-lane shuffle reductions so no
 //      phase has a >32-step dependency chain.  Fuses what the reference ran
 //      as 12 separate kernels per sample.
 //   2. k_wgrad   — all weight/bias gradients, batch-reduced.  Measured
@@ -24,10 +24,11 @@
 //      (k,m) pairs — no atomics, plain accumulate-stores.
 //   3. k_update  — SGD apply (p += dt*scale*g) fused with gradient zeroing.
 //
-// Numerics: activations are stored bf16 (or fp32, template) in global
-// memory; ALL arithmetic is fp32 in registers/LDS; parameters, gradients and
-// backward-data tensors are fp32.  Loss-metric semantics match the reference
-// (sum over samples of ||onehot - y||_2).
+// Numerics: activations AND the large conv preact gradient (dz1) are
+// stored bf16/fp16/fp32 (template); ALL arithmetic is fp32 in
+// registers/LDS; parameters, gradients, dz and dz2 are fp32.  Loss-metric
+// semantics match the reference (sum over samples of ||onehot - y||_2,
+// SURVEY.md §0.1 item 3).
 //
 // Wave width is 64 (CDNA4); block size 256 = 4 waves.
 
