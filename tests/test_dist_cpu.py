@@ -13,22 +13,21 @@ from parallel_cnn_amd.data.mnist import synthetic_mnist
 from parallel_cnn_amd.engine.trainer import Trainer
 from parallel_cnn_amd.parallel import dist as pdist
 
-WORLD = 2
 GLOBAL_BATCH = 16
 STEPS = 3
 
 
-def _worker(rank, port, result_q):
+def _worker(rank, world, port, result_q):
     os.environ.update({
         "RANK": str(rank),
         "LOCAL_RANK": str(rank),
-        "WORLD_SIZE": str(WORLD),
+        "WORLD_SIZE": str(world),
         "MASTER_ADDR": "127.0.0.1",
         "MASTER_PORT": str(port),
     })
     ctx = pdist.init_from_env(device="cpu")
     cfg = TrainConfig(backend="cpu", device="cpu",
-                      batch_size=GLOBAL_BATCH // WORLD, log_interval=0)
+                      batch_size=GLOBAL_BATCH // world, log_interval=0)
     t = Trainer(cfg, ctx=ctx)
     x, y = synthetic_mnist(GLOBAL_BATCH * STEPS, seed=0)
     Bl = cfg.batch_size
@@ -41,8 +40,9 @@ def _worker(rank, port, result_q):
     torch.distributed.destroy_process_group()
 
 
-@pytest.mark.timeout(180)
-def test_dp2_matches_single_rank():
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("world,port", [(2, 29531), (4, 29532)])
+def test_dp_matches_single_rank(world, port):
     # single-rank reference trajectory at the same global batch
     cfg = TrainConfig(backend="cpu", device="cpu", batch_size=GLOBAL_BATCH,
                       log_interval=0)
@@ -56,9 +56,8 @@ def test_dp2_matches_single_rank():
 
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29531
-    procs = [ctx.Process(target=_worker, args=(r, port, q))
-             for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, world, port, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     params, loss, n = q.get()
