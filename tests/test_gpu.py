@@ -288,6 +288,24 @@ def test_gpu_modes_match_oracle(pool, loss, device):
         assert torch.equal(s1grad, s1init)  # pool params untouched
 
 
+def test_step_nondeterminism_bounded(device):
+    """Race/nondeterminism detector: two identical steps from identical
+    state may differ only by atomic-ordering roundoff (a data race or
+    missing sync would blow far past this bound)."""
+    x, y = synthetic_mnist(64, seed=41)
+    cfg = TrainConfig(batch_size=64, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0)
+    t1, t2 = Trainer(cfg), Trainer(cfg)
+    xb1, yb1 = t1.stage_batch(x, y)
+    xb2, yb2 = t2.stage_batch(x, y)
+    for _ in range(3):
+        t1.step(xb1, yb1)
+        t2.step(xb2, yb2)
+    torch.cuda.synchronize()
+    diff = (t1.model.params - t2.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+
+
 def test_graph_step_matches_eager(device):
     """hipGraph-captured step replay == eager step trajectory."""
     x, y = synthetic_mnist(64, seed=19)
