@@ -109,8 +109,6 @@ enum Mode { MODE_TRAIN = 0, MODE_EVAL = 1, MODE_INFER = 2 };
 // matrix stays in global memory (L2-resident, read coalesced once per
 // phase).
 struct FwdLds {
-  float ps[OFF_FW];     // staged conv1+pool parameters
-  float xs[IN_PIX];     // input image (fp32)
   float a2s[S1_OUT];    // pool activation
   float ys[FC_OUT];     // logits (post-sigmoid)
   float dzs[FC_OUT];    // residual gradient
@@ -144,8 +142,10 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   if (b >= B) return;
   const int tid = threadIdx.x;
 
-  // ---- phase 0: stage conv/pool parameters and the input image ----
-  if (tid < OFF_FW) L.ps[tid] = params[tid];
+  // No staging phase: each thread loads its own 8x8 input window straight
+  // from global (overlapping windows hit L1; one 1.5 KB image per block)
+  // and the 173 conv/pool parameters broadcast through L1 — the kernel has
+  // TWO barriers ({conv+pool} {fc+residual} {bwd}).
   if (MODE == MODE_TRAIN && wgrad_fuse) {
     if (tid < C1_CH * (C1_K * C1_K + 1))
       L.gw[tid / (C1_K * C1_K + 1)][tid % (C1_K * C1_K + 1)] = 0.f;
@@ -153,8 +153,6 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
       L.gs1[tid - C1_CH * (C1_K * C1_K + 1)] = 0.f;
   }
   const act_t* xb = x + (size_t)b * IN_PIX;
-  for (int i = tid; i < IN_PIX; i += 256) L.xs[i] = ldf(xb + i);
-  __syncthreads();
 
   // ---- phase 1: conv1 + sigmoid + pool + sigmoid (no barrier between) ----
   const int o = tid / S1_PIX;
@@ -164,18 +162,18 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
   float a1v[S1_K * S1_K];  // this cell's conv activations, kept live to bwd
   float a2v = 0.f;
   if (tid < S1_OUT) {
-    // load the cell's 8x8 input window into registers (16 x b128 LDS reads)
+    // load the cell's 8x8 input window into registers (2 x 8B loads/row)
     float xw[8][8];
 #pragma unroll
     for (int u = 0; u < 8; ++u)
 #pragma unroll
       for (int v4 = 0; v4 < 2; ++v4)
-        ld4f(&L.xs[(pr * S1_K + u) * IN_W + pc * S1_K + v4 * 4],
+        ld4f(xb + (pr * S1_K + u) * IN_W + pc * S1_K + v4 * 4,
              &xw[u][v4 * 4]);
-    const float* w = &L.ps[OFF_C1W + o * C1_K * C1_K];
-    const float cb = L.ps[OFF_C1B + o];
+    const float* w = params + OFF_C1W + o * C1_K * C1_K;
+    const float cb = params[OFF_C1B + o];
     // pool preact: trainable weighted sum (reference) or max
-    float pacc = pool_mode == 1 ? -1e30f : L.ps[OFF_S1B];
+    float pacc = pool_mode == 1 ? -1e30f : params[OFF_S1B];
 #pragma unroll
     for (int i = 0; i < S1_K; ++i) {
 #pragma unroll
@@ -191,7 +189,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
         if (pool_mode == 1)
           pacc = fmaxf(pacc, av);
         else
-          pacc += L.ps[OFF_S1W + i * S1_K + j] * av;
+          pacc += params[OFF_S1W + i * S1_K + j] * av;
       }
     }
     if (MODE == MODE_TRAIN) {
@@ -313,7 +311,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
         const float dd =
             pool_mode == 1
                 ? (i * S1_K + j == best ? d2 : 0.f)
-                : d2 * L.ps[OFF_S1W + i * S1_K + j];
+                : d2 * params[OFF_S1W + i * S1_K + j];
         dz1v[i * S1_K + j] = dd * av * (1.0f - av);
       }
       st4f(dz1g + (size_t)b * C1_OUT + o * C1_PIX + (pr * S1_K + i) * C1_W +
@@ -338,7 +336,7 @@ __global__ __launch_bounds__(256) void k_fwdbwd(
         for (int u = 0; u < C1_K; ++u)
 #pragma unroll
           for (int v = 0; v < C1_K; ++v)
-            cacc[u * C1_K + v] += d * L.xs[(r + u) * IN_W + (c + v)];
+            cacc[u * C1_K + v] += d * ldf(xb + (r + u) * IN_W + (c + v));
       }
 #pragma unroll
       for (int w2 = 0; w2 < C1_K * C1_K; ++w2)
