@@ -18,8 +18,17 @@ xp, yp = t.stage_batch(x, y)
 xp, yp = xp.contiguous(), yp.contiguous()
 torch.cuda.synchronize()
 
+def cpp_null_stream(n):
+    # identical C++ enqueue loop, HIP null stream instead of torch's stream
+    w = t.ws
+    t._C.hip_train_steps(xp, yp, t.model.params, t.model.grads, w.a1, w.a2,
+                         w.y, w.dz, w.dz2, w.dz1, w.loss_accum, 64, n,
+                         t.cfg.wgrad_chunk, t.cfg.dt / 64.0, 0, 0, 0)
+
+
 for name, fn in [
     ("cpp-loop", lambda n: t.run_steps_pooled(xp, yp, n)),
+    ("cpp-null-stream", cpp_null_stream),
     ("py-loop", lambda n: [t.step(xp[(s % 64) * 64:(s % 64 + 1) * 64],
                                   yp[(s % 64) * 64:(s % 64 + 1) * 64])
                            for s in range(n)]),
