@@ -12,9 +12,25 @@ USE_TORCH = "--torch" in sys.argv
 if USE_TORCH:
     import torch  # noqa: F401 - presence of the runtime is the variable
     torch.zeros(1, device="cuda")  # force full HIP init
-
-lib = ctypes.CDLL(glob.glob("parallel_cnn_amd/_C.*.so")[0], mode=ctypes.RTLD_GLOBAL)
-hip = ctypes.CDLL("/opt/rocm/lib/libamdhip64.so")
+    sys.path.insert(0, ".")
+    import parallel_cnn_amd._C as _C_mod  # loads with torch's libs
+    lib = ctypes.CDLL(_C_mod.__file__)
+    hip_path = [l.split()[-1] for l in open("/proc/self/maps")
+                if "amdhip64" in l][0]
+    hip = ctypes.CDLL(hip_path)
+else:
+    # load torch's lib deps first so _C resolves, without torch python init
+    for dep in ("libc10.so", "libtorch_cpu.so"):
+        try:
+            ctypes.CDLL("/usr/local/lib/python3.10/dist-packages/torch/lib/"
+                        + dep, mode=ctypes.RTLD_GLOBAL)
+        except OSError:
+            pass
+    lib = ctypes.CDLL(glob.glob("parallel_cnn_amd/_C.*.so")[0],
+                      mode=ctypes.RTLD_GLOBAL)
+    hip_path = [l.split()[-1] for l in open("/proc/self/maps")
+                if "amdhip64" in l][0]
+    hip = ctypes.CDLL(hip_path)
 
 def hipcheck(r):
     assert r == 0, r
