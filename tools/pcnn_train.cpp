@@ -234,6 +234,7 @@ int main(int argc, char** argv) {
 
   const float scale = a.reduction == "sum" ? 1.0f : 1.0f / (float)B;
   printf("Learning\n");
+  int epochs_done = 0;
   auto t0 = std::chrono::steady_clock::now();
   for (int ep = 0; ep < a.epochs; ++ep) {
     for (int s = 0; s < a.train_count; s += B) {
@@ -253,13 +254,17 @@ int main(int argc, char** argv) {
     double secs = std::chrono::duration<double>(
                       std::chrono::steady_clock::now() - t0).count();
     printf("error: %e, time_on_gpu: %f\n", loss_sum / a.train_count, secs);
-    if (loss_sum / a.train_count < a.threshold) break;
+    ++epochs_done;
+    if (loss_sum / a.train_count < a.threshold) break;  // reference early stop
   }
   CHECK(hipDeviceSynchronize());
   double total = std::chrono::duration<double>(
                      std::chrono::steady_clock::now() - t0).count();
   printf("\n Time - %f ms\n", total * 1e3);
-  printf("images/sec: %.0f\n", a.epochs * a.train_count / total);
+  printf("epochs run: %d (early stop at error < %g)\n", epochs_done,
+         a.threshold);
+  printf("images/sec: %.0f\n",
+         (double)epochs_done * a.train_count / total);
 
   if (!a.ckpt_save.empty()) {
     CHECK(hipMemcpy(params_h.data(), d_params, N_PARAMS * 4,
