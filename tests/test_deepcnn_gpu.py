@@ -127,6 +127,22 @@ def test_hip_step_matches_oracle_bf16(device):
     assert abs(lg - loss) < 5e-2 * max(1.0, loss)
 
 
+def test_hip_step_fp16_close(device):
+    B = 8
+    t, x, labels = hip_step_pieces(B, "fp16", device, seed=13)
+    ref = DeepCNN(seed=t.cfg.seed)
+    xh = x.view(B, 32, 32, 3)
+    acts, pouts, y = deep_ref.forward(xh, ref)
+    grads, _ = deep_ref.backward(xh, ref, acts, pouts, y, labels)
+    with torch.no_grad():
+        ref.params += t.cfg.dt * (1.0 / B) * grads
+    delta_hip = t.model.params.cpu() - DeepCNN(seed=t.cfg.seed).params
+    delta_ref = ref.params - DeepCNN(seed=t.cfg.seed).params
+    diff = (delta_hip - delta_ref).abs().max().item()
+    scale = delta_ref.abs().max().item()
+    assert diff < 0.1 * max(1e-3, scale), (diff, scale)
+
+
 def test_hip_eval(device):
     B = 32
     cfg = TrainConfig(batch_size=B, device="cuda", backend="hip",

@@ -247,6 +247,25 @@ def test_dp2_gloo_single_gpu(overlap, device):
     assert r["value"] > 0
 
 
+def test_trainer_fp16_step_close_to_fp32(device):
+    """Trainer-level fp16 activation mode: same trajectory to fp16
+    tolerance."""
+    x, y = synthetic_mnist(32, seed=51)
+    cfg16 = TrainConfig(batch_size=32, device="cuda", backend="hip",
+                        act_dtype="fp16", log_interval=0)
+    cfg32 = TrainConfig(batch_size=32, device="cuda", backend="hip",
+                        act_dtype="fp32", log_interval=0)
+    t16, t32 = Trainer(cfg16), Trainer(cfg32)
+    t16.step(*t16.stage_batch(x, y))
+    t32.step(*t32.stage_batch(x, y))
+    torch.cuda.synchronize()
+    d16 = t16.model.params - LeNet5(device, seed=0).params
+    d32 = t32.model.params - LeNet5(device, seed=0).params
+    diff = (d16 - d32).abs().max().item()
+    scale = d32.abs().max().item()
+    assert diff < 0.05 * max(1e-3, scale), (diff, scale)
+
+
 def test_native_extension_is_loaded_on_gpu(device):
     """The HIP path must be the one that runs (no silent eager fallback)."""
     assert native.available()

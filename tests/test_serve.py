@@ -54,3 +54,15 @@ def test_predict_validation(client):
     assert r.status_code == 400
     r = client.post("/predict", json={"images": [[0.0] * 10]})
     assert r.status_code == 400
+
+
+def test_deepcnn_serving():
+    app = create_app(TrainConfig(device="cpu", backend="torchref",
+                                 model="deepcnn", log_interval=0))
+    c = TestClient(app)
+    assert c.get("/info").json()["input_pixels"] == 32 * 32 * 3
+    from parallel_cnn_amd.data.mnist import synthetic_images
+    x, _ = synthetic_images(2, 32, 32, 3, seed=1)
+    r = c.post("/predict", json={"images": x.tolist()})
+    assert r.status_code == 200
+    assert len(r.json()["labels"]) == 2
