@@ -26,6 +26,7 @@ class TrainConfig:
     # optimizer (reference semantics: p += dt * grad)
     dt: float = REF_DT
     grad_reduction: str = "mean"     # mean | sum over the global batch
+    grad_accum: int = 1              # micro-batches per optimizer step
     threshold: float = REF_THRESHOLD  # early-stop when mean err-norm < this
 
     # schedule

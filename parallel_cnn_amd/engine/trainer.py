@@ -89,7 +89,8 @@ class Trainer:
     # ------------------------------------------------------------------ util
     def _update_scale(self, local_batch: int) -> float:
         if self.cfg.grad_reduction == "mean":
-            return 1.0 / float(local_batch * self.ctx.world_size)
+            return 1.0 / float(local_batch * self.ctx.world_size *
+                               self.cfg.grad_accum)
         return 1.0
 
     def stage_batch(self, x: torch.Tensor, labels: torch.Tensor
@@ -111,11 +112,18 @@ class Trainer:
 
     # ------------------------------------------------------------------ step
     def step(self, x: torch.Tensor, labels: torch.Tensor) -> None:
-        """One training step on an already-staged batch."""
+        """One micro/step on an already-staged batch.  The weight-grad
+        kernels ACCUMULATE into the flat bucket, so gradient accumulation
+        (cfg.grad_accum > 1) simply defers the all-reduce + update to every
+        grad_accum-th call."""
         if self.timers is not None:
             return self._step_profiled(x, labels)
         B = x.shape[0]
         assert B <= self.ws.max_batch
+        self._accum = getattr(self, "_accum", 0) + 1
+        apply_update = self._accum >= self.cfg.grad_accum
+        if apply_update:
+            self._accum = 0
         m, w = self.model, self.ws
         scale = self._update_scale(B)
         if self.backend == "hip":
@@ -124,7 +132,11 @@ class Trainer:
                                w.dz1, labels, w.loss_accum, w.correct_accum,
                                B, MODE_TRAIN, stream, self._pool_mode,
                                self._loss_mode, m.grads, self._fuse)
-            if self.cfg.overlap_comm and self.ctx.world_size > 1:
+            if not apply_update:
+                self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
+                                        m.grads, B, self.cfg.wgrad_chunk,
+                                        self._wroles, stream)
+            elif self.cfg.overlap_comm and self.ctx.world_size > 1:
                 # two-bucket overlap (SURVEY §5.8 / north star): the first
                 # bucket all-reduces on the RCCL stream while the rest of
                 # the wgrad still computes on the compute stream
@@ -158,8 +170,9 @@ class Trainer:
                                         m.grads, B, self.cfg.wgrad_chunk,
                                         self._wroles, stream)
                 pdist.allreduce_grads(m.grads)
-            self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
-                               stream)
+            if apply_update:
+                self._C.hip_update(m.params, m.grads, self.cfg.dt * scale,
+                                   stream)
         elif self.backend == "cpu":
             # CPU path keeps fp32 activations in the workspace directly.
             a1 = self._cpu_view(w.a1, B)
@@ -171,8 +184,9 @@ class Trainer:
                                         m.grads, self._pool_mode,
                                         self._loss_mode)
             self._loss_host += loss
-            pdist.allreduce_grads(m.grads)
-            self._C.cpu_update(m.params, m.grads, self.cfg.dt, scale)
+            if apply_update:
+                pdist.allreduce_grads(m.grads)
+                self._C.cpu_update(m.params, m.grads, self.cfg.dt, scale)
         else:  # torchref
             a1, a2, y = torch_ref.forward(x, m.params, self.cfg.pool,
                                           self.cfg.loss)
@@ -180,8 +194,9 @@ class Trainer:
                 x, m.params, a1, a2, y, labels, self.cfg.pool, self.cfg.loss)
             self._loss_host += loss
             m.grads += grads
-            pdist.allreduce_grads(m.grads)
-            torch_ref.update(m.params, m.grads, self.cfg.dt, scale)
+            if apply_update:
+                pdist.allreduce_grads(m.grads)
+                torch_ref.update(m.params, m.grads, self.cfg.dt, scale)
         self._samples_seen += B * self.ctx.world_size
         self.global_step += 1
 

@@ -247,6 +247,21 @@ def test_dp2_gloo_single_gpu(overlap, device):
     assert r["value"] > 0
 
 
+def test_grad_accumulation_gpu_matches_big_batch(device):
+    x, y = synthetic_mnist(32, seed=61)
+    cfg_a = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                        act_dtype="fp32", log_interval=0, grad_accum=2)
+    cfg_b = TrainConfig(batch_size=32, device="cuda", backend="hip",
+                        act_dtype="fp32", log_interval=0)
+    ta, tb = Trainer(cfg_a), Trainer(cfg_b)
+    ta.step(*ta.stage_batch(x[:16], y[:16]))
+    ta.step(*ta.stage_batch(x[16:], y[16:]))
+    tb.step(*tb.stage_batch(x, y))
+    torch.cuda.synchronize()
+    diff = (ta.model.params - tb.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+
+
 def test_trainer_fp16_step_close_to_fp32(device):
     """Trainer-level fp16 activation mode: same trajectory to fp16
     tolerance."""

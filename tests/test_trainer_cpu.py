@@ -87,3 +87,19 @@ def test_early_stop_threshold():
     x, y = synthetic_mnist(32, seed=6)
     err = t.train_epoch(x, y, log=lambda *a: None)
     assert err < 1e9
+
+
+def test_grad_accumulation_matches_big_batch():
+    """grad_accum=2 over two bs=8 micro-batches == one bs=16 step with
+    mean reduction (the wgrad path accumulates; update scale folds in
+    the accumulation count)."""
+    from parallel_cnn_amd.data.mnist import synthetic_mnist as sm
+    x, y = sm(16, seed=8)
+    t_acc = make_trainer("cpu", batch_size=8, grad_accum=2)
+    t_big = make_trainer("cpu", batch_size=16)
+    t_acc.step(*t_acc.stage_batch(x[:8], y[:8]))
+    t_acc.step(*t_acc.stage_batch(x[8:], y[8:]))
+    t_big.step(*t_big.stage_batch(x, y))
+    assert torch.allclose(t_acc.model.params, t_big.model.params, atol=1e-6)
+    # grads were consumed at the boundary
+    assert t_acc.model.grads.abs().sum() == 0
