@@ -108,6 +108,8 @@ class Trainer:
         """Per-phase sync-correct timers (the framework analog of the
         reference's per-layer clock() accumulators, but device-synced).
         Adds a device sync per phase — profiling mode only."""
+        if self.cfg.grad_accum != 1:
+            raise RuntimeError("profiling mode assumes grad_accum == 1")
         self.timers = PhaseTimers()
 
     # ------------------------------------------------------------------ step
@@ -238,6 +240,8 @@ class Trainer:
         not capturable)."""
         if self.backend != "hip":
             raise RuntimeError("graph capture requires the hip backend")
+        if self.cfg.grad_accum != 1:
+            raise RuntimeError("graph capture assumes grad_accum == 1")
         if self.ctx.world_size > 1 and \
                 torch.distributed.get_backend() != "nccl":
             raise RuntimeError("graph capture requires RCCL (nccl backend)")
