@@ -69,3 +69,49 @@ def test_dp_matches_single_rank(world, port):
     assert abs(loss - ref_loss) < 1e-3
     assert torch.allclose(params, ref.model.params, atol=1e-5), \
         (params - ref.model.params).abs().max()
+
+def _accum_worker(rank, world, port, result_q):
+    os.environ.update({
+        "RANK": str(rank),
+        "LOCAL_RANK": str(rank),
+        "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    ctx = pdist.init_from_env(device="cpu")
+    # 2 ranks x bs=4 x grad_accum=2 == one global-batch-16 step
+    cfg = TrainConfig(backend="cpu", device="cpu", batch_size=4,
+                      grad_accum=2, log_interval=0)
+    t = Trainer(cfg, ctx=ctx)
+    x, y = synthetic_mnist(16, seed=0)
+    for micro in range(2):
+        lo = micro * 8 + rank * 4
+        t.step(*t.stage_batch(x[lo:lo + 4], y[lo:lo + 4]))
+    if rank == 0:
+        result_q.put(t.model.params.clone())
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp_with_grad_accumulation_matches_big_batch():
+    """DP(2) x grad_accum(2) at bs=4 == one single-rank bs=16 step: the
+    all-reduce must fire only at the accumulation boundary and the update
+    scale must fold in world*accum*micro_batch."""
+    cfg = TrainConfig(backend="cpu", device="cpu", batch_size=16,
+                      log_interval=0)
+    ref = Trainer(cfg)
+    x, y = synthetic_mnist(16, seed=0)
+    ref.step(*ref.stage_batch(x, y))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_accum_worker, args=(r, 2, 29533, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    params = q.get()
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert torch.allclose(params, ref.model.params, atol=1e-6), \
+        (params - ref.model.params).abs().max()
