@@ -780,6 +780,86 @@ __global__ __launch_bounds__(256) void k_pool_wgrad(
                     red[0][16] + red[1][16] + red[2][16] + red[3][16]);
 }
 
+// Vectorized pool wgrad for C % 8 == 0 (every DeepCNN config): each
+// thread owns 8 consecutive channels, so all K*K+1 loads per item are
+// 16B ld8v instead of scalar 2B loads (scalar bf16 access is
+// address-issue-bound on CDNA4 — this is the k_pool_fwd idiom applied
+// to the reduction).  acc[w] folds the 8 channels immediately (the pool
+// weight is shared across channels).  Wave + LDS pre-reduce, 17 atomics
+// per workgroup, as in the scalar kernel above.
+template <typename act_t, int KK>
+__global__ __launch_bounds__(256) void k_pool_wgrad8(
+    const act_t* __restrict__ dppre, const act_t* __restrict__ a,
+    float* __restrict__ dpw, int B, int H, int W, int C, int G) {
+  const int OH = H / KK, OW = W / KK;
+  const int C8 = C / 8;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  float acc[KK * KK];
+#pragma unroll
+  for (int w = 0; w < KK * KK; ++w) acc[w] = 0.f;
+  float bacc = 0.f;
+  const long long N8 = (long long)B * OH * OW * C8;
+  const long long stride = (long long)G * 256;
+
+  auto decode_load = [&](long long it, float* d, float (*av)[8]) {
+    const int c0 = (int)(it % C8) * 8;
+    long long t = it / C8;
+    const int q = (int)(t % OW);
+    t /= OW;
+    const int p = (int)(t % OH);
+    const int b = (int)(t / OH);
+    ld8v(dppre + (((long long)b * OH + p) * OW + q) * C + c0, d);
+#pragma unroll
+    for (int i = 0; i < KK; ++i)
+#pragma unroll
+      for (int j = 0; j < KK; ++j)
+        ld8v(a + (((long long)b * H + p * KK + i) * W + q * KK + j) * C + c0,
+             av[i * KK + j]);
+  };
+
+  for (long long it = (long long)blockIdx.x * 256 + tid; it < N8;
+       it += 2 * stride) {
+    float d1[8], av1[KK * KK][8], d2[8], av2[KK * KK][8];
+    decode_load(it, d1, av1);
+    const long long it2 = it + stride;
+    if (it2 < N8) decode_load(it2, d2, av2);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) bacc += d1[u];
+#pragma unroll
+    for (int w = 0; w < KK * KK; ++w)
+#pragma unroll
+      for (int u = 0; u < 8; ++u) acc[w] += d1[u] * av1[w][u];
+    if (it2 < N8) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) bacc += d2[u];
+#pragma unroll
+      for (int w = 0; w < KK * KK; ++w)
+#pragma unroll
+        for (int u = 0; u < 8; ++u) acc[w] += d2[u] * av2[w][u];
+    }
+  }
+  __shared__ float red[4][KK * KK + 1];
+  const int wv = tid >> 6;
+#pragma unroll
+  for (int w = 0; w < KK * KK; ++w) {
+    float v = acc[w];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][w] = v;
+  }
+  {
+    float v = bacc;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][KK * KK] = v;
+  }
+  __syncthreads();
+  if (tid <= KK * KK)
+    unsafeAtomicAdd(&dpw[tid],
+                    red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid]);
+}
+
 // ---------------------------------------------------------------------------
 // FC head (FCIN -> 10) + residual loss, general fan-in.  One block per
 // sample; 16 lanes per class with shuffle reduction (the LeNet pattern).
@@ -1085,6 +1165,17 @@ int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
                          int H, int W, int C, int K, int G, int actf,
                          void* stream) {
   dim3 grid(G), block(256);
+  if (C % 8 == 0 && K == 2) {
+    // vectorized path: 16B ld8v over 8 consecutive channels per thread.
+    // (K=3/4 instantiations need 200+ VGPRs with the 2-in-flight pipeline
+    // — occupancy 1 — so larger pools keep the scalar kernel.)
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_wgrad8<act_t, 2>), grid,
+                                            block, 0, (hipStream_t)stream,
+                                            (const act_t*)dppre,
+                                            (const act_t*)a, dpw, B, H, W, C,
+                                            G));
+    return (int)hipGetLastError();
+  }
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_wgrad<act_t>), grid, block,
                                           0, (hipStream_t)stream,
                                           (const act_t*)dppre,
