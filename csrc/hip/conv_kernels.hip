@@ -84,72 +84,91 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
 // One thread per cols element; kc is the fast axis (coalesced stores, and
 // coalesced loads since ci is the fast axis of NHWC x).
 // ---------------------------------------------------------------------------
-// Row-piece formulation: one thread owns an 8-element span of a kernel row
-// (kc = i*K*Cin + t, t in [0, K*Cin)).  Within one kernel row the source
-// addresses are CONTIGUOUS in NHWC x (x row ih, starting at column ow-P),
-// so the whole span is one 16B load + one 16B store, with scalar clipping
-// only at image borders / row boundaries.  Index decode happens once per
-// thread (the per-element version was 83% VALUBusy on index math).
+// Span-walk formulation: one thread owns FOUR consecutive 8-element
+// spans (32 elements, KcP % 32 == 0 by construction) of one cols row.
+// The (b, oh, ow) decode and the kc -> (i, j, ci) decode happen ONCE;
+// the walk across spans is incremental (no divisions).  For Cin % 8 == 0
+// a span always lies inside one input pixel (ci % 8 == 0 and 8 <= Cin),
+// so it is a single 16B load; smaller Cin walks elementwise with
+// incremental counters (the per-element div/mod version was 83% VALUBusy
+// on index math).
 template <typename act_t>
 __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
                          int B, int H, int W, int Cin, int K, int P,
                          int KcP) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long long total = ((long long)B * H * W * KcP) / 8;
+  const long long total = ((long long)B * H * W * KcP) / 32;
   if (idx >= total) return;
-  const int kc0 = (int)((idx * 8) % KcP);
-  const long long m = (idx * 8) / KcP;
+  int kc = (int)((idx * 32) % KcP);
+  const long long m = (idx * 32) / KcP;
   const int Kc = K * K * Cin;
-  const int rowc = K * Cin;  // elements per kernel row
+  const int rowc = K * Cin;
   const int ow = (int)(m % W);
   const long long bh = m / W;
   const int oh = (int)(bh % H);
   const int b = (int)(bh / H);
+  const act_t* __restrict__ xb = x + (long long)b * H * W * Cin;
+  act_t* __restrict__ orow = cols + m * KcP;
 
-  act_t out[8];
-  bool fast = false;
-  if (kc0 + 8 <= Kc && (kc0 % rowc) + 8 <= rowc && (Cin % 8) == 0) {
-    const int i = kc0 / rowc;
-    const int ih = oh + i - P;
-    const int t0 = kc0 - i * rowc;       // offset within the kernel row
-    // span covers x row ih, columns iw = ow - P + t/Cin, elements
-    // contiguous from ((b,ih,ow-P))*Cin + t0
-    const int j0 = t0 / Cin;
-    const int j7 = (t0 + 7) / Cin;
-    const int iw0 = ow + j0 - P;
-    const int iw7 = ow + j7 - P;
-    if (ih >= 0 && ih < H && iw0 >= 0 && iw7 < W) {
-      float v8[8];
-      ld8v(x + (((long long)b * H + ih) * W + (ow - P)) * Cin + t0, v8);
+  // span-0 decode — the only divisions in the kernel
+  int i = kc / rowc;
+  int t = kc - i * rowc;
+  int j = t / Cin;
+  int ci = t - j * Cin;
+  const bool vec = (Cin % 8) == 0;
+
+  for (int s = 0; s < 4; ++s, kc += 8) {
+    act_t out[8];
+    if (kc >= Kc) {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
-      fast = true;
-    }
-  }
-  if (!fast) {
+      for (int u = 0; u < 8; ++u) out[u] = (act_t)0.f;
+    } else if (vec) {
+      const int ih = oh + i - P;
+      const int iw = ow + j - P;
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W) {
+        float v8[8];
+        ld8v(xb + ((long long)ih * W + iw) * Cin + ci, v8);
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int kc = kc0 + u;
-      float val = 0.f;
-      if (kc < Kc) {
-        const int i = kc / rowc;
-        const int t = kc - i * rowc;
-        const int j = t / Cin;
-        const int ci = t - j * Cin;
-        const int ih = oh + i - P;
-        const int iw = ow + j - P;
-        if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-          val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
+        for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) out[u] = (act_t)0.f;
       }
-      out[u] = (act_t)val;
-    }
-  }
-  if (sizeof(act_t) == 2)
-    *reinterpret_cast<uint4*>(cols + m * KcP + kc0) =
-        *reinterpret_cast<const uint4*>(out);
-  else
+      ci += 8;
+      if (ci >= Cin) {
+        ci = 0;
+        if (++j >= K) {
+          j = 0;
+          ++i;
+        }
+      }
+    } else {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) cols[m * KcP + kc0 + u] = out[u];
+      for (int u = 0; u < 8; ++u) {
+        float val = 0.f;
+        if (kc + u < Kc) {
+          const int ih = oh + i - P;
+          const int iw = ow + j - P;
+          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+            val = ldf(xb + ((long long)ih * W + iw) * Cin + ci);
+        }
+        out[u] = (act_t)val;
+        if (++ci >= Cin) {
+          ci = 0;
+          if (++j >= K) {
+            j = 0;
+            ++i;
+          }
+        }
+      }
+    }
+    if (sizeof(act_t) == 2)
+      *reinterpret_cast<uint4*>(orow + kc) =
+          *reinterpret_cast<const uint4*>(out);
+    else
+#pragma unroll
+      for (int u = 0; u < 8; ++u) orow[kc + u] = out[u];
+  }
 }
 
 // 8-wide im2col gather for implicit-GEMM staging: returns cols[m][kc0..+8)
@@ -556,6 +575,17 @@ __global__ __launch_bounds__(256) void k_colsum(const act_t* __restrict__ dpre,
                                                     // 2048%N==0 for N<=256)
   long long flat = ((long long)blockIdx.x * 256 + tid) * 8;
   float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  // 4 independent streams in flight: a single-stream loop is a load
+  // latency chain at the grid sizes this runs at (~1 workgroup/CU).
+  for (; flat + 3 * stride + 8 <= total; flat += 4 * stride) {
+    float v[4][8];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) ld8v(dpre + flat + q * stride, v[q]);
+#pragma unroll
+    for (int q = 0; q < 4; ++q)
+#pragma unroll
+      for (int u = 0; u < 8; ++u) acc[u] += v[q][u];
+  }
   for (; flat + 8 <= total; flat += stride) {
     float v8[8];
     ld8v(dpre + flat, v8);
@@ -1031,7 +1061,7 @@ extern "C" {
 
 int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
                      int K, int P, int KcP, int actf, void* stream) {
-  const long long total = (long long)B * H * W * KcP / 8;  // 8 elems/thread
+  const long long total = (long long)B * H * W * KcP / 32;  // 32 elems/thread
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL(
                           (k_im2col<act_t>), grid, block, 0,

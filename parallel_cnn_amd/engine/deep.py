@@ -172,7 +172,10 @@ class DeepTrainer:
             dapre = w.acts[i]
             ktiles = (st.kcp + 63) // 64
             ntiles = (st.cout + 63) // 64
-            ms = max(1, min(128, 512 // (ktiles * ntiles)))
+            # ~512 total workgroups is the measured sweet spot for the
+            # atomic-combine cost (tools/deep_sweep.py); the old 128 cap
+            # starved the small-KcP stage (88us -> 54us at ms=256)
+            ms = max(1, min(256, 512 // (ktiles * ntiles)))
             x_in = x if i == 0 else w.pouts[i - 1]
             implicit = False
             self._C.deep_wgrad_gemm(w.cols[i], dapre,
