@@ -66,7 +66,8 @@ int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                          long long M, int KcP, int N, int MS, int actf,
                          void* stream);
-int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
+int pcnn_deep_colsum(const void* dpre, float* part, float* db, long long M,
+                     int N,
                      int slices, int actf, void* stream);
 int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
                             int B, int H, int W, int Cin, int K, int P,
@@ -271,11 +272,12 @@ void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
             "deep_wgrad_gemm");
 }
 
-void deep_colsum(at::Tensor dpre, at::Tensor db, int64_t M, int64_t N,
-                 int64_t slices, int64_t stream) {
-  check_hip(pcnn_deep_colsum(dpre.data_ptr(), db.data_ptr<float>(), M,
-                             (int)N, (int)slices, act_flag(dpre),
-                             (void*)stream),
+void deep_colsum(at::Tensor dpre, at::Tensor part, at::Tensor db, int64_t M,
+                 int64_t N, int64_t slices, int64_t stream) {
+  TORCH_CHECK(part.numel() >= slices * N, "colsum scratch too small");
+  check_hip(pcnn_deep_colsum(dpre.data_ptr(), part.data_ptr<float>(),
+                             db.data_ptr<float>(), M, (int)N, (int)slices,
+                             act_flag(dpre), (void*)stream),
             "deep_colsum");
 }
 

@@ -567,7 +567,7 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 // the block tree-reduces per column at the end.
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_colsum(const act_t* __restrict__ dpre,
-                                                float* __restrict__ db,
+                                                float* __restrict__ part,
                                                 long long M, int N, int G) {
   const int tid = threadIdx.x;
   const long long total = M * N;
@@ -604,10 +604,49 @@ __global__ __launch_bounds__(256) void k_colsum(const act_t* __restrict__ dpre,
     }
     __syncthreads();
   }
+  // per-workgroup partials to scratch (atomics to the handful of db
+  // cache lines serialize globally at ~3ns/op — measured to dominate
+  // this kernel at any G; k_colsum_fin combines the partials instead)
   if (tid < N / 8) {
+    float* dst = part + (long long)blockIdx.x * N + tid * 8;
 #pragma unroll
-    for (int u = 0; u < 8; ++u)
-      unsafeAtomicAdd(&db[tid * 8 + u], sred[tid][u]);
+    for (int u = 0; u < 8; ++u) dst[u] = sred[tid][u];
+  }
+}
+
+// Combine the [G][N] column partials into db (+=).  One workgroup: 256/N
+// lanes per column sum a G/L stride each (4 streams in flight), then a
+// small LDS fold.  N > 256 falls back to one thread per column.
+__global__ __launch_bounds__(256) void k_colsum_fin(
+    const float* __restrict__ part, float* __restrict__ db, int N, int G) {
+  const int tid = threadIdx.x;
+  if (N >= 256) {
+    for (int n = tid; n < N; n += 256) {
+      float t = 0.f;
+      for (int g = 0; g < G; ++g) t += part[(long long)g * N + n];
+      db[n] += t;
+    }
+    return;
+  }
+  const int L = 256 / N;  // lanes per column (N % 8 == 0)
+  const int n = tid % N;
+  const int sl = tid / N;
+  float s = 0.f;
+  if (sl < L) {
+    int g = sl;
+    for (; g + 3 * L < G; g += 4 * L)
+      s += part[(long long)g * N + n] + part[(long long)(g + L) * N + n] +
+           part[(long long)(g + 2 * L) * N + n] +
+           part[(long long)(g + 3 * L) * N + n];
+    for (; g < G; g += L) s += part[(long long)g * N + n];
+  }
+  __shared__ float red[256];
+  red[tid] = s;
+  __syncthreads();
+  if (tid < N) {
+    float tot = 0.f;
+    for (int s2 = 0; s2 < L; ++s2) tot += red[tid + s2 * N];
+    db[tid] += tot;
   }
 }
 
@@ -1144,13 +1183,15 @@ int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                                  0, 0, 0, 0, actf, stream);
 }
 
-int pcnn_deep_colsum(const void* dpre, float* db, long long M, int N,
-                     int slices, int actf, void* stream) {
+int pcnn_deep_colsum(const void* dpre, float* part, float* db, long long M,
+                     int N, int slices, int actf, void* stream) {
   dim3 grid((unsigned)slices), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_colsum<act_t>), grid, block, 0,
                                           (hipStream_t)stream,
-                                          (const act_t*)dpre, db, M, N,
+                                          (const act_t*)dpre, part, M, N,
                                           slices));
+  hipLaunchKernelGGL(k_colsum_fin, dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, part, db, N, slices);
   return (int)hipGetLastError();
 }
 

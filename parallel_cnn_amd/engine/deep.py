@@ -54,6 +54,10 @@ class DeepWorkspace:
                              device=device)
         self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
                               device=device)
+        # colsum per-workgroup partials (up to 512 slices x widest stage)
+        maxc = max(st.cout for st in spec.stages)
+        self.colsum_part = torch.empty(512 * maxc, dtype=torch.float32,
+                                       device=device)
         self.loss_accum = torch.zeros(1, dtype=torch.float32, device=device)
         self.correct_accum = torch.zeros(1, dtype=torch.int32, device=device)
 
@@ -160,7 +164,10 @@ class DeepTrainer:
         for i in range(nstage - 1, -1, -1):
             st = spec.stages[i]
             M = B * st.h * st.w
-            G = max(8, min(512, (B * st.oh * st.ow * st.cout) // (256 * 16)))
+            # ~halved from the 2-items/thread target: the 17 per-WG
+            # atomics land on one cache line (~3ns/op serialized), so the
+            # sweet spot trades compute depth against the atomic tail
+            G = max(64, min(512, (B * st.oh * st.ow * st.cout) // (256 * 32)))
             self._C.deep_pool_wgrad(w.dppre[i], w.acts[i],
                                     m.grad_view(f"pool{i}_w"), B, st.h, st.w,
                                     st.cout, st.pool_k, G, st_h)
@@ -183,9 +190,10 @@ class DeepTrainer:
                                     st.cout, ms, st_h,
                                     x_in if implicit else torch.empty(0),
                                     st.h, st.w, st.cin, st.k, st.pad)
-            gsum = max(32, min(512, (M * st.cout) // (256 * 96)))
-            self._C.deep_colsum(dapre, m.grad_view(f"conv{i}_b"), M,
-                                st.cout, gsum, st_h)
+            gsum = max(32, min(128, (M * st.cout) // (256 * 96)))
+            self._C.deep_colsum(dapre, w.colsum_part,
+                                m.grad_view(f"conv{i}_b"), M, st.cout, gsum,
+                                st_h)
             if i > 0:
                 # dgrad into the cols buffer (its forward use is done)
                 self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),

@@ -61,11 +61,12 @@ def main():
         M = B * h * w
         dpre = torch.randn(M, cout, device=dev).to(torch.bfloat16)
         db = torch.zeros(cout, device=dev)
+        part = torch.empty(512 * cout, device=dev)
         base = max(32, min(512, (M * cout) // (256 * 96)))
         row = []
-        for g in sorted({base, 128, 256, 512, 1024, 2048, 4096}):
-            us = t_kernel(lambda st, g=g: _C.deep_colsum(dpre, db, M, cout,
-                                                         g, st))
+        for g in sorted({base, 64, 128, 256, 512}):
+            us = t_kernel(lambda st, g=g: _C.deep_colsum(dpre, part, db, M,
+                                                         cout, g, st))
             row.append((g, us))
         best = min(row, key=lambda p: p[1])
         print("stage%d M=%6d N=%2d cur(G=%d)=%.1fus best(G=%d)=%.1fus" %
@@ -73,5 +74,28 @@ def main():
         print("   ", " ".join("G=%d:%.1f" % p for p in row))
 
 
+def pool_sweep():
+    dev = "cuda:0"
+    B = 256
+    print("== k_pool_wgrad8 G sweep (bs=%d) ==" % B)
+    for si, (h, w, cout) in enumerate([(32, 32, 32), (16, 16, 64),
+                                       (8, 8, 64)]):
+        dppre = torch.randn(B * (h // 2) * (w // 2) * cout,
+                            device=dev).to(torch.bfloat16)
+        acts = torch.rand(B * h * w * cout, device=dev).to(torch.bfloat16)
+        dpw = torch.zeros(5, device=dev)
+        base = max(8, min(512, (B * (h // 2) * (w // 2) * cout) // (256 * 16)))
+        row = []
+        for g in sorted({base, 32, 64, 128, 256, 512}):
+            us = t_kernel(lambda st, g=g: _C.deep_pool_wgrad(
+                dppre, acts, dpw, B, h, w, cout, 2, g, st))
+            row.append((g, us))
+        best = min(row, key=lambda p: p[1])
+        print("stage%d cur(G=%d)=%.1fus best(G=%d)=%.1fus" %
+              (si, base, dict(row)[base], best[0], best[1]))
+        print("   ", " ".join("G=%d:%.1f" % p for p in row))
+
+
 if __name__ == "__main__":
     main()
+    pool_sweep()
