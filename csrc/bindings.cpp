@@ -58,14 +58,14 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       const void* imx, int XH, int XW, int XC, int XK,
                       int XP, int actf, void* stream);
 int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
-                            long long M, int KcP, int N, int MS,
+                            float* part, long long M, int KcP, int N, int MS,
                             const void* imx, int XH, int XW, int XC, int XK,
                             int XP, int actf, void* stream);
 int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
                       void* stream);
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
-                         long long M, int KcP, int N, int MS, int actf,
-                         void* stream);
+                         float* part, long long M, int KcP, int N, int MS,
+                         int actf, void* stream);
 int pcnn_deep_colsum(const void* dpre, float* part, float* db, long long M,
                      int N,
                      int slices, int actf, void* stream);
@@ -263,9 +263,14 @@ void deep_cast_wt(at::Tensor W, at::Tensor out, at::Tensor outT, int64_t R,
 void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
                      int64_t M, int64_t KcP, int64_t N, int64_t MS,
                      int64_t stream, at::Tensor imx, int64_t XH, int64_t XW,
-                     int64_t XC, int64_t XK, int64_t XP) {
+                     int64_t XC, int64_t XK, int64_t XP, at::Tensor part) {
+  float* pp = nullptr;
+  if (part.numel()) {
+    TORCH_CHECK(part.numel() >= MS * KcP * N, "wgrad slab scratch too small");
+    pp = part.data_ptr<float>();
+  }
   check_hip(pcnn_deep_wgrad_gemm_ex(
-                cols.data_ptr(), dpre.data_ptr(), dW.data_ptr<float>(), M,
+                cols.data_ptr(), dpre.data_ptr(), dW.data_ptr<float>(), pp, M,
                 (int)KcP, (int)N, (int)MS,
                 imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
                 (int)XC, (int)XK, (int)XP, act_flag(cols), (void*)stream),
@@ -418,7 +423,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("N"), py::arg("MS"), py::arg("stream"),
         py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
         py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
-        py::arg("XP") = 0);
+        py::arg("XP") = 0, py::arg("part") = at::empty({0}));
   m.def("deep_colsum", &deep_colsum);
   m.def("deep_col2im_sigbwd", &deep_col2im_sigbwd);
   m.def("deep_pool_fwd", &deep_pool_fwd);

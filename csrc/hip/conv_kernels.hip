@@ -217,17 +217,19 @@ __device__ __forceinline__ void im2col8(const act_t* __restrict__ x, int H,
 //   C[16x16]: lane l, reg r (0..3):  C[row = (l>>4)*4 + r][col = l&15]
 // ---------------------------------------------------------------------------
 
-constexpr int BM = 64, BN = 64, BK = 64;  // 64-deep K-step: 2 MFMAs per
-                                          // fragment per barrier pair
-constexpr int LDP = BK + 8;  // LDS row stride (bf16) — conflict-free b128
+constexpr int BM = 64, BN = 64, BK = 64;   // wgrad M-chunk per barrier pair
+constexpr int BKC = 64;  // k_gemm K-step (measured: 128 halves the
+                         // iteration count but costs 187 VGPRs -> occupancy
+                         // 2 waves/SIMD and loses 15% end-to-end; 64 keeps
+                         // 68 VGPRs / occupancy 5)
 
-template <int TBM>
+template <int TBM, int KD>
 struct GemmLdsT {
-  __bf16 As[TBM][LDP];  // A tile, [m][k]
-  __bf16 Bs[BN][LDP];   // B tile, [n][k]  (transposed image: frag reads
-                        // are contiguous along k for both operands)
+  __bf16 As[TBM][KD + 8];  // A tile, [m][k]; +8 rows: conflict-free b128
+  __bf16 Bs[BN][KD + 8];   // B tile, [n][k]  (transposed image: frag reads
+                           // are contiguous along k for both operands)
 };
-using GemmLds = GemmLdsT<BM>;
+using GemmLds = GemmLdsT<BM, BK>;
 
 // Load an 8-element bf16 fragment from an LDS row.
 __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
@@ -256,11 +258,13 @@ __global__ __launch_bounds__(256) void k_gemm(
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
     int XC, int XK, int XP) {
-  __shared__ GemmLdsT<TBM> L;
+  __shared__ GemmLdsT<TBM, BKC> L;
   constexpr int RF = TBM / 64;        // row fragments per wave
   constexpr int TPR = 256 / TBM;      // staging threads per A row
-  constexpr int SPAN = 64 / TPR;      // k-span per staging thread
+  constexpr int SPAN = BKC / TPR;     // k-span per staging thread
   constexpr int NCH = SPAN / 8;       // 8-chunks per staging thread
+  constexpr int SPANB = BKC / 4;      // k-span per B staging thread
+  constexpr int NCHB = SPANB / 8;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -279,12 +283,12 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   const int row_a = tid / TPR;
   const int kq = (tid % TPR) * SPAN;
-  const int bkq = (tid & 3) * 16;     // B staging: 4 threads per row
+  const int bkq = (tid & 3) * SPANB;  // B staging: 4 threads per row
   // T14-style software pipeline: next tile's global loads are issued into
   // registers BEFORE the MFMA block; the LDS write happens after the read
   // barrier.
   float ra[NCH][8];
-  float rb[2][8];
+  float rb[NCHB][8];
   const long long m_a = m0 + row_a;
   // implicit-A: decode this thread's im2col row position once
   int ib = 0, ioh = 0, iow = 0, iKc = 0;
@@ -300,48 +304,66 @@ __global__ __launch_bounds__(256) void k_gemm(
 #pragma unroll
     for (int h = 0; h < NCH; ++h) {
       const int kk = kq + h * 8;
+      const bool ok = m_a < M && (kt + kk) < K;
+      if (imx != nullptr) {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
-      if (m_a < M && (kt + kk) < K) {
-        if (imx != nullptr)
+        for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
+        if (ok)
           im2col8(imx, XH, XW, XC, XK, XP, ib, ioh, iow, kt + kk, iKc,
                   ra[h]);
-        else
-          ld8v(A + m_a * ldA + kt + kk, ra[h]);
+      } else {
+        // clamped unconditional load + select-to-zero: a load behind a
+        // thread-varying guard de-pipelines the whole K-loop (hipcc
+        // branches around it; CDNA4 lesson from the implicit-im2col
+        // experiment).  K % 8 == 0 and K <= ldA keep the clamp in-bounds.
+        const long long mm = m_a < M ? m_a : M - 1;
+        const int kcl = (kt + kk) < K ? (kt + kk) : K - 8;
+        ld8v(A + mm * ldA + kcl, ra[h]);
+        if (!ok) {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
+        }
       }
     }
     if (Bpre != nullptr) {
-      // bf16 [N][K] rows: two vector loads per thread
+      // bf16 [N][K] rows: two clamped unconditional vector loads per
+      // thread, zeros selected in afterwards (see the A-path note)
       const int n = tid >> 2;
-      const bool ok = (n0 + n) < N;
+      const int nn = (n0 + n) < N ? (n0 + n) : N - 1;
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
+      for (int h = 0; h < NCHB; ++h) {
         const int kk = bkq + h * 8;
+        const int kcl = (kt + kk) < K ? (kt + kk) : K - 8;
+        ld8v(reinterpret_cast<const bf16*>(Bpre + (long long)nn * K + kcl),
+             rb[h]);
+        if (!((n0 + n) < N && (kt + kk) < K)) {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) rb[h][u] = 0.f;
-        if (ok && (kt + kk) < K)
-          ld8v(reinterpret_cast<const bf16*>(
-                   Bpre + (long long)(n0 + n) * K + kt + kk),
-               rb[h]);
+          for (int u = 0; u < 8; ++u) rb[h][u] = 0.f;
+        }
       }
     } else if (b_kxn) {
-      // Bsrc[K][N]: read rows k (coalesced along n), write transposed
-      const int k = tid >> 2;
-      const int nq = (tid & 3) * 16;
-      const bool kok = (kt + k) < K;
+      // Bsrc[K][N]: read rows k (coalesced along n), write transposed;
+      // each thread covers BKC/64 k-rows 64 apart
 #pragma unroll
-      for (int h = 0; h < 2; ++h)
+      for (int kb = 0; kb < BKC / 64; ++kb) {
+        const int k = (tid >> 2) + kb * 64;
+        const int nq = (tid & 3) * 16;
+        const bool kok = (kt + k) < K;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int n = n0 + nq + h * 8 + u;
-          rb[h][u] = (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
-        }
+        for (int h = 0; h < 2; ++h)
+#pragma unroll
+          for (int u = 0; u < 8; ++u) {
+            const int n = n0 + nq + h * 8 + u;
+            rb[kb * 2 + h][u] =
+                (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
+          }
+      }
     } else {
       // Bsrc[N][K]: row per n
       const int n = tid >> 2;
       const bool ok = (n0 + n) < N;
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
+      for (int h = 0; h < NCHB; ++h) {
         const int kk = bkq + h * 8;
         const float* src = Bsrc + (long long)(n0 + n) * K + kt + kk;
         const bool kok = ok && (kt + kk) < K;
@@ -359,17 +381,20 @@ __global__ __launch_bounds__(256) void k_gemm(
       for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)ra[h][u];
     }
     if (Bpre == nullptr && b_kxn) {
-      const int k = tid >> 2;
-      const int nq = (tid & 3) * 16;
 #pragma unroll
-      for (int h = 0; h < 2; ++h)
+      for (int kb = 0; kb < BKC / 64; ++kb) {
+        const int k = (tid >> 2) + kb * 64;
+        const int nq = (tid & 3) * 16;
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          L.Bs[nq + h * 8 + u][k] = (__bf16)rb[h][u];
+        for (int h = 0; h < 2; ++h)
+#pragma unroll
+          for (int u = 0; u < 8; ++u)
+            L.Bs[nq + h * 8 + u][k] = (__bf16)rb[kb * 2 + h][u];
+      }
     } else {
       const int n = tid >> 2;
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
+      for (int h = 0; h < NCHB; ++h) {
         const int kk = bkq + h * 8;
 #pragma unroll
         for (int u = 0; u < 8; ++u) L.Bs[n][kk + u] = (__bf16)rb[h][u];
@@ -379,13 +404,13 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   load_regs(0);
   write_lds();
-  for (int kt = 0; kt < K; kt += BK) {
+  for (int kt = 0; kt < K; kt += BKC) {
     __syncthreads();  // tile kt visible in LDS
-    const bool more = kt + BK < K;
-    if (more) load_regs(kt + BK);
-    // wave wv owns C rows [wv*16*RF, +16*RF); two 32-deep MFMA sub-steps
+    const bool more = kt + BKC < K;
+    if (more) load_regs(kt + BKC);
+    // wave wv owns C rows [wv*16*RF, +16*RF); BKC/32 32-deep MFMA sub-steps
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int kk = 0; kk < BKC / 32; ++kk) {
 #pragma unroll
       for (int rf = 0; rf < RF; ++rf) {
         const bf16x8 a0 =
@@ -437,8 +462,9 @@ __global__ __launch_bounds__(256) void k_gemm(
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad_gemm(
     const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
-    float* __restrict__ dW, long long M, int KcP, int N, int MS,
-    const act_t* __restrict__ imx, int XH, int XW, int XC, int XK, int XP) {
+    float* __restrict__ dW, float* __restrict__ part, long long M, int KcP,
+    int N, int MS, const act_t* __restrict__ imx, int XH, int XW, int XC,
+    int XK, int XP) {
   __shared__ GemmLds L;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -471,37 +497,41 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 
   auto load_regs = [&](long long mt) {
     const long long m = mt + row_s;
-    const bool ok = m < m_hi && m < M;
+    // m < m_hi implies m < M (m_hi = M*(slice+1)/MS <= M); clamped
+    // unconditional loads + select-to-zero keep the K-loop load pipeline
+    // intact (a guarded load makes hipcc branch around it).  KcP % 32 == 0
+    // and N % 16 == 0, so an 8-span is either fully in or fully out.
+    const bool ok = m < m_hi;
+    const long long mm = ok ? m : m_hi - 1;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int c = cq + h * 8;
+      if (imx != nullptr) {
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        rc[h][u] = 0.f;
-        rd[h][u] = 0.f;
-      }
-      if (ok && imx != nullptr) {
-        if (kc0 + c < KcP) {
+        for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
+        if (ok && kc0 + c < KcP) {
           const int iow = (int)(m % XW);
           const long long bh = m / XW;
           im2col8(imx, XH, XW, XC, XK, XP, (int)(bh / XH),
                   (int)(bh % XH), iow, kc0 + c, iKc, rc[h]);
         }
-      } else if (ok && (kc0 + c + 8) <= KcP) {
-        ld8v(cols + m * KcP + kc0 + c, rc[h]);
-      } else if (ok) {
-        const act_t* src = cols + m * KcP + kc0 + c;
+      } else {
+        const bool cok = (kc0 + c + 8) <= KcP;
+        const int ccl = cok ? kc0 + c : KcP - 8;
+        ld8v(cols + mm * KcP + ccl, rc[h]);
+        if (!(ok && cok)) {
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          if ((kc0 + c + u) < KcP) rc[h][u] = (float)src[u];
+          for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
+        }
       }
-      if (ok && (n0 + c + 8) <= N) {
-        ld8v(dpre + m * N + n0 + c, rd[h]);
-      } else if (ok) {
-        const act_t* src = dpre + m * N + n0 + c;
+      {
+        const bool nok = (n0 + c + 8) <= N;
+        const int ncl = nok ? n0 + c : N - 8;
+        ld8v(dpre + mm * N + ncl, rd[h]);
+        if (!(ok && nok)) {
 #pragma unroll
-        for (int u = 0; u < 8; ++u)
-          if ((n0 + c + u) < N) rd[h][u] = (float)src[u];
+          for (int u = 0; u < 8; ++u) rd[h][u] = 0.f;
+        }
       }
     }
   };
@@ -544,6 +574,11 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
 
   const int crow = wv * 16 + (lane >> 4) * 4;  // kc within tile
   const int ccol = lane & 15;                  // n within fragment
+  // part != nullptr: slab mode — every (kct, ntile) pair owns a disjoint
+  // region of slab `slice`, so these are plain stores and the slice count
+  // is decoupled from any atomic-combine cost (k_wsum folds the slabs).
+  float* const slab =
+      part ? part + (long long)slice * KcP * N : nullptr;
 #pragma unroll
   for (int f = 0; f < BN / 16; ++f) {
     if (f < nf) {
@@ -553,11 +588,40 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
         const int kc = kc0 + crow + r;
         if (kc < KcP && n < N) {
           const float v = acc[f][r];
-          if (v != 0.f) unsafeAtomicAdd(&dW[(long long)kc * N + n], v);
+          if (slab)
+            slab[(long long)kc * N + n] = v;
+          else if (v != 0.f)
+            unsafeAtomicAdd(&dW[(long long)kc * N + n], v);
         }
       }
     }
   }
+}
+
+// Fold MS weight-grad slabs into dW (+=): 8 fp32 per thread, the slab
+// axis walked with 2 streams in flight.
+__global__ __launch_bounds__(256) void k_wsum(const float* __restrict__ part,
+                                              float* __restrict__ dW,
+                                              long long total, int MS) {
+  const long long i8 = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i8 >= total) return;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  int sl = 0;
+  for (; sl + 2 <= MS; sl += 2) {
+    float a[8], b[8];
+    ld8v(part + sl * total + i8, a);
+    ld8v(part + (sl + 1) * total + i8, b);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc[u] += a[u] + b[u];
+  }
+  if (sl < MS) {
+    float a[8];
+    ld8v(part + sl * total + i8, a);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc[u] += a[u];
+  }
+#pragma unroll
+  for (int u = 0; u < 8; ++u) dW[i8 + u] += acc[u];
 }
 
 // Column-sum for the conv bias grad: db[n] += sum_m dpre[m][n].
@@ -1161,26 +1225,32 @@ int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
 }
 
 int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
-                            long long M, int KcP, int N, int MS,
+                            float* part, long long M, int KcP, int N, int MS,
                             const void* imx, int XH, int XW, int XC, int XK,
                             int XP, int actf, void* stream) {
-  const int ntiles = (N + BN - 1) / BN;
   const int ktiles = (KcP + BM - 1) / BM;
+  const int ntiles = (N + BN - 1) / BN;
   dim3 grid((unsigned)(ktiles * ntiles * MS)), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_wgrad_gemm<act_t>), grid, block,
                                           0, (hipStream_t)stream,
                                           (const act_t*)cols,
-                                          (const act_t*)dpre, dW, M, KcP, N,
-                                          MS, (const act_t*)imx, XH, XW, XC,
+                                          (const act_t*)dpre, dW, part, M, KcP,
+                                          N, MS, (const act_t*)imx, XH, XW, XC,
                                           XK, XP));
+  if (part) {
+    const long long total = (long long)KcP * N;
+    dim3 g2((unsigned)((total / 8 + 255) / 256));
+    hipLaunchKernelGGL(k_wsum, g2, block, 0, (hipStream_t)stream, part, dW,
+                       total, MS);
+  }
   return (int)hipGetLastError();
 }
 
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
-                         long long M, int KcP, int N, int MS, int actf,
-                         void* stream) {
-  return pcnn_deep_wgrad_gemm_ex(cols, dpre, dW, M, KcP, N, MS, nullptr, 0,
-                                 0, 0, 0, 0, actf, stream);
+                         float* part, long long M, int KcP, int N, int MS,
+                         int actf, void* stream) {
+  return pcnn_deep_wgrad_gemm_ex(cols, dpre, dW, part, M, KcP, N, MS, nullptr,
+                                 0, 0, 0, 0, 0, actf, stream);
 }
 
 int pcnn_deep_colsum(const void* dpre, float* part, float* db, long long M,

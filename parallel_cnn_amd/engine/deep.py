@@ -24,6 +24,16 @@ MODE_TRAIN, MODE_EVAL, MODE_INFER = 0, 1, 2
 
 
 class DeepWorkspace:
+    @staticmethod
+    def wgrad_ms(st, M: int) -> int:
+        """K-slice count for the weight-grad GEMM: ~512 total workgroups
+        (measured optimum — per-WG fixed costs rise beyond it and the
+        fp32 atomic combine over the dW tile lines stays cheap), never
+        slicing thinner than one 64-row BK chunk."""
+        ktiles = (st.kcp + 63) // 64
+        ntiles = (st.cout + 63) // 64
+        return max(1, min(256, 512 // (ktiles * ntiles), M // 64))
+
     def __init__(self, model: DeepCNN, max_batch: int, device, act_dtype):
         spec = model.spec
         B = max_batch
@@ -58,6 +68,7 @@ class DeepWorkspace:
         maxc = max(st.cout for st in spec.stages)
         self.colsum_part = torch.empty(512 * maxc, dtype=torch.float32,
                                        device=device)
+
         self.loss_accum = torch.zeros(1, dtype=torch.float32, device=device)
         self.correct_accum = torch.zeros(1, dtype=torch.int32, device=device)
 
@@ -177,12 +188,10 @@ class DeepTrainer:
                                   m.view(f"pool{i}_w"), w.acts[i], B, st.h,
                                   st.w, st.cout, st.pool_k, st_h)
             dapre = w.acts[i]
-            ktiles = (st.kcp + 63) // 64
-            ntiles = (st.cout + 63) // 64
-            # ~512 total workgroups is the measured sweet spot for the
-            # atomic-combine cost (tools/deep_sweep.py); the old 128 cap
-            # starved the small-KcP stage (88us -> 54us at ms=256)
-            ms = max(1, min(256, 512 // (ktiles * ntiles)))
+            ms = self.ws.wgrad_ms(st, M)
+            # measured (tools/deep_sweep.py): atomic combine beats the
+            # slab+reduce mode at every MS, and ~512 total WGs is optimal
+            # (per-WG fixed costs dominate beyond that)
             x_in = x if i == 0 else w.pouts[i - 1]
             implicit = False
             self._C.deep_wgrad_gemm(w.cols[i], dapre,

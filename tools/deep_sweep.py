@@ -41,20 +41,25 @@ def main():
         ktiles = (kcp + 63) // 64
         empty = torch.empty(0, device=dev)
         base = max(1, min(128, 512 // ktiles))
-        cands = sorted({base, 64, 128, 256, 512, 1024, 2048,
-                        max(1, 2048 // ktiles), max(1, 4096 // ktiles)})
-        row = []
-        for ms in cands:
-            if M // ms < 64:  # slice thinner than one BK chunk: skip
-                continue
-            us = t_kernel(lambda st, ms=ms: _C.deep_wgrad_gemm(
-                cols, dpre, dW, M, kcp, cout, ms, st, empty, h, w, cin, 5, 2))
-            row.append((ms, us))
-        cur = dict(row).get(base)
-        best = min(row, key=lambda p: p[1])
-        print("stage%d M=%6d KcP=%4d ktiles=%2d cur(ms=%d)=%.1fus best(ms=%d)=%.1fus" %
-              (si, M, kcp, ktiles, base, cur if cur else -1, best[0], best[1]))
-        print("   ", " ".join("ms=%d:%.1f" % p for p in row))
+        cands = sorted({base, 64, 128, 256,
+                        max(1, 1024 // ktiles), max(1, 2048 // ktiles),
+                        max(1, 4096 // ktiles)})
+        for slab in (False, True):
+            part = (torch.empty(max(cands) * kcp * cout, device=dev)
+                    if slab else torch.empty(0, device=dev))
+            row = []
+            for ms in cands:
+                if M // ms < 64:  # slice thinner than one BK chunk: skip
+                    continue
+                us = t_kernel(lambda st, ms=ms: _C.deep_wgrad_gemm(
+                    cols, dpre, dW, M, kcp, cout, ms, st, empty, h, w, cin,
+                    5, 2, part))
+                row.append((ms, us))
+            best = min(row, key=lambda p: p[1])
+            print("stage%d %s M=%6d KcP=%4d ktiles=%2d best(ms=%d)=%.1fus" %
+                  (si, "slab" if slab else "atom", M, kcp, ktiles,
+                   best[0], best[1]))
+            print("   ", " ".join("ms=%d:%.1f" % p for p in row))
 
     print("== k_colsum G sweep ==")
     for si, (h, w, cin, cout, kcp) in enumerate(stages):
