@@ -204,7 +204,13 @@ class DeepTrainer:
                                 m.grad_view(f"conv{i}_b"), M, st.cout, gsum,
                                 st_h)
             if i > 0:
-                # dgrad into the cols buffer (its forward use is done)
+                # dgrad into the cols buffer (its forward use is done).
+                # NOT torch.matmul/hipBLASLt: standalone Lt kernels beat
+                # ours 1.6-3.3x on these shapes (tools/membench.py), but
+                # IN-STEP Lt picks a generic tile (110 us vs our 112 for
+                # both dgrads) and the torch dispatch adds ~110 us of
+                # host wall at this step scale — measured 724 -> 835
+                # us/step.  Round-2 option: drive hipBLASLt from C++.
                 self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
                                   torch.empty(0), w.cols[i], M, st.cout,
                                   st.kcp, st.cout, st.kcp, 0, 0, st_h,
