@@ -34,6 +34,7 @@ using bf16 = __hip_bfloat16;
 using fp16 = __half;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 __device__ __forceinline__ float sigmoidf_dev(float v) {
   return 1.0f / (1.0f + __expf(-v));
@@ -163,6 +164,8 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
       }
     }
     if (sizeof(act_t) == 2)
+      // NOT __builtin_nontemporal_store: cols is re-read immediately by
+      // the GEMM and partially L2-resident — NT stores cost 15% end-to-end
       *reinterpret_cast<uint4*>(orow + kc) =
           *reinterpret_cast<const uint4*>(out);
     else
@@ -1183,7 +1186,7 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
   // (the 2-barrier loop is staging/latency bound, not MFMA bound — see
   // profiles/: MfmaUtil ~3%); the instantiation stays available but the
   // selector is off.
-  const bool big = false && M >= 128 * 512;
+  const bool big = false && M >= 128 * 512;  // measured slower (twice)
   const int tbm = big ? 128 : BM;
   const long long mtiles = (M + tbm - 1) / tbm;
   dim3 grid((unsigned)(mtiles * ntiles)), block(256);
