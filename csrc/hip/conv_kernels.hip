@@ -261,7 +261,7 @@ __global__ __launch_bounds__(256) void k_gemm(
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
     int XC, int XK, int XP) {
-  __shared__ GemmLdsT<TBM, BKC> L;
+  __shared__ GemmLdsT<TBM, BKC> Lb[2];  // double-buffered tiles
   constexpr int RF = TBM / 64;        // row fragments per wave
   constexpr int TPR = 256 / TBM;      // staging threads per A row
   constexpr int SPAN = BKC / TPR;     // k-span per staging thread
@@ -376,7 +376,7 @@ __global__ __launch_bounds__(256) void k_gemm(
     }
   };
 
-  auto write_lds = [&]() {
+  auto write_lds = [&](GemmLdsT<TBM, BKC>& L) {
 #pragma unroll
     for (int h = 0; h < NCH; ++h) {
       const int kk = kq + h * 8;
@@ -405,12 +405,18 @@ __global__ __launch_bounds__(256) void k_gemm(
     }
   };
 
+  // Double-buffered pipeline, ONE barrier per K-iteration: while the
+  // MFMAs read tile t from Lb[p], tile t+1 (issued a full iteration ago)
+  // is written to Lb[1-p] and tile t+2's loads are issued — the
+  // s_waitcnt for a tile's global loads lands ~one iteration after
+  // issue instead of right after the MFMA block.
   load_regs(0);
-  write_lds();
-  for (int kt = 0; kt < K; kt += BKC) {
-    __syncthreads();  // tile kt visible in LDS
-    const bool more = kt + BKC < K;
-    if (more) load_regs(kt + BKC);
+  write_lds(Lb[0]);
+  if (BKC < K) load_regs(BKC);  // issue tile 1 (skip for single-tile K)
+  __syncthreads();              // Lb[0] visible
+  int p = 0;
+  for (int kt = 0; kt < K; kt += BKC, p ^= 1) {
+    auto& L = Lb[p];
     // wave wv owns C rows [wv*16*RF, +16*RF); BKC/32 32-deep MFMA sub-steps
 #pragma unroll
     for (int kk = 0; kk < BKC / 32; ++kk) {
@@ -430,8 +436,11 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
       }
     }
-    __syncthreads();  // all reads of tile kt done
-    if (more) write_lds();
+    if (kt + BKC < K) {
+      write_lds(Lb[p ^ 1]);  // waits on tile t+1's loads here
+      if (kt + 2 * BKC < K) load_regs(kt + 2 * BKC);
+    }
+    __syncthreads();  // reads of Lb[p] done AND Lb[1-p] complete
   }
 
   // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
