@@ -477,7 +477,7 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     float* __restrict__ dW, float* __restrict__ part, long long M, int KcP,
     int N, int MS, const act_t* __restrict__ imx, int XH, int XW, int XC,
     int XK, int XP) {
-  __shared__ GemmLds L;
+  __shared__ GemmLds Lb[2];  // double-buffered M-chunks
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -548,7 +548,7 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     }
   };
 
-  auto write_lds = [&]() {
+  auto write_lds = [&](GemmLds& L) {
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int c = cq + h * 8;
@@ -560,12 +560,14 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     }
   };
 
+  // Double-buffered pipeline, one barrier per M-chunk (see k_gemm)
   load_regs(m_lo);
-  write_lds();
-  for (long long mt = m_lo; mt < m_hi; mt += BK) {
-    __syncthreads();
-    const bool more = mt + BK < m_hi;
-    if (more) load_regs(mt + BK);
+  write_lds(Lb[0]);
+  if (m_lo + BK < m_hi) load_regs(m_lo + BK);
+  __syncthreads();
+  int p = 0;
+  for (long long mt = m_lo; mt < m_hi; mt += BK, p ^= 1) {
+    auto& L = Lb[p];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
@@ -580,8 +582,11 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
         }
       }
     }
+    if (mt + BK < m_hi) {
+      write_lds(Lb[p ^ 1]);
+      if (mt + 2 * BK < m_hi) load_regs(mt + 2 * BK);
+    }
     __syncthreads();
-    if (more) write_lds();
   }
 
   const int crow = wv * 16 + (lane >> 4) * 4;  // kc within tile
