@@ -79,6 +79,20 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
   out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
 }
 
+// Magic-multiply unsigned division (divisor known on the host): exact
+// for n < 2^38/d and n*M < 2^64, i.e. n < 2^26 with d <= 4096 — the
+// launcher checks the domain and falls back to the 64-bit-div kernel
+// variant beyond it.  A runtime u32 divide is ~25 VALU ops (and a 64-bit
+// one a ~100-op libcall); this is a 32x32->64 mul + shift.
+constexpr int FDIV_SH = 38;
+__host__ __device__ __forceinline__ unsigned long long fdiv_magic(
+    unsigned d) {
+  return ((1ULL << FDIV_SH) + d - 1) / d;
+}
+__device__ __forceinline__ unsigned fdiv(unsigned n, unsigned long long M) {
+  return (unsigned)((n * M) >> FDIV_SH);
+}
+
 // ---------------------------------------------------------------------------
 // im2col (NHWC, same-padding):
 //   cols[(b*OH+oh)*OW+ow][ (i*K+j)*Cin+ci ] = x[b][oh+i-P][ow+j-P][ci] or 0
@@ -93,21 +107,37 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
 // so it is a single 16B load; smaller Cin walks elementwise with
 // incremental counters (the per-element div/mod version was 83% VALUBusy
 // on index math).
-template <typename act_t>
+template <typename act_t, bool FAST>
 __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
                          int B, int H, int W, int Cin, int K, int P,
-                         int KcP) {
+                         int KcP, unsigned long long fd_tpr,
+                         unsigned long long fd_w, unsigned long long fd_h) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long total = ((long long)B * H * W * KcP) / 32;
   if (idx >= total) return;
-  int kc = (int)((idx * 32) % KcP);
-  const long long m = (idx * 32) / KcP;
+  int kc, ow, oh, b;
+  long long m;
+  if (FAST) {
+    const unsigned n = (unsigned)idx;
+    const unsigned tpr = (unsigned)(KcP / 32);  // threads per cols row
+    const unsigned mq = fdiv(n, fd_tpr);
+    kc = (int)(n - mq * tpr) * 32;
+    const unsigned bhq = fdiv(mq, fd_w);
+    ow = (int)(mq - bhq * (unsigned)W);
+    const unsigned bq = fdiv(bhq, fd_h);
+    oh = (int)(bhq - bq * (unsigned)H);
+    b = (int)bq;
+    m = mq;
+  } else {
+    kc = (int)((idx * 32) % KcP);
+    m = (idx * 32) / KcP;
+    ow = (int)(m % W);
+    const long long bh = m / W;
+    oh = (int)(bh % H);
+    b = (int)(bh / H);
+  }
   const int Kc = K * K * Cin;
   const int rowc = K * Cin;
-  const int ow = (int)(m % W);
-  const long long bh = m / W;
-  const int oh = (int)(bh % H);
-  const int b = (int)(bh / H);
   const act_t* __restrict__ xb = x + (long long)b * H * W * Cin;
   act_t* __restrict__ orow = cols + m * KcP;
 
@@ -738,22 +768,38 @@ __global__ __launch_bounds__(256) void k_colsum_fin(
 //   out[b,h,w,ci] = g * pout*(1-pout)        (pout = that pixel's value)
 // When pout == nullptr the sigmoid factor is skipped (plain dX).
 // ---------------------------------------------------------------------------
-template <typename act_t>
+template <typename act_t, bool FAST>
 __global__ void k_col2im_sigbwd(const act_t* __restrict__ dcols,
                                 const act_t* __restrict__ pout,
                                 act_t* __restrict__ out, int B, int H, int W,
-                                int Cin, int K, int P, int KcP) {
+                                int Cin, int K, int P, int KcP,
+                                unsigned long long fd_cpr,
+                                unsigned long long fd_w,
+                                unsigned long long fd_h) {
   // 8 consecutive ci per thread (Cin % 8 == 0 for every caller): the K*K
   // gather loads become 8-wide vector loads.
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long total = (long long)B * H * W * Cin / 8;
   if (idx >= total) return;
-  const int ci0 = (int)((idx * 8) % Cin);
-  long long t = (idx * 8) / Cin;
-  const int w = (int)(t % W);
-  t /= W;
-  const int h = (int)(t % H);
-  const int b = (int)(t / H);
+  int ci0, w, h, b;
+  if (FAST) {
+    const unsigned n = (unsigned)idx;
+    const unsigned cpr = (unsigned)(Cin / 8);
+    const unsigned tq = fdiv(n, fd_cpr);
+    ci0 = (int)(n - tq * cpr) * 8;
+    const unsigned hq = fdiv(tq, fd_w);
+    w = (int)(tq - hq * (unsigned)W);
+    const unsigned bq = fdiv(hq, fd_h);
+    h = (int)(hq - bq * (unsigned)H);
+    b = (int)bq;
+  } else {
+    ci0 = (int)((idx * 8) % Cin);
+    long long t = (idx * 8) / Cin;
+    w = (int)(t % W);
+    t /= W;
+    h = (int)(t % H);
+    b = (int)(t / H);
+  }
   float g[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   for (int i = 0; i < K; ++i) {
     const int oh = h + P - i;
@@ -1183,10 +1229,24 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
                      int K, int P, int KcP, int actf, void* stream) {
   const long long total = (long long)B * H * W * KcP / 32;  // 32 elems/thread
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
-  PCNN_DISPATCH(actf, hipLaunchKernelGGL(
-                          (k_im2col<act_t>), grid, block, 0,
-                          (hipStream_t)stream, (const act_t*)x, (act_t*)cols,
-                          B, H, W, Cin, K, P, KcP));
+  const bool fast = total < (1LL << 26) && (KcP / 32) <= 4096 && W <= 4096 &&
+                    H <= 4096;
+  const unsigned long long fd_tpr = fdiv_magic((unsigned)(KcP / 32));
+  const unsigned long long fd_w = fdiv_magic((unsigned)W);
+  const unsigned long long fd_h = fdiv_magic((unsigned)H);
+  if (fast) {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL(
+                            (k_im2col<act_t, true>), grid, block, 0,
+                            (hipStream_t)stream, (const act_t*)x,
+                            (act_t*)cols, B, H, W, Cin, K, P, KcP, fd_tpr,
+                            fd_w, fd_h));
+  } else {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL(
+                            (k_im2col<act_t, false>), grid, block, 0,
+                            (hipStream_t)stream, (const act_t*)x,
+                            (act_t*)cols, B, H, W, Cin, K, P, KcP, fd_tpr,
+                            fd_w, fd_h));
+  }
   return (int)hipGetLastError();
 }
 
@@ -1287,11 +1347,28 @@ int pcnn_deep_col2im_sigbwd(const void* dcols, const void* pout, void* out,
                             int KcP, int actf, void* stream) {
   const long long total = (long long)B * H * W * Cin / 8;
   dim3 grid((unsigned)((total + 255) / 256)), block(256);
-  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_col2im_sigbwd<act_t>), grid,
-                                          block, 0, (hipStream_t)stream,
-                                          (const act_t*)dcols,
-                                          (const act_t*)pout, (act_t*)out, B,
-                                          H, W, Cin, K, P, KcP));
+  const bool fast = total < (1LL << 26) && (Cin / 8) <= 4096 && W <= 4096 &&
+                    H <= 4096;
+  const unsigned long long fd_cpr = fdiv_magic((unsigned)(Cin / 8));
+  const unsigned long long fd_w = fdiv_magic((unsigned)W);
+  const unsigned long long fd_h = fdiv_magic((unsigned)H);
+  if (fast) {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_col2im_sigbwd<act_t, true>),
+                                            grid, block, 0,
+                                            (hipStream_t)stream,
+                                            (const act_t*)dcols,
+                                            (const act_t*)pout, (act_t*)out,
+                                            B, H, W, Cin, K, P, KcP, fd_cpr,
+                                            fd_w, fd_h));
+  } else {
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_col2im_sigbwd<act_t, false>),
+                                            grid, block, 0,
+                                            (hipStream_t)stream,
+                                            (const act_t*)dcols,
+                                            (const act_t*)pout, (act_t*)out,
+                                            B, H, W, Cin, K, P, KcP, fd_cpr,
+                                            fd_w, fd_h));
+  }
   return (int)hipGetLastError();
 }
 
