@@ -91,9 +91,11 @@ def main() -> int:
 
     def run(n_steps: int):
         if args.model == "deepcnn":
+            fn = (trainer.step_graph if getattr(trainer, "_graph", None)
+                  is not None else trainer.step)
             for st in range(n_steps):
                 i = (st % n_pool_batches) * B
-                trainer.step(x_pool[i:i + B], y_pool[i:i + B])
+                fn(x_pool[i:i + B], y_pool[i:i + B])
         else:
             trainer.run_steps_pooled(x_pool, y_pool, n_steps)
 

@@ -167,3 +167,29 @@ def test_bench_deepcnn_contract(device):
     assert r["value"] > 0
     assert "DeepCNN" in r["config"]["model"]
     assert r["config"]["global_batch"] == 256
+
+
+def test_deep_graph_matches_eager(device):
+    """hipGraph-captured deep step == eager step trajectory."""
+    B = 16
+    cfg = TrainConfig(batch_size=B, device="cuda", backend="hip",
+                      act_dtype="bf16", log_interval=0)
+    x, labels = synthetic_images(B * 3, 32, 32, 3, seed=21, structured=False)
+    te = DeepTrainer(cfg)
+    tg = DeepTrainer(cfg)
+    tg.enable_graph()
+    assert torch.allclose(te.model.params, tg.model.params)
+    for s in range(3):
+        xb, lb = te.stage_batch(x[s * B:(s + 1) * B],
+                                labels[s * B:(s + 1) * B])
+        te.step(xb, lb)
+        xg, lg = tg.stage_batch(x[s * B:(s + 1) * B],
+                                labels[s * B:(s + 1) * B])
+        tg.step_graph(xg, lg)
+    torch.cuda.synchronize()
+    diff = (te.model.params - tg.model.params).abs().max().item()
+    assert diff < 1e-5, diff
+    le, ne = te.consume_loss()
+    lg_, ng = tg.consume_loss()
+    assert ne == ng == 3 * B
+    assert abs(le - lg_) < 1e-3 * max(1.0, abs(le))
