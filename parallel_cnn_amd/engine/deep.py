@@ -215,7 +215,6 @@ class DeepTrainer:
                                   torch.empty(0), w.cols[i], M, st.cout,
                                   st.kcp, st.cout, st.kcp, 0, 0, st_h,
                                   w.wbf[i])
-                prev = spec.stages[i - 1]
                 self._C.deep_col2im_sigbwd(w.cols[i], w.pouts[i - 1],
                                            w.dppre[i - 1], B, st.h, st.w,
                                            st.cin, st.k, st.pad, st.kcp,
