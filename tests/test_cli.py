@@ -57,3 +57,28 @@ def test_cli_mnist_idx_files(tmp_path):
                    "--log-interval", "0"])
     assert out.returncode == 0, out.stdout + out.stderr
     assert "Error Rate: " in out.stdout
+
+
+def test_example_configs_parse():
+    """Every YAML under examples/ loads into a valid TrainConfig."""
+    import glob
+    from parallel_cnn_amd.config import TrainConfig
+    paths = sorted(glob.glob("examples/*.yaml"))
+    assert len(paths) >= 4
+    for p in paths:
+        cfg = TrainConfig.from_yaml(p)
+        assert cfg.batch_size > 0
+        assert cfg.model in ("lenet5", "deepcnn")
+
+
+def test_cli_runs_with_example_config(tmp_path):
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "-m", "parallel_cnn_amd.train", "--config",
+         "examples/lenet_reference.yaml", "--device", "cpu", "--backend",
+         "cpu", "--train-count", "64", "--test-count", "32",
+         "--log-interval", "0"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    assert "Error Rate" in r.stdout
