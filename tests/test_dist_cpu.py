@@ -115,3 +115,52 @@ def test_dp_with_grad_accumulation_matches_big_batch():
         assert p.exitcode == 0
     assert torch.allclose(params, ref.model.params, atol=1e-6), \
         (params - ref.model.params).abs().max()
+
+
+def _deep_worker(rank, world, port, result_q):
+    os.environ.update({
+        "RANK": str(rank),
+        "LOCAL_RANK": str(rank),
+        "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    from parallel_cnn_amd.data.mnist import synthetic_images
+    from parallel_cnn_amd.engine.deep import DeepTrainer
+    ctx = pdist.init_from_env(device="cpu")
+    cfg = TrainConfig(backend="torchref", device="cpu", model="deepcnn",
+                      batch_size=8, log_interval=0)
+    t = DeepTrainer(cfg, ctx=ctx)
+    x, y = synthetic_images(16, 32, 32, 3, seed=0, structured=False)
+    lo = rank * 8
+    t.step(*t.stage_batch(x[lo:lo + 8], y[lo:lo + 8]))
+    if rank == 0:
+        result_q.put(t.model.params.clone())
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_deep_dp_matches_single_rank():
+    """DeepCNN DP(2) at bs=8/rank == one single-rank bs=16 step (covers
+    the deep trainer's fused-bucket all-reduce path the 8-GPU scale run
+    exercises with --model deepcnn)."""
+    from parallel_cnn_amd.data.mnist import synthetic_images
+    from parallel_cnn_amd.engine.deep import DeepTrainer
+    cfg = TrainConfig(backend="torchref", device="cpu", model="deepcnn",
+                      batch_size=16, log_interval=0)
+    ref = DeepTrainer(cfg)
+    x, y = synthetic_images(16, 32, 32, 3, seed=0, structured=False)
+    ref.step(*ref.stage_batch(x, y))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_deep_worker, args=(r, 2, 29534, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    params = q.get()
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert torch.allclose(params, ref.model.params, atol=1e-5), \
+        (params - ref.model.params).abs().max()
