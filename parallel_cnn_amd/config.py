@@ -95,9 +95,14 @@ class TrainConfig:
             v = getattr(args, f.name, None)
             if v is None:
                 continue
+            if isinstance(v, bool):
+                # bool flags are already parsed by the CLI lambda; coercing
+                # through str() would turn False into the truthy "False"
+                setattr(cfg, f.name, v)
+                continue
             typ = {int: int, float: float, str: str}.get(
                 type(getattr(cfg, f.name)), str)
-            setattr(cfg, f.name, typ(v) if v is not None else v)
+            setattr(cfg, f.name, typ(v))
         return cfg
 
 

@@ -46,6 +46,10 @@ class DevicePrefetcher:
 
     def _issue_copy(self, slot: int, off: int) -> None:
         b = self.batch
+        # the previous async H2D from this pinned buffer may still be in
+        # flight (the host loop runs arbitrarily far ahead of the device);
+        # wait for it before overwriting the staging buffer
+        self._copy_done[slot].synchronize()
         self._hx[slot].copy_(self.x[off:off + b])
         self._hl[slot].copy_(self.labels[off:off + b].to(torch.int32))
         with torch.cuda.stream(self.copy_stream):

@@ -111,7 +111,8 @@ class DeepTrainer:
     # ------------------------------------------------------------------ util
     def _scale(self, B: int) -> float:
         if self.cfg.grad_reduction == "mean":
-            return 1.0 / float(B * self.ctx.world_size)
+            return 1.0 / float(B * self.ctx.world_size *
+                               self.cfg.grad_accum)
         return 1.0
 
     def stage_batch(self, x: torch.Tensor, labels: torch.Tensor):
@@ -279,8 +280,16 @@ class DeepTrainer:
 
     # ------------------------------------------------------------------ step
     def step(self, x: torch.Tensor, labels: torch.Tensor) -> None:
+        """One micro/step.  The wgrad kernels ACCUMULATE into the flat
+        gradient bucket (deep_update zeroes it), so cfg.grad_accum > 1
+        simply defers the all-reduce + update to every grad_accum-th
+        call — same contract as the LeNet Trainer."""
         B = x.shape[0]
         assert B <= self.ws.max_batch
+        self._accum = getattr(self, "_accum", 0) + 1
+        apply_update = self._accum >= self.cfg.grad_accum
+        if apply_update:
+            self._accum = 0
         scale = self._scale(B)
         if self.backend == "hip":
             if self.timers is not None:
@@ -288,21 +297,24 @@ class DeepTrainer:
                     self._hip_forward(x, labels, B, MODE_TRAIN)
                 with self.timers.phase("backward"):
                     self._hip_backward(x, B)
-                with self.timers.phase("all-reduce"):
-                    pdist.allreduce_grads(self.model.grads)
-                with self.timers.phase("update"):
-                    self._C.deep_update(self.model.params, self.model.grads,
-                                        self.cfg.dt * scale,
-                                        native.current_stream_handle())
+                if apply_update:
+                    with self.timers.phase("all-reduce"):
+                        pdist.allreduce_grads(self.model.grads)
+                    with self.timers.phase("update"):
+                        self._C.deep_update(self.model.params,
+                                            self.model.grads,
+                                            self.cfg.dt * scale,
+                                            native.current_stream_handle())
                 self._samples_seen += B * self.ctx.world_size
                 self.global_step += 1
                 return
             self._hip_forward(x, labels, B, MODE_TRAIN)
             self._hip_backward(x, B)
-            pdist.allreduce_grads(self.model.grads)
-            self._C.deep_update(self.model.params, self.model.grads,
-                                self.cfg.dt * scale,
-                                native.current_stream_handle())
+            if apply_update:
+                pdist.allreduce_grads(self.model.grads)
+                self._C.deep_update(self.model.params, self.model.grads,
+                                    self.cfg.dt * scale,
+                                    native.current_stream_handle())
         else:
             spec = self.model.spec
             xh = x.view(B, spec.in_h, spec.in_w, spec.in_ch)
@@ -311,10 +323,12 @@ class DeepTrainer:
                                             labels)
             self._loss_host += loss
             self.model.grads += grads
-            pdist.allreduce_grads(self.model.grads)
-            with torch.no_grad():
-                self.model.params += self.cfg.dt * scale * self.model.grads
-                self.model.grads.zero_()
+            if apply_update:
+                pdist.allreduce_grads(self.model.grads)
+                with torch.no_grad():
+                    self.model.params += (self.cfg.dt * scale *
+                                          self.model.grads)
+                    self.model.grads.zero_()
         self._samples_seen += B * self.ctx.world_size
         self.global_step += 1
 
@@ -353,6 +367,35 @@ class DeepTrainer:
         total_loss += loss
         total_n += cnt
         return total_loss / max(1, total_n)
+
+    @torch.no_grad()
+    def forward_logits(self, x: torch.Tensor) -> torch.Tensor:
+        """Forward a host fp32 batch [N, H*W*Cin] of ANY size, chunked by
+        the workspace max batch (the deep twin of Trainer.classify's
+        chunking — serving requests must never exceed the device
+        workspace sizing, see serve.py).  Returns host fp32 logits
+        [N, n_classes]."""
+        w, spec = self.ws, self.model.spec
+        outs = []
+        bs = w.max_batch
+        for i in range(0, x.shape[0], bs):
+            nb = min(bs, x.shape[0] - i)
+            xb, lb = self.stage_batch(
+                x[i:i + nb], torch.zeros(nb, dtype=torch.int64))
+            B = xb.shape[0]
+            if self.backend == "hip":
+                self._hip_forward(xb, lb, B, MODE_INFER)
+                outs.append(w.y[:B].cpu().clone())
+            else:
+                _, _, y = deep_ref.forward(
+                    xb.view(B, spec.in_h, spec.in_w, spec.in_ch), self.model)
+                outs.append(y)
+        return torch.cat(outs)
+
+    @torch.no_grad()
+    def classify(self, x: torch.Tensor) -> torch.Tensor:
+        """Predicted labels for a host fp32 batch [N, H*W*Cin] (chunked)."""
+        return self.forward_logits(x).argmax(dim=1)
 
     @torch.no_grad()
     def evaluate(self, x: torch.Tensor, labels: torch.Tensor,

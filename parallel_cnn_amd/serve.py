@@ -61,21 +61,9 @@ def create_app(cfg: Optional[TrainConfig] = None,
                 400, f"each image must be a flat list of {spec_pixels} "
                      f"floats, got shape {tuple(x.shape)}")
         if cfg.model == "deepcnn":
-            # deep path: oracle/hip evaluate per batch
-            xb, _ = trainer.stage_batch(
-                x, torch.zeros(x.shape[0], dtype=torch.int64))
-            B = x.shape[0]
-            if trainer.backend == "hip":
-                trainer._hip_forward(xb, torch.zeros(
-                    B, dtype=torch.int32, device=trainer.device), B, 2)
-                torch.cuda.synchronize()
-                y = trainer.ws.y[:B].cpu()
-            else:
-                from .ops import deep_ref
-                spec = trainer.model.spec
-                _, _, y = deep_ref.forward(
-                    xb.view(B, spec.in_h, spec.in_w, spec.in_ch),
-                    trainer.model)
+            # chunked by the device workspace max batch — a request larger
+            # than cfg.batch_size must never reach the raw forward
+            y = trainer.forward_logits(x)
             labels = y.argmax(1)
         else:
             labels = trainer.classify(x)

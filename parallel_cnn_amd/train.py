@@ -23,6 +23,11 @@ from .parallel import dist as pdist
 
 def load_datasets(cfg: TrainConfig):
     if cfg.data == "mnist":
+        if cfg.model == "deepcnn":
+            raise ValueError(
+                "--model deepcnn expects 32x32x3 inputs; MNIST is 28x28x1 "
+                "(use --data synthetic with deepcnn, or --model lenet5 "
+                "with --data mnist)")
         xtr, ytr = load_mnist(
             os.path.join(cfg.data_dir, "train-images.idx3-ubyte"),
             os.path.join(cfg.data_dir, "train-labels.idx1-ubyte"))
@@ -48,8 +53,7 @@ def main(argv=None) -> int:
     args = p.parse_args(argv)
     cfg = TrainConfig.from_args(args)
 
-    import sys
-    profile = "--profile" in (argv or sys.argv)
+    profile = bool(args.profile)
     ctx = pdist.init_from_env(cfg.resolved_device())
     trainer = (DeepTrainer(cfg, ctx=ctx) if cfg.model == "deepcnn"
                else Trainer(cfg, ctx=ctx))

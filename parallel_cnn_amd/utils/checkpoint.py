@@ -36,5 +36,16 @@ def load_checkpoint(trainer, path: str) -> Optional[dict]:
         raise ValueError(
             f"checkpoint {path!r} is for a model with {meta['n_params']} "
             f"params, this model has {trainer.model.params.numel()}")
+    # a matching parameter COUNT does not make a matching model: a
+    # pool=max or loss=softmax_ce checkpoint would load silently into a
+    # trainable/residual config and predict garbage
+    for key, cur in (("model", trainer.cfg.model),
+                     ("pool", getattr(trainer.cfg, "pool", "trainable")),
+                     ("loss", getattr(trainer.cfg, "loss", "residual"))):
+        want = meta.get(key)
+        if want is not None and want != cur:
+            raise ValueError(
+                f"checkpoint {path!r} was trained with {key}={want!r}, "
+                f"but the running config has {key}={cur!r}")
     trainer.global_step = int(meta.get("global_step", 0))
     return meta

@@ -82,3 +82,12 @@ def test_cli_runs_with_example_config(tmp_path):
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr
     assert "Error Rate" in r.stdout
+
+
+def test_deepcnn_mnist_mismatch_errors():
+    """--model deepcnn --data mnist used to silently feed 784-pixel MNIST
+    to a 3072-pixel model; it must fail with a clear shape error."""
+    out = run_cli(["--device", "cpu", "--model", "deepcnn",
+                   "--data", "mnist", "--epochs", "1"])
+    assert out.returncode != 0
+    assert "28x28x1" in (out.stdout + out.stderr)

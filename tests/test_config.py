@@ -41,6 +41,33 @@ def test_cli_overrides_yaml(tmp_path):
     assert cfg.dt == 0.2
 
 
+def test_cli_bool_false_is_false():
+    """Regression: bool flags used to be re-coerced through str(), so
+    `--overlap-comm false` yielded the truthy string 'False'."""
+    ap = argparse.ArgumentParser()
+    TrainConfig.add_cli_args(ap)
+    args = ap.parse_args(["--overlap-comm", "false",
+                          "--fuse-wgrad", "false"])
+    cfg = TrainConfig.from_args(args)
+    assert cfg.overlap_comm is False
+    assert cfg.fuse_wgrad is False
+    args = ap.parse_args(["--overlap-comm", "true"])
+    cfg = TrainConfig.from_args(args)
+    assert cfg.overlap_comm is True
+
+
+def test_cli_bool_matches_yaml(tmp_path):
+    p = tmp_path / "cfg.yaml"
+    p.write_text("overlap_comm: true\nfuse_wgrad: false\n")
+    ap = argparse.ArgumentParser()
+    TrainConfig.add_cli_args(ap)
+    # CLI overrides YAML; explicit false must win over a YAML true
+    args = ap.parse_args(["--config", str(p), "--overlap-comm", "false"])
+    cfg = TrainConfig.from_args(args)
+    assert cfg.overlap_comm is False
+    assert cfg.fuse_wgrad is False
+
+
 def test_resolution():
     cfg = TrainConfig(device="cpu", backend="auto")
     assert cfg.resolved_backend() == "cpu"

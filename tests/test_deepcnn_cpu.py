@@ -151,3 +151,21 @@ def test_channel_width_validation():
     from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
     with pytest.raises(ValueError, match="multiples of 16"):
         DeepCNNSpec(channels=(20, 64, 64))
+
+
+def test_deep_grad_accumulation_matches_big_batch():
+    """ADVICE regression: DeepTrainer used to ignore cfg.grad_accum.
+    grad_accum=2 over two bs=4 micro-batches == one bs=8 step with mean
+    reduction (deep wgrad accumulates; the update scale folds in the
+    accumulation count)."""
+    x, y = synthetic_images(8, 32, 32, 3, seed=11)
+    cfg_a = TrainConfig(backend="torchref", device="cpu", batch_size=4,
+                        grad_accum=2, log_interval=0)
+    cfg_b = TrainConfig(backend="torchref", device="cpu", batch_size=8,
+                        log_interval=0)
+    ta, tb = DeepTrainer(cfg_a), DeepTrainer(cfg_b)
+    ta.step(*ta.stage_batch(x[:4], y[:4]))
+    ta.step(*ta.stage_batch(x[4:], y[4:]))
+    tb.step(*tb.stage_batch(x, y))
+    assert torch.allclose(ta.model.params, tb.model.params, atol=1e-6)
+    assert ta.model.grads.abs().sum() == 0  # consumed at the boundary

@@ -88,3 +88,21 @@ def test_checkpoint_sidecar_wrong_model(tmp_path):
         json.dump(meta, f)
     with pytest.raises(ValueError, match="999"):
         load_checkpoint(Trainer(cfg), path)
+
+
+def test_checkpoint_sidecar_config_mismatch(tmp_path):
+    """ADVICE regression: a pool=max / loss=softmax_ce checkpoint must not
+    load silently into a trainable/residual config of the same size."""
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    cfg = TrainConfig(device="cpu", backend="cpu", batch_size=8,
+                      log_interval=0, pool="max", loss="softmax_ce")
+    t = Trainer(cfg)
+    path = str(tmp_path / "ck.bin")
+    save_checkpoint(t, path)
+    other = TrainConfig(device="cpu", backend="cpu", batch_size=8,
+                        log_interval=0)  # trainable/residual defaults
+    with pytest.raises(ValueError, match="pool"):
+        load_checkpoint(Trainer(other), path)

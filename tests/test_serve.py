@@ -66,3 +66,32 @@ def test_deepcnn_serving():
     r = c.post("/predict", json={"images": x.tolist()})
     assert r.status_code == 200
     assert len(r.json()["labels"]) == 2
+
+
+def test_deepcnn_request_larger_than_batch():
+    """Regression: a deep /predict with B > cfg.batch_size used to feed the
+    raw forward (workspaces sized for batch_size) — OOB GPU writes on the
+    hip path.  The chunked classify must return the same labels as
+    per-chunk requests."""
+    from parallel_cnn_amd.data.mnist import synthetic_images
+    cfg = TrainConfig(device="cpu", backend="torchref", model="deepcnn",
+                      batch_size=4, log_interval=0)
+    app = create_app(cfg)
+    c = TestClient(app)
+    x, _ = synthetic_images(11, 32, 32, 3, seed=3)  # 11 > 4, ragged tail
+    r = c.post("/predict", json={"images": x.tolist()})
+    assert r.status_code == 200
+    big = r.json()["labels"]
+    assert len(big) == 11
+    small = []
+    for i in range(0, 11, 4):
+        rr = c.post("/predict", json={"images": x[i:i + 4].tolist()})
+        small += rr.json()["labels"]
+    assert big == small
+
+
+def test_lenet_request_larger_than_batch(client):
+    x, _ = synthetic_mnist(70, seed=5)  # 70 > server batch of 64
+    r = client.post("/predict", json={"images": x.tolist()})
+    assert r.status_code == 200
+    assert len(r.json()["labels"]) == 70
