@@ -102,6 +102,7 @@ class DeepTrainer:
         self._loss_host = 0.0
         self._samples_seen = 0
         self.global_step = 0
+        self.epoch = 0              # completed-epoch cursor (exact resume)
         self.timers = None  # set by enable_profiling
 
     def enable_profiling(self) -> None:
@@ -366,6 +367,7 @@ class DeepTrainer:
         loss, cnt = self.consume_loss()
         total_loss += loss
         total_n += cnt
+        self.epoch += 1
         return total_loss / max(1, total_n)
 
     @torch.no_grad()

@@ -71,6 +71,7 @@ class Trainer:
         self._loss_host = 0.0       # cpu/torchref backends accumulate here
         self._samples_seen = 0
         self.global_step = 0
+        self.epoch = 0              # completed-epoch cursor (exact resume)
         # cached stream handle for the hot per-step path (the graph-capture
         # body resolves the live stream instead)
         self._sh = native.current_stream_handle() if self.backend == "hip" \
@@ -445,4 +446,5 @@ class Trainer:
         loss, cnt = self.consume_loss()
         total_loss += loss
         total_n += cnt
+        self.epoch += 1
         return total_loss / max(1, total_n)

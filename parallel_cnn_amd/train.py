@@ -68,7 +68,9 @@ def main(argv=None) -> int:
     if ctx.is_main:
         print("Learning", flush=True)
     t0 = time.perf_counter()
-    for epoch in range(cfg.epochs):
+    # a loaded checkpoint restores trainer.epoch — resume the remaining
+    # epochs, not the full schedule
+    for epoch in range(trainer.epoch, cfg.epochs):
         err = trainer.train_epoch(xtr, ytr)
         dt = time.perf_counter() - t0
         if ctx.is_main:

@@ -106,3 +106,52 @@ def test_checkpoint_sidecar_config_mismatch(tmp_path):
                         log_interval=0)  # trainable/residual defaults
     with pytest.raises(ValueError, match="pool"):
         load_checkpoint(Trainer(other), path)
+
+
+def test_exact_resume_equivalence(tmp_path):
+    """train(2 epochs) == train(1) -> save -> load -> train(1): the sidecar
+    persists the epoch cursor and RNG state, so resume continues the exact
+    trajectory."""
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    from parallel_cnn_amd.data.mnist import synthetic_mnist
+    x, y = synthetic_mnist(64, seed=2)
+    cfg = TrainConfig(device="cpu", backend="cpu", batch_size=16,
+                      log_interval=0, epochs=2)
+    full = Trainer(cfg)
+    full.train_epoch(x, y, log=lambda *a: None)
+    full.train_epoch(x, y, log=lambda *a: None)
+
+    half = Trainer(cfg)
+    half.train_epoch(x, y, log=lambda *a: None)
+    path = str(tmp_path / "resume.bin")
+    save_checkpoint(half, path)
+
+    resumed = Trainer(cfg)
+    meta = load_checkpoint(resumed, path)
+    assert meta["epoch"] == 1 and resumed.epoch == 1
+    assert "rng" in meta and "torch_cpu" in meta["rng"]
+    resumed.train_epoch(x, y, log=lambda *a: None)
+    assert torch.equal(full.model.params, resumed.model.params)
+    assert resumed.epoch == full.epoch == 2
+    assert resumed.global_step == full.global_step
+
+
+def test_resume_restores_rng_stream(tmp_path):
+    """RNG draws after load reproduce the draws after save."""
+    from parallel_cnn_amd.config import TrainConfig
+    from parallel_cnn_amd.engine.trainer import Trainer
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    cfg = TrainConfig(device="cpu", backend="cpu", log_interval=0)
+    t = Trainer(cfg)
+    torch.manual_seed(77)
+    torch.rand(3)  # advance the stream
+    path = str(tmp_path / "r.bin")
+    save_checkpoint(t, path)
+    expect = torch.rand(4)
+    torch.manual_seed(0)  # clobber
+    load_checkpoint(Trainer(cfg), path)
+    assert torch.equal(torch.rand(4), expect)
