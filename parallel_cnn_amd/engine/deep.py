@@ -236,7 +236,7 @@ class DeepTrainer:
             raise RuntimeError("graph capture requires the hip backend")
         if self.cfg.grad_accum != 1:
             raise RuntimeError("graph capture assumes grad_accum == 1")
-        if self.ctx.world_size > 1 and \
+        if pdist.is_distributed() and \
                 torch.distributed.get_backend() != "nccl":
             raise RuntimeError("graph capture requires RCCL (nccl backend)")
         B = self.ws.max_batch

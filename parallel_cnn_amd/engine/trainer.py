@@ -139,7 +139,7 @@ class Trainer:
                 self._C.hip_wgrad_roles(x, w.a1, w.a2, w.dz, w.dz2, w.dz1,
                                         m.grads, B, self.cfg.wgrad_chunk,
                                         self._wroles, stream)
-            elif self.cfg.overlap_comm and self.ctx.world_size > 1:
+            elif self.cfg.overlap_comm and pdist.is_distributed():
                 # two-bucket overlap (SURVEY §5.8 / north star): the first
                 # bucket all-reduces on the RCCL stream while the rest of
                 # the wgrad still computes on the compute stream
@@ -243,7 +243,7 @@ class Trainer:
             raise RuntimeError("graph capture requires the hip backend")
         if self.cfg.grad_accum != 1:
             raise RuntimeError("graph capture assumes grad_accum == 1")
-        if self.ctx.world_size > 1 and \
+        if pdist.is_distributed() and \
                 torch.distributed.get_backend() != "nccl":
             raise RuntimeError("graph capture requires RCCL (nccl backend)")
         B = self.ws.max_batch
@@ -305,7 +305,7 @@ class Trainer:
                 i = (s % P) * B
                 self.step_graph(x_pool[i:i + B], labels_pool[i:i + B])
             return
-        if self.backend == "hip" and self.ctx.world_size == 1:
+        if self.backend == "hip" and not pdist.is_distributed():
             w = self.ws
             self._C.hip_train_steps(
                 x_pool, labels_pool, self.model.params, self.model.grads,

@@ -41,10 +41,22 @@ def init_from_env(device: str = "auto") -> DistContext:
 
     Backend: RCCL ("nccl") when each rank has its own GPU; gloo otherwise.
     PCNN_DIST_BACKEND overrides (e.g. gloo on a 1-GPU box to exercise the
-    multi-rank engine path with both ranks sharing the device)."""
+    multi-rank engine path with both ranks sharing the device).
+
+    With PCNN_DIST_BACKEND set explicitly, a real process group is
+    initialised even at WORLD_SIZE=1: this runs the exact RCCL code of
+    the 8-GPU job (comm init, in-graph all-reduce, async work handles)
+    on a single-GPU lease, de-risking the first multi-GPU run."""
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    if world <= 1:
+    forced = os.environ.get("PCNN_DIST_BACKEND")
+    if world <= 1 and not forced:
         return DistContext()
+    if world <= 1:
+        # single-rank forced group still needs a rendezvous
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29641")
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     use_gpu = device != "cpu" and torch.cuda.is_available()

@@ -164,3 +164,18 @@ def test_deep_dp_matches_single_rank():
         assert p.exitcode == 0
     assert torch.allclose(params, ref.model.params, atol=1e-5), \
         (params - ref.model.params).abs().max()
+
+
+@pytest.mark.timeout(300)
+def test_scale_check_gloo_ws2():
+    """The multi-GPU preflight tool runs green over gloo at world_size 2
+    (the CPU stand-in for the nccl run the GPU test does)."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29653", "tools/scale_check.py"],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "READY" in out.stdout and "FAIL" not in out.stdout

@@ -414,3 +414,58 @@ def test_bench_contract_single_gpu(device):
     assert r["value"] > 0 and r["ms_per_step"] > 0
     assert r["dtype"] == "bf16"
     assert r["config"]["global_batch"] == 64
+
+
+def _torchrun_ws1_nccl(extra, timeout=600):
+    """Launch bench.py under torchrun at world_size=1 with a FORCED nccl
+    process group: the RCCL comm init, in-step dist.all_reduce, async
+    work handles and (with --use-graph) in-graph collectives all execute
+    on hardware — the same code the 8-GPU driver run uses (VERDICT r1 #1:
+    de-risk the first SCALE run)."""
+    import os
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "1", "--master-addr", "127.0.0.1",
+         "--master-port", "29651", "bench.py", "--gpus", "1",
+         "--steps", "20", "--warmup", "5"] + extra,
+        capture_output=True, text=True, timeout=timeout,
+        env={**os.environ, "PCNN_DIST_BACKEND": "nccl"})
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith("{")][-1]
+    return json.loads(line)
+
+
+def test_nccl_ws1_bench(device):
+    r = _torchrun_ws1_nccl([])
+    assert r["value"] > 0 and r["config"]["backend"] == "hip"
+
+
+def test_nccl_ws1_overlap_comm(device):
+    """overlap_comm over a real RCCL group: async all-reduce on the comm
+    stream + Work.wait ordering before the update."""
+    r = _torchrun_ws1_nccl(["--overlap-comm"])
+    assert r["value"] > 0
+
+
+def test_nccl_ws1_graph_capture(device):
+    """hipGraph capture with the RCCL all-reduce inside the graph body."""
+    r = _torchrun_ws1_nccl(["--use-graph"])
+    assert r["value"] > 0 and r["config"]["hipgraph"] is True
+
+
+def test_nccl_ws1_deepcnn(device):
+    r = _torchrun_ws1_nccl(["--model", "deepcnn", "--batch-size", "64"])
+    assert r["value"] > 0
+
+
+def test_scale_check_preflight(device):
+    """tools/scale_check.py must report READY on a GPU box (nccl group,
+    bucket all-reduces verified)."""
+    out = subprocess.run(
+        [sys.executable, "tools/scale_check.py"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "READY" in out.stdout
+    assert "backend=nccl" in out.stdout
+    assert "FAIL" not in out.stdout
