@@ -24,11 +24,15 @@ HIP_SOURCES = [
 ]
 
 
+FORCE = os.environ.get("PCNN_FORCE_REBUILD") == "1"
+
+
 def compile_hip_objects():
     objs = []
     for src in HIP_SOURCES:
         obj = os.path.splitext(src)[0] + ".o"
-        stale = (not os.path.exists(obj)
+        stale = (FORCE
+                 or not os.path.exists(obj)
                  or os.path.getmtime(obj) < os.path.getmtime(src)
                  or os.path.getmtime(obj) < os.path.getmtime(
                      os.path.join(HERE, "csrc", "lenet_dims.h")))
@@ -57,7 +61,8 @@ def build_native_cli(hip_objs):
     src = os.path.join(HERE, "tools", "pcnn_train.cpp")
     out = os.path.join(HERE, "tools", "pcnn_train")
     try:
-        stale = (not os.path.exists(out)
+        stale = (FORCE
+                 or not os.path.exists(out)
                  or os.path.getmtime(out) < os.path.getmtime(src)
                  or any(os.path.getmtime(out) < os.path.getmtime(o)
                         for o in hip_objs))
