@@ -52,17 +52,23 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    void* C, long long M, int K, int N, int ldA, int ldC,
                    int b_kxn, int epilogue, int actf, void* stream);
-int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
-                      const float* bias, void* C, long long M, int K, int N,
-                      int ldA, int ldC, int b_kxn, int epilogue,
-                      const void* imx, int XH, int XW, int XC, int XK,
-                      int XP, int actf, void* stream);
-int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
-                            float* part, long long M, int KcP, int N, int MS,
-                            const void* imx, int XH, int XW, int XC, int XK,
-                            int XP, int actf, void* stream);
+int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
+                       const float* bias, void* C, long long M, int K, int N,
+                       int ldA, int ldC, int b_kxn, int epilogue,
+                       const void* imx, int XH, int XW, int XC, int XK,
+                       int XP, const void* epi, int actf, void* stream);
+int pcnn_deep_wgrad_gemm_ex2(const void* cols, const void* dpre, float* dW,
+                             float* part, long long M, int KcP, int N,
+                             int MS, const void* imx, int XH, int XW, int XC,
+                             int XK, int XP, float* db, int actf,
+                             void* stream);
 int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
                       void* stream);
+int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
+                       const int* R, const int* C, const int* K,
+                       const int* Cin, const long long* w_off,
+                       const long long* bf_off, const long long* bfT_off,
+                       const long long* rot_off, void* stream);
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                          float* part, long long M, int KcP, int N, int MS,
                          int actf, void* stream);
@@ -240,15 +246,17 @@ void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                int64_t M, int64_t K, int64_t N, int64_t ldA, int64_t ldC,
                int64_t b_kxn, int64_t epilogue, int64_t stream,
                at::Tensor Bpre, at::Tensor imx, int64_t XH, int64_t XW,
-               int64_t XC, int64_t XK, int64_t XP) {
-  check_hip(pcnn_deep_gemm_ex(
+               int64_t XC, int64_t XK, int64_t XP, at::Tensor epi) {
+  check_hip(pcnn_deep_gemm_ex2(
                 A.data_ptr(), Bsrc.data_ptr<float>(),
                 Bpre.numel() ? Bpre.data_ptr() : nullptr,
                 bias.numel() ? bias.data_ptr<float>() : nullptr,
                 C.data_ptr(), M, (int)K, (int)N, (int)ldA, (int)ldC,
                 (int)b_kxn, (int)epilogue,
                 imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
-                (int)XC, (int)XK, (int)XP, act_flag(A), (void*)stream),
+                (int)XC, (int)XK, (int)XP,
+                epi.numel() ? epi.data_ptr() : nullptr, act_flag(A),
+                (void*)stream),
             "deep_gemm");
 }
 
@@ -260,20 +268,53 @@ void deep_cast_wt(at::Tensor W, at::Tensor out, at::Tensor outT, int64_t R,
             "deep_cast_wt");
 }
 
+void deep_cast_all(at::Tensor params, at::Tensor wbuf,
+                   std::vector<int64_t> R, std::vector<int64_t> C,
+                   std::vector<int64_t> K, std::vector<int64_t> Cin,
+                   std::vector<int64_t> w_off, std::vector<int64_t> bf_off,
+                   std::vector<int64_t> bfT_off,
+                   std::vector<int64_t> rot_off, int64_t stream) {
+  const size_t n = R.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "deep_cast_all: 1..8 stages");
+  TORCH_CHECK(C.size() == n && K.size() == n && Cin.size() == n &&
+                  w_off.size() == n && bf_off.size() == n &&
+                  bfT_off.size() == n && rot_off.size() == n,
+              "deep_cast_all: descriptor length mismatch");
+  int Ri[8], Ci[8], Ki[8], Cini[8];
+  long long wo[8], bo[8], bto[8], ro[8];
+  for (size_t s = 0; s < n; ++s) {
+    Ri[s] = (int)R[s];
+    Ci[s] = (int)C[s];
+    Ki[s] = (int)K[s];
+    Cini[s] = (int)Cin[s];
+    wo[s] = w_off[s];
+    bo[s] = bf_off[s];
+    bto[s] = bfT_off[s];
+    ro[s] = rot_off[s];
+  }
+  check_hip(pcnn_deep_cast_all(params.data_ptr<float>(), wbuf.data_ptr(),
+                               (int)n, Ri, Ci, Ki, Cini, wo, bo, bto, ro,
+                               (void*)stream),
+            "deep_cast_all");
+}
+
 void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
                      int64_t M, int64_t KcP, int64_t N, int64_t MS,
                      int64_t stream, at::Tensor imx, int64_t XH, int64_t XW,
-                     int64_t XC, int64_t XK, int64_t XP, at::Tensor part) {
+                     int64_t XC, int64_t XK, int64_t XP, at::Tensor part,
+                     at::Tensor db) {
   float* pp = nullptr;
   if (part.numel()) {
     TORCH_CHECK(part.numel() >= MS * KcP * N, "wgrad slab scratch too small");
     pp = part.data_ptr<float>();
   }
-  check_hip(pcnn_deep_wgrad_gemm_ex(
+  check_hip(pcnn_deep_wgrad_gemm_ex2(
                 cols.data_ptr(), dpre.data_ptr(), dW.data_ptr<float>(), pp, M,
                 (int)KcP, (int)N, (int)MS,
                 imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
-                (int)XC, (int)XK, (int)XP, act_flag(cols), (void*)stream),
+                (int)XC, (int)XK, (int)XP,
+                db.numel() ? db.data_ptr<float>() : nullptr,
+                act_flag(cols), (void*)stream),
             "deep_wgrad_gemm");
 }
 
@@ -416,14 +457,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("Bpre") = at::empty({0}),
         py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
         py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
-        py::arg("XP") = 0);
+        py::arg("XP") = 0, py::arg("epi") = at::empty({0}));
   m.def("deep_cast_wt", &deep_cast_wt);
+  m.def("deep_cast_all", &deep_cast_all);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm, py::arg("cols"),
         py::arg("dpre"), py::arg("dW"), py::arg("M"), py::arg("KcP"),
         py::arg("N"), py::arg("MS"), py::arg("stream"),
         py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
         py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
-        py::arg("XP") = 0, py::arg("part") = at::empty({0}));
+        py::arg("XP") = 0, py::arg("part") = at::empty({0}),
+        py::arg("db") = at::empty({0}));
   m.def("deep_colsum", &deep_colsum);
   m.def("deep_col2im_sigbwd", &deep_col2im_sigbwd);
   m.def("deep_pool_fwd", &deep_pool_fwd);

@@ -204,41 +204,37 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
   }
 }
 
-// 8-wide im2col gather for implicit-GEMM staging: returns cols[m][kc0..+8)
-// computed directly from NHWC x (fast path: one 16B load when the span
-// stays in one kernel row and in-image and Cin % 8 == 0).
+// 8-wide implicit-im2col gather (GEMM A-operand staged straight from the
+// NHWC activation, no materialized cols buffer).  Requires Cin % 8 == 0
+// (launcher-enforced): an 8-span then always lies inside ONE input pixel,
+// so it is a single clamped UNCONDITIONAL 16B load with a select-to-zero
+// afterwards — the round-1 implicit path lost to the materialized one
+// because its guarded gather de-pipelined the K-loop (hipcc branches
+// around conditional loads); the clamp+select idiom is what fixed the
+// materialized path and is applied here.  The (i, j, ci) decode is two
+// magic-multiply divisions (fd_cin = fdiv_magic(Cin), fd_k =
+// fdiv_magic(K)).  valid=false spans (image border, kc >= Kc pad, m past
+// M) read a clamped in-bounds address and are zeroed after the load.
 template <typename act_t>
-__device__ __forceinline__ void im2col8(const act_t* __restrict__ x, int H,
-                                        int W, int Cin, int K, int P, int b,
-                                        int oh, int ow, int kc0, int Kc,
-                                        float* v8) {
-  const int rowc = K * Cin;
-  if ((Cin % 8) == 0 && kc0 + 8 <= Kc && (kc0 % rowc) + 8 <= rowc) {
-    const int i = kc0 / rowc;
-    const int ih = oh + i - P;
-    const int t0 = kc0 - i * rowc;
-    const int j0 = t0 / Cin;
-    const int j7 = (t0 + 7) / Cin;
-    if (ih >= 0 && ih < H && (ow + j0 - P) >= 0 && (ow + j7 - P) < W) {
-      ld8v(x + (((long long)b * H + ih) * W + (ow - P)) * Cin + t0, v8);
-      return;
-    }
-  }
+__device__ __forceinline__ void im2col8f(const act_t* __restrict__ x, int H,
+                                         int W, int Cin, int K, int P, int b,
+                                         int oh, int ow, int kc0, int Kc,
+                                         unsigned long long fd_cin,
+                                         unsigned long long fd_k, bool valid,
+                                         float* v8) {
+  const unsigned p = fdiv((unsigned)kc0, fd_cin);
+  const int ci = kc0 - (int)p * Cin;
+  const unsigned pi = fdiv(p, fd_k);
+  const int j = (int)(p - pi * (unsigned)K);
+  const int i = (int)pi;
+  const int ih = oh + i - P;
+  const int iw = ow + j - P;
+  const int ihc = min(max(ih, 0), H - 1);
+  const int iwc = min(max(iw, 0), W - 1);
+  ld8v(x + (((long long)b * H + ihc) * W + iwc) * Cin + ci, v8);
+  if (!(valid && kc0 < Kc && ih == ihc && iw == iwc)) {
 #pragma unroll
-  for (int u = 0; u < 8; ++u) {
-    const int kc = kc0 + u;
-    float val = 0.f;
-    if (kc < Kc) {
-      const int i = kc / rowc;
-      const int t = kc - i * rowc;
-      const int j = t / Cin;
-      const int ci = t - j * Cin;
-      const int ih = oh + i - P;
-      const int iw = ow + j - P;
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-        val = ldf(x + (((long long)b * H + ih) * W + iw) * Cin + ci);
-    }
-    v8[u] = val;
+    for (int u = 0; u < 8; ++u) v8[u] = 0.f;
   }
 }
 
@@ -273,14 +269,20 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 //   A: act_t, row-major, leading dim ldA (>= K, multiple of 32 used).
 //   Bsrc: fp32.  b_kxn = true: Bsrc is [K][N] (stage-transposed to LDS);
 //                 false: Bsrc is [N][K] (staged directly).
-//   epilogue: 0 = plain store, 1 = bias + sigmoid.
+//   epilogue: 0 = plain store, 1 = bias + sigmoid,
+//             2 = sigmoid-backward:  C = acc * epi * (1 - epi)   (epi is
+//                 the sigmoid OUTPUT tensor, same [M][ldC] shape as C —
+//                 used by the implicit dgrad-as-conv, which writes the
+//                 previous stage's pool preact grad directly and kills
+//                 the dcols round-trip + k_col2im_sigbwd pass).
 //   C: act_t [M][ldC].
 // Bpre (optional): pre-cast bf16 B in [N][K] row-per-output-column layout;
 // when non-null it replaces Bsrc/b_kxn and stages with two 16B copies per
 // thread per tile.
 // imx != null: the A operand is the im2col view of NHWC imx (implicit
 // GEMM — no materialized cols buffer); A/ldA are ignored, K = KcP, and
-// the imx geometry is (XH, XW, XC, XK, XP).
+// the imx geometry is (XH, XW, XC, XK, XP) with XC % 8 == 0
+// (launcher-enforced; fd_cin/fd_k are the magic divisors).
 // TBM: M-tile (64 or 128).  At 128 each wave owns two 16-row fragments
 // (16 MFMAs per K-step — double the compute per barrier pair) at +9 KB
 // LDS; the launcher picks it for large-M calls.
@@ -290,7 +292,8 @@ __global__ __launch_bounds__(256) void k_gemm(
     const __bf16* __restrict__ Bpre, const float* __restrict__ bias,
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
-    int XC, int XK, int XP) {
+    int XC, int XK, int XP, const act_t* __restrict__ epi,
+    unsigned long long fd_cin, unsigned long long fd_k) {
   __shared__ GemmLdsT<TBM, BKC> Lb[2];  // double-buffered tiles
   constexpr int RF = TBM / 64;        // row fragments per wave
   constexpr int TPR = 256 / TBM;      // staging threads per A row
@@ -323,11 +326,14 @@ __global__ __launch_bounds__(256) void k_gemm(
   float ra[NCH][8];
   float rb[NCHB][8];
   const long long m_a = m0 + row_a;
-  // implicit-A: decode this thread's im2col row position once
+  // implicit-A: decode this thread's im2col row position once (from the
+  // CLAMPED row — loads are unconditional, so the address must stay
+  // in-bounds even for m_a >= M rows, which are zero-selected)
   int ib = 0, ioh = 0, iow = 0, iKc = 0;
   if (imx != nullptr) {
-    iow = (int)(m_a % XW);
-    const long long bh = m_a / XW;
+    const long long mm = m_a < M ? m_a : M - 1;
+    iow = (int)(mm % XW);
+    const long long bh = mm / XW;
     ioh = (int)(bh % XH);
     ib = (int)(bh / XH);
     iKc = XK * XK * XC;
@@ -339,11 +345,8 @@ __global__ __launch_bounds__(256) void k_gemm(
       const int kk = kq + h * 8;
       const bool ok = m_a < M && (kt + kk) < K;
       if (imx != nullptr) {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
-        if (ok)
-          im2col8(imx, XH, XW, XC, XK, XP, ib, ioh, iow, kt + kk, iKc,
-                  ra[h]);
+        im2col8f(imx, XH, XW, XC, XK, XP, ib, ioh, iow, kt + kk, iKc,
+                 fd_cin, fd_k, ok, ra[h]);
       } else {
         // clamped unconditional load + select-to-zero: a load behind a
         // thread-varying guard de-pipelines the whole K-loop (hipcc
@@ -487,7 +490,12 @@ __global__ __launch_bounds__(256) void k_gemm(
           const long long m = m0 + crow + r;
           if (m < M && n < N) {
             float v = acc[rf][f][r];
-            if (epilogue == 1) v = sigmoidf_dev(v + bias[n]);
+            if (epilogue == 1) {
+              v = sigmoidf_dev(v + bias[n]);
+            } else if (epilogue == 2) {
+              const float e = ldf(epi + m * ldC + n);
+              v = v * e * (1.0f - e);
+            }
             stf(C + m * ldC + n, v);
           }
         }
@@ -500,13 +508,23 @@ __global__ __launch_bounds__(256) void k_gemm(
 // Both operands are transpose-staged into the [row][k=m] LDS image; the
 // M dimension is the MFMA K axis.  Grid: (kc-tiles) x (n-tiles) x MS
 // M-slices; fp32 hardware atomics combine slices.
-// imx != null: the cols operand is the im2col view of NHWC imx.
+// imx != null: the cols operand is the im2col view of NHWC imx
+// (XC % 8 == 0, launcher-enforced): each thread's kc-span is FIXED, so
+// its (i, j, ci) decode hoists out of the M-walk entirely; per chunk only
+// the (b, oh, ow) row decode (two magic divisions) remains, and the
+// gather is a clamped unconditional 16B load + select-to-zero.
+// db != null: the conv BIAS grad db[n] += sum_m dpre[m][n] is folded in
+// (replacing the separate k_colsum/k_colsum_fin passes): every dpre
+// element is staged through write_lds exactly once per (kct == 0) WG,
+// accumulated per-thread and LDS-reduced after the MFMA loop.
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_wgrad_gemm(
     const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
     float* __restrict__ dW, float* __restrict__ part, long long M, int KcP,
     int N, int MS, const act_t* __restrict__ imx, int XH, int XW, int XC,
-    int XK, int XP) {
+    int XK, int XP, float* __restrict__ db, unsigned long long fd_cin,
+    unsigned long long fd_k, unsigned long long fd_xw,
+    unsigned long long fd_xh) {
   __shared__ GemmLds Lb[2];  // double-buffered M-chunks
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -536,6 +554,27 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
   float rd[2][8];  // dpre prefetch
 
   const int iKc = XK * XK * XC;
+  // implicit-A: this thread's kc spans are FIXED — hoist the (i, j, ci)
+  // decode out of the M-walk (two magic divisions per span, once)
+  int hi_i[2], hi_j[2], hi_ci[2];
+  bool hi_in[2];
+  if (imx != nullptr) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int kc = kc0 + cq + h * 8;
+      const unsigned p = fdiv((unsigned)kc, fd_cin);
+      hi_ci[h] = kc - (int)p * XC;
+      const unsigned pi = fdiv(p, fd_k);
+      hi_j[h] = (int)(p - pi * (unsigned)XK);
+      hi_i[h] = (int)pi;
+      hi_in[h] = kc < iKc;
+    }
+  }
+  // fused colsum partials: thread owns columns [n0+cq, +16)
+  float dbacc[16];
+  const bool do_db = db != nullptr && kct == 0;
+#pragma unroll
+  for (int u = 0; u < 16; ++u) dbacc[u] = 0.f;
 
   auto load_regs = [&](long long mt) {
     const long long m = mt + row_s;
@@ -545,17 +584,27 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     // and N % 16 == 0, so an 8-span is either fully in or fully out.
     const bool ok = m < m_hi;
     const long long mm = ok ? m : m_hi - 1;
+    int ib = 0, ioh = 0, iow = 0;
+    if (imx != nullptr) {
+      const unsigned bhq = fdiv((unsigned)mm, fd_xw);
+      iow = (int)((unsigned)mm - bhq * (unsigned)XW);
+      const unsigned bq = fdiv(bhq, fd_xh);
+      ioh = (int)(bhq - bq * (unsigned)XH);
+      ib = (int)bq;
+    }
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int c = cq + h * 8;
       if (imx != nullptr) {
+        const int ih = ioh + hi_i[h] - XP;
+        const int iw = iow + hi_j[h] - XP;
+        const int ihc = min(max(ih, 0), XH - 1);
+        const int iwc = min(max(iw, 0), XW - 1);
+        ld8v(imx + (((long long)ib * XH + ihc) * XW + iwc) * XC + hi_ci[h],
+             rc[h]);
+        if (!(ok && hi_in[h] && ih == ihc && iw == iwc)) {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
-        if (ok && kc0 + c < KcP) {
-          const int iow = (int)(m % XW);
-          const long long bh = m / XW;
-          im2col8(imx, XH, XW, XC, XK, XP, (int)(bh / XH),
-                  (int)(bh % XH), iow, kc0 + c, iKc, rc[h]);
+          for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
         }
       } else {
         const bool cok = (kc0 + c + 8) <= KcP;
@@ -586,6 +635,7 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
       for (int u = 0; u < 8; ++u) {
         L.As[c + u][row_s] = (__bf16)rc[h][u];
         L.Bs[c + u][row_s] = (__bf16)rd[h][u];
+        if (do_db) dbacc[h * 8 + u] += rd[h][u];
       }
     }
   };
@@ -617,6 +667,22 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
       if (mt + 2 * BK < m_hi) load_regs(mt + 2 * BK);
     }
     __syncthreads();
+  }
+
+  // fused colsum epilogue: fold the per-thread dpre partials across the
+  // 64 staging rows (LDS reuse of the now-idle A tile) and combine into
+  // db with one atomic per column per (kct==0, slice) WG.
+  if (do_db) {
+    float* sc = reinterpret_cast<float*>(&Lb[0]);  // [64 rows][64 cols] fp32
+#pragma unroll
+    for (int u = 0; u < 16; ++u) sc[row_s * 64 + cq + u] = dbacc[u];
+    __syncthreads();
+    if (tid < 64 && (n0 + tid) < N) {
+      float s = 0.f;
+      for (int r = 0; r < 64; ++r) s += sc[r * 64 + tid];
+      if (s != 0.f) unsafeAtomicAdd(&db[n0 + tid], s);
+    }
+    __syncthreads();  // scratch reads done before any later reuse
   }
 
   const int crow = wv * 16 + (lane >> 4) * 4;  // kc within tile
@@ -1173,6 +1239,45 @@ __global__ void k_cast_wt(const float* __restrict__ W,
   outT[(long long)c * R + r] = v;
 }
 
+// Batched weight pre-cast for ALL conv stages in ONE launch (replaces
+// n_stages k_cast_wt dispatches).  For each stage s with fp32 master
+// W[R][C] (R = KcP rows, C = Cout) at params + w_off[s], emits into the
+// shared bf16 scratch buffer:
+//   wbuf + bf_off[s]  : bf16 [R][C]        (dgrad fallback B operand)
+//   wbuf + bfT_off[s] : bf16 [C][R]        (forward GEMM B operand)
+//   wbuf + rot_off[s] : bf16 [Cin][K*K*C]  (implicit dgrad-as-conv B:
+//       row ci, col (i'*K+j')*C + c with i' = K-1-i, j' = K-1-j — the
+//       180-degree-rotated, channel-transposed kernel; pad rows kc >= Kc
+//       have no image here)
+struct CastDescs {
+  int n;                       // stages (<= 8)
+  int R[8], C[8], K[8], Cin[8];
+  long long w_off[8], bf_off[8], bfT_off[8], rot_off[8];
+  long long cum[9];            // cumulative R*C
+};
+__global__ void k_cast_wt_all(const float* __restrict__ params,
+                              __bf16* __restrict__ wbuf, CastDescs d) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= d.cum[d.n]) return;
+  int s = 0;
+  while (i >= d.cum[s + 1]) ++s;
+  const long long q = i - d.cum[s];
+  const int C = d.C[s];
+  const int kc = (int)(q / C);
+  const int c = (int)(q - (long long)kc * C);
+  const __bf16 v = (__bf16)params[d.w_off[s] + q];
+  wbuf[d.bf_off[s] + q] = v;
+  wbuf[d.bfT_off[s] + (long long)c * d.R[s] + kc] = v;
+  const int Cin = d.Cin[s], K = d.K[s];
+  if (kc < K * K * Cin) {
+    const int ci = kc % Cin;
+    const int p = kc / Cin;
+    const int ki = p / K, kj = p - ki * K;
+    const int col = ((K - 1 - ki) * K + (K - 1 - kj)) * C + c;
+    wbuf[d.rot_off[s] + (long long)ci * (K * K * C) + col] = v;
+  }
+}
+
 // Generic SGD apply + zero:  p += step*g; g = 0  over n params.
 __global__ void k_update_n(float* __restrict__ params,
                            float* __restrict__ grads, long long n,
@@ -1250,12 +1355,18 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
-                      const float* bias, void* C, long long M, int K, int N,
-                      int ldA, int ldC, int b_kxn, int epilogue,
-                      const void* imx, int XH, int XW, int XC, int XK,
-                      int XP, int actf, void* stream) {
+int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
+                       const float* bias, void* C, long long M, int K, int N,
+                       int ldA, int ldC, int b_kxn, int epilogue,
+                       const void* imx, int XH, int XW, int XC, int XK,
+                       int XP, const void* epi, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
+  if (imx != nullptr &&
+      ((XC % 8) != 0 || M >= (1LL << 26) || XC > 4096 || XK > 4096))
+    return -2;  // implicit fast path preconditions (engine falls back)
+  if (epilogue == 2 && epi == nullptr) return -3;
+  const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
+  const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
   // 128-row tiles measured neutral-to-negative at this family's shapes
   // (the 2-barrier loop is staging/latency bound, not MFMA bound — see
   // profiles/: MfmaUtil ~3%); the instantiation stays available but the
@@ -1271,7 +1382,7 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                                       (const __bf16*)Bpre, bias, (act_t*)C,
                                       M, K, N, ldA, ldC, b_kxn, epilogue,
                                       (const act_t*)imx, XH, XW, XC, XK,
-                                      XP));
+                                      XP, (const act_t*)epi, fd_cin, fd_k));
   } else {
     PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, BM>), grid, block,
                                       0, (hipStream_t)stream,
@@ -1279,9 +1390,19 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                                       (const __bf16*)Bpre, bias, (act_t*)C,
                                       M, K, N, ldA, ldC, b_kxn, epilogue,
                                       (const act_t*)imx, XH, XW, XC, XK,
-                                      XP));
+                                      XP, (const act_t*)epi, fd_cin, fd_k));
   }
   return (int)hipGetLastError();
+}
+
+int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
+                      const float* bias, void* C, long long M, int K, int N,
+                      int ldA, int ldC, int b_kxn, int epilogue,
+                      const void* imx, int XH, int XW, int XC, int XK,
+                      int XP, int actf, void* stream) {
+  return pcnn_deep_gemm_ex2(A, Bsrc, Bpre, bias, C, M, K, N, ldA, ldC,
+                            b_kxn, epilogue, imx, XH, XW, XC, XK, XP,
+                            nullptr, actf, stream);
 }
 
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
@@ -1301,10 +1422,44 @@ int pcnn_deep_cast_wt(const float* W, void* out, void* outT, int R, int C,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
-                            float* part, long long M, int KcP, int N, int MS,
-                            const void* imx, int XH, int XW, int XC, int XK,
-                            int XP, int actf, void* stream) {
+int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
+                       const int* R, const int* C, const int* K,
+                       const int* Cin, const long long* w_off,
+                       const long long* bf_off, const long long* bfT_off,
+                       const long long* rot_off, void* stream) {
+  if (n_stages < 1 || n_stages > 8) return -2;
+  CastDescs d;
+  d.n = n_stages;
+  d.cum[0] = 0;
+  for (int s = 0; s < n_stages; ++s) {
+    d.R[s] = R[s];
+    d.C[s] = C[s];
+    d.K[s] = K[s];
+    d.Cin[s] = Cin[s];
+    d.w_off[s] = w_off[s];
+    d.bf_off[s] = bf_off[s];
+    d.bfT_off[s] = bfT_off[s];
+    d.rot_off[s] = rot_off[s];
+    d.cum[s + 1] = d.cum[s] + (long long)R[s] * C[s];
+  }
+  dim3 grid((unsigned)((d.cum[n_stages] + 255) / 256)), block(256);
+  hipLaunchKernelGGL(k_cast_wt_all, grid, block, 0, (hipStream_t)stream,
+                     params, (__bf16*)wbuf, d);
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_wgrad_gemm_ex2(const void* cols, const void* dpre, float* dW,
+                             float* part, long long M, int KcP, int N,
+                             int MS, const void* imx, int XH, int XW, int XC,
+                             int XK, int XP, float* db, int actf,
+                             void* stream) {
+  if (imx != nullptr && ((XC % 8) != 0 || M >= (1LL << 26) || XC > 4096 ||
+                         XK > 4096 || XW > 4096 || XH > 4096))
+    return -2;  // implicit fast path preconditions (engine falls back)
+  const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
+  const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
+  const unsigned long long fd_xw = fdiv_magic((unsigned)(XW > 0 ? XW : 1));
+  const unsigned long long fd_xh = fdiv_magic((unsigned)(XH > 0 ? XH : 1));
   const int ktiles = (KcP + BM - 1) / BM;
   const int ntiles = (N + BN - 1) / BN;
   dim3 grid((unsigned)(ktiles * ntiles * MS)), block(256);
@@ -1313,7 +1468,8 @@ int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
                                           (const act_t*)cols,
                                           (const act_t*)dpre, dW, part, M, KcP,
                                           N, MS, (const act_t*)imx, XH, XW, XC,
-                                          XK, XP));
+                                          XK, XP, db, fd_cin, fd_k, fd_xw,
+                                          fd_xh));
   if (part) {
     const long long total = (long long)KcP * N;
     dim3 g2((unsigned)((total / 8 + 255) / 256));
@@ -1321,6 +1477,14 @@ int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
                        total, MS);
   }
   return (int)hipGetLastError();
+}
+
+int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
+                            float* part, long long M, int KcP, int N, int MS,
+                            const void* imx, int XH, int XW, int XC, int XK,
+                            int XP, int actf, void* stream) {
+  return pcnn_deep_wgrad_gemm_ex2(cols, dpre, dW, part, M, KcP, N, MS, imx,
+                                  XH, XW, XC, XK, XP, nullptr, actf, stream);
 }
 
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,

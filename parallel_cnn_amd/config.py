@@ -47,6 +47,11 @@ class TrainConfig:
                                      # bucket while the other wgrad runs
     fuse_wgrad: bool = False         # experimental: conv/pool wgrad inside
                                      # the fwdbwd kernel (measured slower)
+    deep_implicit: bool = True       # DeepCNN: implicit-im2col GEMMs (no
+                                     # materialized cols for Cin%8==0
+                                     # stages, dgrad-as-conv w/ fused
+                                     # sigmoid-bwd); False = round-1
+                                     # materialized path
 
     # io / observability
     log_interval: int = 100          # steps between loss readouts

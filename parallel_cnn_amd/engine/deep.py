@@ -3,11 +3,15 @@
 Backends: "hip" (gfx950 kernels, csrc/hip/conv_kernels.hip) and "torchref"
 (the fp32 oracle, also the CPU execution path for this family).
 
-Step sequence (hip): per stage {im2col, fused GEMM+bias+sigmoid, pool};
-fc fwd (+ residual loss); then backward {fc bwd/wgrad; per stage: pool
-wgrad, pool bwd (in-place over the conv activation), conv wgrad GEMM +
-colsum bias, dgrad GEMM (into the cols buffer) + col2im}, one fused DP
-all-reduce of the flat gradient bucket, SGD update.
+Step sequence (hip, deep_implicit=True — the default): one fused weight
+cast; per stage {implicit-im2col GEMM+bias+sigmoid (stage 0 with Cin=3
+materializes cols), pool}; fc fwd (+ residual loss); backward {fc
+bwd/wgrad; per stage: pool wgrad, pool bwd (in-place over the conv
+activation), implicit wgrad GEMM with the bias colsum folded in, and for
+i>0 an implicit dgrad-as-conv GEMM (rotated weight image, sigmoid-bwd
+epilogue) writing dppre[i-1] directly}; one fused DP all-reduce of the
+flat gradient bucket; SGD update.  deep_implicit=False keeps the round-1
+materialized-cols path (dgrad into dcols + col2im).
 """
 from __future__ import annotations
 
@@ -34,18 +38,25 @@ class DeepWorkspace:
         ntiles = (st.cout + 63) // 64
         return max(1, min(256, 512 // (ktiles * ntiles), M // 64))
 
-    def __init__(self, model: DeepCNN, max_batch: int, device, act_dtype):
+    def __init__(self, model: DeepCNN, max_batch: int, device, act_dtype,
+                 implicit: bool = True):
         spec = model.spec
         B = max_batch
         self.max_batch = B
         self.act_dtype = act_dtype
-        self.cols = []    # [M, KcP] per stage (reused as dcols in backward)
+        # a stage can run the implicit-im2col GEMM fast path when its input
+        # channel count is a multiple of 8 (16B spans inside one pixel)
+        self.stage_implicit = [implicit and st.cin % 8 == 0
+                               for st in spec.stages]
+        self.cols = []    # [M, KcP] per materialized stage (None when the
+                          # stage runs implicit — no cols buffer exists)
         self.acts = []    # [B*H*W, Cout] per stage (conv act == NHWC)
         self.pouts = []   # [B*OH*OW, Cout] per stage
         self.dppre = []   # pool preact grads, same shape as pouts
-        for st in spec.stages:
+        for i, st in enumerate(spec.stages):
             M = B * st.h * st.w
-            self.cols.append(torch.empty(M, st.kcp, dtype=act_dtype,
+            self.cols.append(None if self.stage_implicit[i] else
+                             torch.empty(M, st.kcp, dtype=act_dtype,
                                          device=device))
             self.acts.append(torch.empty(M, st.cout, dtype=act_dtype,
                                          device=device))
@@ -54,12 +65,40 @@ class DeepWorkspace:
                                           device=device))
             self.dppre.append(torch.empty(mo, st.cout, dtype=act_dtype,
                                           device=device))
-        # pre-cast bf16 weight copies ([KcP][Cout] and transposed), refreshed
-        # after each update so GEMM B-staging is plain bf16 row copies
-        self.wbf = [torch.empty(st.kcp * st.cout, dtype=torch.bfloat16,
-                                device=device) for st in spec.stages]
-        self.wbfT = [torch.empty(st.cout * st.kcp, dtype=torch.bfloat16,
-                                 device=device) for st in spec.stages]
+        # pre-cast bf16 weight images, refreshed once per step by ONE
+        # k_cast_wt_all launch: per stage [KcP][Cout] (wbf, dgrad
+        # fallback), [Cout][KcP] (wbfT, forward B), and [Cin][K*K*Cout]
+        # (wrot, the rotated/channel-transposed image the implicit
+        # dgrad-as-conv consumes).  All slices of one flat buffer.
+        self.cast_desc = {"R": [], "C": [], "K": [], "Cin": [], "w_off": [],
+                          "bf_off": [], "bfT_off": [], "rot_off": []}
+        off = 0
+        self.wbf, self.wbfT, self.wrot = [], [], []
+        offs = []
+        for i, st in enumerate(spec.stages):
+            d = self.cast_desc
+            d["R"].append(st.kcp)
+            d["C"].append(st.cout)
+            d["K"].append(st.k)
+            d["Cin"].append(st.cin)
+            d["w_off"].append(spec.offsets[f"conv{i}_w"][0])
+            d["bf_off"].append(off)
+            offs.append((off, st.kcp * st.cout))
+            off += st.kcp * st.cout
+            d["bfT_off"].append(off)
+            offs.append((off, st.cout * st.kcp))
+            off += st.cout * st.kcp
+            d["rot_off"].append(off)
+            offs.append((off, st.cin * st.k * st.k * st.cout))
+            off += st.cin * st.k * st.k * st.cout
+        self.wbuf = torch.zeros(off, dtype=torch.bfloat16, device=device)
+        for i in range(len(spec.stages)):
+            o0, n0 = offs[3 * i]
+            o1, n1 = offs[3 * i + 1]
+            o2, n2 = offs[3 * i + 2]
+            self.wbf.append(self.wbuf[o0:o0 + n0])
+            self.wbfT.append(self.wbuf[o1:o1 + n1])
+            self.wrot.append(self.wbuf[o2:o2 + n2])
         self.y = torch.empty(B, spec.n_classes, dtype=torch.float32,
                              device=device)
         self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
@@ -98,7 +137,8 @@ class DeepTrainer:
         self.act_dtype = (act_map[cfg.act_dtype] if backend == "hip"
                           else torch.float32)
         self.ws = DeepWorkspace(self.model, max_batch or cfg.batch_size,
-                                self.device, self.act_dtype)
+                                self.device, self.act_dtype,
+                                implicit=cfg.deep_implicit)
         self._loss_host = 0.0
         self._samples_seen = 0
         self.global_step = 0
@@ -127,11 +167,13 @@ class DeepTrainer:
 
     # ------------------------------------------------------------- hip paths
     def _hip_cast_weights(self):
-        st_h = native.current_stream_handle()
-        for i, st in enumerate(self.model.spec.stages):
-            self._C.deep_cast_wt(self.model.view(f"conv{i}_w"),
-                                 self.ws.wbf[i], self.ws.wbfT[i], st.kcp,
-                                 st.cout, st_h)
+        # ONE launch casts every stage's weights into all three bf16
+        # images (wbf / wbfT / wrot) — was one k_cast_wt per stage
+        d = self.ws.cast_desc
+        self._C.deep_cast_all(self.model.params, self.ws.wbuf, d["R"],
+                              d["C"], d["K"], d["Cin"], d["w_off"],
+                              d["bf_off"], d["bfT_off"], d["rot_off"],
+                              native.current_stream_handle())
 
     def _hip_forward(self, x: torch.Tensor, labels: torch.Tensor, B: int,
                      mode: int):
@@ -140,16 +182,19 @@ class DeepTrainer:
         self._hip_cast_weights()
         src = x
         for i, st in enumerate(spec.stages):
-            # implicit-im2col staging measured SLOWER (256k vs 299k img/s
-            # at bs=256): the conditional gather inside the K-loop breaks
-            # the load pipeline (hipcc branches around the loads).  The
-            # kernels keep the capability; the engine materializes cols.
-            implicit = False
+            # Implicit-im2col staging (clamped unconditional gather +
+            # hoisted magic-div decode) for Cin % 8 == 0 stages: no cols
+            # buffer is written or re-read.  Round 1's implicit attempt
+            # lost because its guarded gather de-pipelined the K-loop;
+            # see im2col8f in conv_kernels.hip.  Stage 0 (Cin=3)
+            # materializes cols — also reused by its wgrad.
+            implicit = w.stage_implicit[i]
             if not implicit:
                 self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin,
                                     st.k, st.pad, st.kcp, st_h)
             M = B * st.h * st.w
-            self._C.deep_gemm(w.cols[i], m.view(f"conv{i}_w"),
+            self._C.deep_gemm(src if implicit else w.cols[i],
+                              m.view(f"conv{i}_w"),
                               m.view(f"conv{i}_b"), w.acts[i], M, st.kcp,
                               st.cout, st.kcp, st.cout, 1, 1, st_h,
                               w.wbfT[i],
@@ -193,34 +238,41 @@ class DeepTrainer:
             ms = self.ws.wgrad_ms(st, M)
             # measured (tools/deep_sweep.py): atomic combine beats the
             # slab+reduce mode at every MS, and ~512 total WGs is optimal
-            # (per-WG fixed costs dominate beyond that)
+            # (per-WG fixed costs dominate beyond that).  The conv BIAS
+            # grad is folded into the wgrad GEMM (db argument) — the
+            # separate k_colsum/k_colsum_fin passes are gone.
             x_in = x if i == 0 else w.pouts[i - 1]
-            implicit = False
-            self._C.deep_wgrad_gemm(w.cols[i], dapre,
+            implicit = w.stage_implicit[i]
+            self._C.deep_wgrad_gemm(x_in if implicit else w.cols[i], dapre,
                                     m.grad_view(f"conv{i}_w"), M, st.kcp,
                                     st.cout, ms, st_h,
                                     x_in if implicit else torch.empty(0),
-                                    st.h, st.w, st.cin, st.k, st.pad)
-            gsum = max(32, min(128, (M * st.cout) // (256 * 96)))
-            self._C.deep_colsum(dapre, w.colsum_part,
-                                m.grad_view(f"conv{i}_b"), M, st.cout, gsum,
-                                st_h)
+                                    st.h, st.w, st.cin, st.k, st.pad,
+                                    db=m.grad_view(f"conv{i}_b"))
             if i > 0:
-                # dgrad into the cols buffer (its forward use is done).
-                # NOT torch.matmul/hipBLASLt: standalone Lt kernels beat
-                # ours 1.6-3.3x on these shapes (tools/membench.py), but
-                # IN-STEP Lt picks a generic tile (110 us vs our 112 for
-                # both dgrads) and the torch dispatch adds ~110 us of
-                # host wall at this step scale — measured 724 -> 835
-                # us/step.  Round-2 option: drive hipBLASLt from C++.
-                self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
-                                  torch.empty(0), w.cols[i], M, st.cout,
-                                  st.kcp, st.cout, st.kcp, 0, 0, st_h,
-                                  w.wbf[i])
-                self._C.deep_col2im_sigbwd(w.cols[i], w.pouts[i - 1],
-                                           w.dppre[i - 1], B, st.h, st.w,
-                                           st.cin, st.k, st.pad, st.kcp,
-                                           st_h)
+                if implicit:
+                    # dgrad-as-conv: implicit im2col of dapre against the
+                    # rotated/channel-transposed weight image, sigmoid-bwd
+                    # fused in the epilogue — writes the previous stage's
+                    # pool preact grad DIRECTLY (the round-1 path's
+                    # dcols write + k_col2im_sigbwd re-read are gone).
+                    kd = st.k * st.k * st.cout
+                    self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
+                                      torch.empty(0), w.dppre[i - 1], M,
+                                      kd, st.cin, kd, st.cin, 0, 2, st_h,
+                                      w.wrot[i], dapre, st.h, st.w,
+                                      st.cout, st.k, st.pad,
+                                      epi=w.pouts[i - 1])
+                else:
+                    # dgrad into the cols buffer (its forward use is done)
+                    self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
+                                      torch.empty(0), w.cols[i], M, st.cout,
+                                      st.kcp, st.cout, st.kcp, 0, 0, st_h,
+                                      w.wbf[i])
+                    self._C.deep_col2im_sigbwd(w.cols[i], w.pouts[i - 1],
+                                               w.dppre[i - 1], B, st.h,
+                                               st.w, st.cin, st.k, st.pad,
+                                               st.kcp, st_h)
 
     # ----------------------------------------------------------------- graph
     def enable_graph(self) -> None:

@@ -77,9 +77,10 @@ def test_deep_im2col_matches_ref(device):
     assert diff < 1e-2, diff
 
 
-def hip_step_pieces(B, act_dtype, device, seed=7):
+def hip_step_pieces(B, act_dtype, device, seed=7, implicit=True):
     cfg = TrainConfig(batch_size=B, device="cuda", backend="hip",
-                      act_dtype=act_dtype, log_interval=0)
+                      act_dtype=act_dtype, log_interval=0,
+                      deep_implicit=implicit)
     t = DeepTrainer(cfg)
     x, labels = synthetic_images(B, 32, 32, 3, seed=seed, structured=False)
     xb, lb = t.stage_batch(x, labels)
@@ -88,11 +89,13 @@ def hip_step_pieces(B, act_dtype, device, seed=7):
     return t, x, labels
 
 
-def test_hip_step_matches_oracle_fp32(device):
+@pytest.mark.parametrize("implicit", [True, False])
+def test_hip_step_matches_oracle_fp32(implicit, device):
     """Full fp32-activation training step vs the torchref oracle: the whole
-    im2col/GEMM/pool/fc forward+backward+update chain."""
+    im2col/GEMM/pool/fc forward+backward+update chain — both the implicit
+    (no cols buffer, dgrad-as-conv) and the materialized path."""
     B = 8
-    t, x, labels = hip_step_pieces(B, "fp32", device)
+    t, x, labels = hip_step_pieces(B, "fp32", device, implicit=implicit)
     ref = DeepCNN(seed=t.cfg.seed)
     xh = x.view(B, 32, 32, 3)
     acts, pouts, y = deep_ref.forward(xh, ref)
@@ -193,3 +196,34 @@ def test_deep_graph_matches_eager(device):
     lg_, ng = tg.consume_loss()
     assert ne == ng == 3 * B
     assert abs(le - lg_) < 1e-3 * max(1.0, abs(le))
+
+
+def test_implicit_matches_materialized(device):
+    """The implicit-im2col step and the round-1 materialized-cols step are
+    the same math: one bf16 step from the same init must land within
+    rounding-grouping tolerance (the implicit dgrad skips the bf16 dcols
+    rounding, so results are close, not bit-equal)."""
+    B = 32
+    ti, x, labels = hip_step_pieces(B, "bf16", device, seed=31,
+                                    implicit=True)
+    tm, _, _ = hip_step_pieces(B, "bf16", device, seed=31, implicit=False)
+    base = DeepCNN(seed=ti.cfg.seed).params
+    di = ti.model.params.cpu() - base
+    dm = tm.model.params.cpu() - base
+    diff = (di - dm).abs().max().item()
+    scale = dm.abs().max().item()
+    assert diff < 3e-2 * max(1e-3, scale), (diff, scale)
+
+
+def test_implicit_infer_matches_materialized(device):
+    """Forward/eval parity between the two engine paths (logits route)."""
+    cfg_i = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                        act_dtype="bf16", log_interval=0)
+    cfg_m = TrainConfig(batch_size=16, device="cuda", backend="hip",
+                        act_dtype="bf16", log_interval=0,
+                        deep_implicit=False)
+    x, _ = synthetic_images(16, 32, 32, 3, seed=41, structured=False)
+    yi = DeepTrainer(cfg_i).forward_logits(x)
+    ym = DeepTrainer(cfg_m).forward_logits(x)
+    diff = (yi - ym).abs().max().item()
+    assert diff < 2e-2, diff
