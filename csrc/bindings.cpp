@@ -52,11 +52,12 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    void* C, long long M, int K, int N, int ldA, int ldC,
                    int b_kxn, int epilogue, int actf, void* stream);
-int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
+int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
                        const float* bias, void* C, long long M, int K, int N,
                        int ldA, int ldC, int b_kxn, int epilogue,
                        const void* imx, int XH, int XW, int XC, int XK,
-                       int XP, const void* epi, int actf, void* stream);
+                       int XP, const void* epi, const float* pw, void* pout,
+                       int PK, int actf, void* stream);
 int pcnn_deep_wgrad_gemm_ex2(const void* cols, const void* dpre, float* dW,
                              float* part, long long M, int KcP, int N,
                              int MS, const void* imx, int XH, int XW, int XC,
@@ -68,7 +69,12 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
                        const int* R, const int* C, const int* K,
                        const int* Cin, const long long* w_off,
                        const long long* bf_off, const long long* bfT_off,
-                       const long long* rot_off, void* stream);
+                       const long long* rot_off, const long long* p8_off,
+                       void* stream);
+int pcnn_deep_pad_channels(const void* x, void* x8, long long npix, int Cin,
+                           int actf, void* stream);
+int pcnn_deep_remap_dw8(float* dW8, float* dW, int KK, int Cin, int Cout,
+                        void* stream);
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,
                          float* part, long long M, int KcP, int N, int MS,
                          int actf, void* stream);
@@ -246,8 +252,9 @@ void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                int64_t M, int64_t K, int64_t N, int64_t ldA, int64_t ldC,
                int64_t b_kxn, int64_t epilogue, int64_t stream,
                at::Tensor Bpre, at::Tensor imx, int64_t XH, int64_t XW,
-               int64_t XC, int64_t XK, int64_t XP, at::Tensor epi) {
-  check_hip(pcnn_deep_gemm_ex2(
+               int64_t XC, int64_t XK, int64_t XP, at::Tensor epi,
+               at::Tensor pw, at::Tensor pout, int64_t PK) {
+  check_hip(pcnn_deep_gemm_ex3(
                 A.data_ptr(), Bsrc.data_ptr<float>(),
                 Bpre.numel() ? Bpre.data_ptr() : nullptr,
                 bias.numel() ? bias.data_ptr<float>() : nullptr,
@@ -255,8 +262,10 @@ void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                 (int)b_kxn, (int)epilogue,
                 imx.numel() ? imx.data_ptr() : nullptr, (int)XH, (int)XW,
                 (int)XC, (int)XK, (int)XP,
-                epi.numel() ? epi.data_ptr() : nullptr, act_flag(A),
-                (void*)stream),
+                epi.numel() ? epi.data_ptr() : nullptr,
+                pw.numel() ? pw.data_ptr<float>() : nullptr,
+                pout.numel() ? pout.data_ptr() : nullptr, (int)PK,
+                act_flag(A), (void*)stream),
             "deep_gemm");
 }
 
@@ -273,15 +282,17 @@ void deep_cast_all(at::Tensor params, at::Tensor wbuf,
                    std::vector<int64_t> K, std::vector<int64_t> Cin,
                    std::vector<int64_t> w_off, std::vector<int64_t> bf_off,
                    std::vector<int64_t> bfT_off,
-                   std::vector<int64_t> rot_off, int64_t stream) {
+                   std::vector<int64_t> rot_off,
+                   std::vector<int64_t> p8_off, int64_t stream) {
   const size_t n = R.size();
   TORCH_CHECK(n >= 1 && n <= 8, "deep_cast_all: 1..8 stages");
   TORCH_CHECK(C.size() == n && K.size() == n && Cin.size() == n &&
                   w_off.size() == n && bf_off.size() == n &&
-                  bfT_off.size() == n && rot_off.size() == n,
+                  bfT_off.size() == n && rot_off.size() == n &&
+                  (p8_off.empty() || p8_off.size() == n),
               "deep_cast_all: descriptor length mismatch");
   int Ri[8], Ci[8], Ki[8], Cini[8];
-  long long wo[8], bo[8], bto[8], ro[8];
+  long long wo[8], bo[8], bto[8], ro[8], po[8];
   for (size_t s = 0; s < n; ++s) {
     Ri[s] = (int)R[s];
     Ci[s] = (int)C[s];
@@ -291,11 +302,27 @@ void deep_cast_all(at::Tensor params, at::Tensor wbuf,
     bo[s] = bf_off[s];
     bto[s] = bfT_off[s];
     ro[s] = rot_off[s];
+    po[s] = p8_off.empty() ? -1 : p8_off[s];
   }
   check_hip(pcnn_deep_cast_all(params.data_ptr<float>(), wbuf.data_ptr(),
                                (int)n, Ri, Ci, Ki, Cini, wo, bo, bto, ro,
-                               (void*)stream),
+                               po, (void*)stream),
             "deep_cast_all");
+}
+
+void deep_pad_channels(at::Tensor x, at::Tensor x8, int64_t npix,
+                       int64_t Cin, int64_t stream) {
+  check_hip(pcnn_deep_pad_channels(x.data_ptr(), x8.data_ptr(), npix,
+                                   (int)Cin, act_flag(x), (void*)stream),
+            "deep_pad_channels");
+}
+
+void deep_remap_dw8(at::Tensor dW8, at::Tensor dW, int64_t KK, int64_t Cin,
+                    int64_t Cout, int64_t stream) {
+  check_hip(pcnn_deep_remap_dw8(dW8.data_ptr<float>(), dW.data_ptr<float>(),
+                                (int)KK, (int)Cin, (int)Cout,
+                                (void*)stream),
+            "deep_remap_dw8");
 }
 
 void deep_wgrad_gemm(at::Tensor cols, at::Tensor dpre, at::Tensor dW,
@@ -457,9 +484,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("Bpre") = at::empty({0}),
         py::arg("imx") = at::empty({0}), py::arg("XH") = 0,
         py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
-        py::arg("XP") = 0, py::arg("epi") = at::empty({0}));
+        py::arg("XP") = 0, py::arg("epi") = at::empty({0}),
+        py::arg("pw") = at::empty({0}), py::arg("pout") = at::empty({0}),
+        py::arg("PK") = 0);
   m.def("deep_cast_wt", &deep_cast_wt);
   m.def("deep_cast_all", &deep_cast_all);
+  m.def("deep_pad_channels", &deep_pad_channels);
+  m.def("deep_remap_dw8", &deep_remap_dw8);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm, py::arg("cols"),
         py::arg("dpre"), py::arg("dW"), py::arg("M"), py::arg("KcP"),
         py::arg("N"), py::arg("MS"), py::arg("stream"),

@@ -274,7 +274,15 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 //                 the sigmoid OUTPUT tensor, same [M][ldC] shape as C —
 //                 used by the implicit dgrad-as-conv, which writes the
 //                 previous stage's pool preact grad directly and kills
-//                 the dcols round-trip + k_col2im_sigbwd pass).
+//                 the dcols round-trip + k_col2im_sigbwd pass),
+//             3 = mode 1 PLUS the trainable pool forward fused in: the
+//                 sigmoid activations are also staged to LDS and each WG
+//                 emits its tile's pool outputs (pw = [PK*PK] kernel then
+//                 scalar bias, pout = pooled sigmoid output).  Launcher-
+//                 enforced preconditions: TBM == 64, one n-tile
+//                 (N <= 64), 64 % XW == 0, (64/XW) % PK == 0,
+//                 XH % PK == 0 — every 2x2 window then lies inside one
+//                 M-tile for all DeepCNN shapes (XW in {8,16,32}).
 //   C: act_t [M][ldC].
 // Bpre (optional): pre-cast bf16 B in [N][K] row-per-output-column layout;
 // when non-null it replaces Bsrc/b_kxn and stages with two 16B copies per
@@ -293,7 +301,8 @@ __global__ __launch_bounds__(256) void k_gemm(
     act_t* __restrict__ C, long long M, int K, int N, int ldA, int ldC,
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
     int XC, int XK, int XP, const act_t* __restrict__ epi,
-    unsigned long long fd_cin, unsigned long long fd_k) {
+    unsigned long long fd_cin, unsigned long long fd_k,
+    const float* __restrict__ pw, act_t* __restrict__ pout, int PK) {
   __shared__ GemmLdsT<TBM, BKC> Lb[2];  // double-buffered tiles
   constexpr int RF = TBM / 64;        // row fragments per wave
   constexpr int TPR = 256 / TBM;      // staging threads per A row
@@ -478,6 +487,9 @@ __global__ __launch_bounds__(256) void k_gemm(
 
   // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
   const int ccol = lane & 15;
+  // mode 3 stages the sigmoid acts to LDS for the fused pool (reuse of
+  // the now-idle tile buffers; [64][64] fp32 = 16 KB < sizeof Lb[0..1])
+  float* const sc = reinterpret_cast<float*>(&Lb[0]);
 #pragma unroll
   for (int rf = 0; rf < RF; ++rf) {
     const int crow = (wv * RF + rf) * 16 + (lane >> 4) * 4;
@@ -490,16 +502,46 @@ __global__ __launch_bounds__(256) void k_gemm(
           const long long m = m0 + crow + r;
           if (m < M && n < N) {
             float v = acc[rf][f][r];
-            if (epilogue == 1) {
+            if (epilogue >= 1 && epilogue != 2) {
               v = sigmoidf_dev(v + bias[n]);
             } else if (epilogue == 2) {
               const float e = ldf(epi + m * ldC + n);
               v = v * e * (1.0f - e);
             }
             stf(C + m * ldC + n, v);
+            if (epilogue == 3) sc[(crow + r) * 64 + n] = v;
           }
         }
       }
+    }
+  }
+  if (epilogue == 3 && TBM == 64) {
+    __syncthreads();
+    // this tile covers whole conv rows (64 % XW == 0) pool-aligned
+    // ((64/XW) % PK == 0): emit its (64 / PK^2) * N pool outputs
+    const int OW = XW / PK;
+    const long long row0 = m0 / XW;       // first (b*XH + h) row
+    const int npos = 64 / (PK * PK);      // pooled positions in the tile
+    const float pb = pw[PK * PK];
+    for (int o = tid; o < npos * N; o += 256) {
+      const int lp = o / N;               // local pooled index
+      const int n = o - lp * N;
+      const int pr = lp / OW;             // local pooled row
+      const int pc = lp - pr * OW;
+      const long long crow0 = row0 + (long long)pr * PK;  // conv row
+      const long long b = crow0 / XH;
+      const int h = (int)(crow0 - b * XH);
+      float a = pb;
+#pragma unroll 4
+      for (int i = 0; i < PK; ++i)
+#pragma unroll 4
+        for (int j = 0; j < PK; ++j) {
+          const int lr = (pr * PK + i) * XW + pc * PK + j;
+          a += pw[i * PK + j] * sc[lr * 64 + n];
+        }
+      // M % 64 == 0 (launcher precondition) — every tile is full
+      const long long pm = (b * (XH / PK) + h / PK) * OW + pc;
+      stf(pout + pm * ldC + n, sigmoidf_dev(a));
     }
   }
 }
@@ -1239,6 +1281,47 @@ __global__ void k_cast_wt(const float* __restrict__ W,
   outT[(long long)c * R + r] = v;
 }
 
+// Channel-pad: x[B*HW][Cin] -> x8[B*HW][8] with zeros in channels >= Cin
+// (Cin < 8).  Lets a Cin=3 input stage run the implicit-im2col GEMM fast
+// path (which needs Cin % 8 == 0) instead of materializing a 50 MB cols
+// buffer: 8-channel padded input is ~4 MB at bs=256 and L2-resident for
+// every re-gather.  One thread per pixel, one 16B store.
+template <typename act_t>
+__global__ void k_pad_channels(const act_t* __restrict__ x,
+                               act_t* __restrict__ x8, long long npix,
+                               int Cin) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= npix) return;
+  act_t out[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u)
+    out[u] = u < Cin ? x[i * Cin + u] : (act_t)0.f;
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(x8 + i * 8) =
+        *reinterpret_cast<const uint4*>(out);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) x8[i * 8 + u] = out[u];
+}
+
+// Remap the 8-padded weight-grad image dW8[(p*8+ci)][Cout] back into the
+// model's [(p*Cin+ci)][Cout] flat-gradient layout (+=), zeroing dW8 for
+// the next step.  Pad-channel rows of dW8 are never touched (their cols
+// values are identically zero and the wgrad atomic skips zeros).
+__global__ void k_remap_dw8(float* __restrict__ dW8,
+                            float* __restrict__ dW, int KK, int Cin,
+                            int Cout) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= KK * Cin * Cout) return;
+  const int c = i % Cout;
+  const int t = i / Cout;
+  const int ci = t % Cin;
+  const int p = t / Cin;
+  float* src = dW8 + ((long long)(p * 8 + ci) * Cout + c);
+  dW[i] += *src;
+  *src = 0.f;
+}
+
 // Batched weight pre-cast for ALL conv stages in ONE launch (replaces
 // n_stages k_cast_wt dispatches).  For each stage s with fp32 master
 // W[R][C] (R = KcP rows, C = Cout) at params + w_off[s], emits into the
@@ -1253,6 +1336,10 @@ struct CastDescs {
   int n;                       // stages (<= 8)
   int R[8], C[8], K[8], Cin[8];
   long long w_off[8], bf_off[8], bfT_off[8], rot_off[8];
+  long long p8_off[8];         // 8-padded [C][K*K*8] image (-1 = none;
+                               // emitted for Cin < 8 stages so their
+                               // forward GEMM can consume the padded
+                               // x8 input via the implicit fast path)
   long long cum[9];            // cumulative R*C
 };
 __global__ void k_cast_wt_all(const float* __restrict__ params,
@@ -1275,6 +1362,10 @@ __global__ void k_cast_wt_all(const float* __restrict__ params,
     const int ki = p / K, kj = p - ki * K;
     const int col = ((K - 1 - ki) * K + (K - 1 - kj)) * C + c;
     wbuf[d.rot_off[s] + (long long)ci * (K * K * C) + col] = v;
+    if (d.p8_off[s] >= 0)
+      // [C][K*K*8] row per output channel; pad entries (ci >= Cin)
+      // stay zero from the buffer's one-time zero init
+      wbuf[d.p8_off[s] + (long long)c * (K * K * 8) + p * 8 + ci] = v;
   }
 }
 
@@ -1355,16 +1446,21 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
+int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
                        const float* bias, void* C, long long M, int K, int N,
                        int ldA, int ldC, int b_kxn, int epilogue,
                        const void* imx, int XH, int XW, int XC, int XK,
-                       int XP, const void* epi, int actf, void* stream) {
+                       int XP, const void* epi, const float* pw, void* pout,
+                       int PK, int actf, void* stream) {
   const int ntiles = (N + BN - 1) / BN;
   if (imx != nullptr &&
       ((XC % 8) != 0 || M >= (1LL << 26) || XC > 4096 || XK > 4096))
     return -2;  // implicit fast path preconditions (engine falls back)
   if (epilogue == 2 && epi == nullptr) return -3;
+  if (epilogue == 3 &&
+      (pw == nullptr || pout == nullptr || N > BN || PK < 1 || XW < 1 ||
+       64 % XW != 0 || (64 / XW) % PK != 0 || XH % PK != 0 || M % 64 != 0))
+    return -4;  // fused-pool preconditions (engine falls back)
   const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
   const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
   // 128-row tiles measured neutral-to-negative at this family's shapes
@@ -1382,7 +1478,8 @@ int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
                                       (const __bf16*)Bpre, bias, (act_t*)C,
                                       M, K, N, ldA, ldC, b_kxn, epilogue,
                                       (const act_t*)imx, XH, XW, XC, XK,
-                                      XP, (const act_t*)epi, fd_cin, fd_k));
+                                      XP, (const act_t*)epi, fd_cin, fd_k,
+                                      pw, (act_t*)pout, PK));
   } else {
     PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, BM>), grid, block,
                                       0, (hipStream_t)stream,
@@ -1390,7 +1487,8 @@ int pcnn_deep_gemm_ex2(const void* A, const float* Bsrc, const void* Bpre,
                                       (const __bf16*)Bpre, bias, (act_t*)C,
                                       M, K, N, ldA, ldC, b_kxn, epilogue,
                                       (const act_t*)imx, XH, XW, XC, XK,
-                                      XP, (const act_t*)epi, fd_cin, fd_k));
+                                      XP, (const act_t*)epi, fd_cin, fd_k,
+                                      pw, (act_t*)pout, PK));
   }
   return (int)hipGetLastError();
 }
@@ -1400,9 +1498,9 @@ int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,
                       int ldA, int ldC, int b_kxn, int epilogue,
                       const void* imx, int XH, int XW, int XC, int XK,
                       int XP, int actf, void* stream) {
-  return pcnn_deep_gemm_ex2(A, Bsrc, Bpre, bias, C, M, K, N, ldA, ldC,
+  return pcnn_deep_gemm_ex3(A, Bsrc, Bpre, bias, C, M, K, N, ldA, ldC,
                             b_kxn, epilogue, imx, XH, XW, XC, XK, XP,
-                            nullptr, actf, stream);
+                            nullptr, nullptr, nullptr, 0, actf, stream);
 }
 
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
@@ -1426,7 +1524,8 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
                        const int* R, const int* C, const int* K,
                        const int* Cin, const long long* w_off,
                        const long long* bf_off, const long long* bfT_off,
-                       const long long* rot_off, void* stream) {
+                       const long long* rot_off, const long long* p8_off,
+                       void* stream) {
   if (n_stages < 1 || n_stages > 8) return -2;
   CastDescs d;
   d.n = n_stages;
@@ -1440,6 +1539,7 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
     d.bf_off[s] = bf_off[s];
     d.bfT_off[s] = bfT_off[s];
     d.rot_off[s] = rot_off[s];
+    d.p8_off[s] = p8_off ? p8_off[s] : -1;
     d.cum[s + 1] = d.cum[s] + (long long)R[s] * C[s];
   }
   dim3 grid((unsigned)((d.cum[n_stages] + 255) / 256)), block(256);
@@ -1485,6 +1585,26 @@ int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
                             int XP, int actf, void* stream) {
   return pcnn_deep_wgrad_gemm_ex2(cols, dpre, dW, part, M, KcP, N, MS, imx,
                                   XH, XW, XC, XK, XP, nullptr, actf, stream);
+}
+
+int pcnn_deep_pad_channels(const void* x, void* x8, long long npix, int Cin,
+                           int actf, void* stream) {
+  if (Cin < 1 || Cin > 8) return -2;
+  dim3 grid((unsigned)((npix + 255) / 256)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pad_channels<act_t>), grid,
+                                          block, 0, (hipStream_t)stream,
+                                          (const act_t*)x, (act_t*)x8, npix,
+                                          Cin));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_remap_dw8(float* dW8, float* dW, int KK, int Cin, int Cout,
+                        void* stream) {
+  const int total = KK * Cin * Cout;
+  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  hipLaunchKernelGGL(k_remap_dw8, grid, block, 0, (hipStream_t)stream, dW8,
+                     dW, KK, Cin, Cout);
+  return (int)hipGetLastError();
 }
 
 int pcnn_deep_wgrad_gemm(const void* cols, const void* dpre, float* dW,

@@ -48,16 +48,38 @@ class DeepWorkspace:
         # channel count is a multiple of 8 (16B spans inside one pixel)
         self.stage_implicit = [implicit and st.cin % 8 == 0
                                for st in spec.stages]
+        # Cin < 8 stages (the 3-channel input) run implicit too, against a
+        # zero-padded 8-channel copy of their input (k_pad_channels) and
+        # an 8-padded weight image; their weight grads come back through
+        # k_remap_dw8.  Kills the last materialized im2col.
+        self.stage_pad8 = [implicit and 0 < st.cin < 8
+                           for st in spec.stages]
+        # pool-forward fuses into the conv GEMM epilogue when every pool
+        # window lies inside one 64-row M-tile (true for all standard
+        # shapes: W in {8,16,32}, pool 2x2) and the stage fits one n-tile
+        self.stage_fusepool = [
+            implicit and st.cout <= 64 and st.w <= 64 and 64 % st.w == 0
+            and (64 // st.w) % st.pool_k == 0 and st.h % st.pool_k == 0
+            and (st.h * st.w) % 64 == 0
+            for st in spec.stages]
         self.cols = []    # [M, KcP] per materialized stage (None when the
                           # stage runs implicit — no cols buffer exists)
         self.acts = []    # [B*H*W, Cout] per stage (conv act == NHWC)
         self.pouts = []   # [B*OH*OW, Cout] per stage
         self.dppre = []   # pool preact grads, same shape as pouts
+        self.x8 = [None] * len(spec.stages)   # padded inputs (pad8 stages)
+        self.dw8 = [None] * len(spec.stages)  # 8-padded wgrad scratch
         for i, st in enumerate(spec.stages):
             M = B * st.h * st.w
-            self.cols.append(None if self.stage_implicit[i] else
-                             torch.empty(M, st.kcp, dtype=act_dtype,
-                                         device=device))
+            self.cols.append(
+                None if (self.stage_implicit[i] or self.stage_pad8[i]) else
+                torch.empty(M, st.kcp, dtype=act_dtype, device=device))
+            if self.stage_pad8[i]:
+                self.x8[i] = torch.empty(M, 8, dtype=act_dtype,
+                                         device=device)
+                self.dw8[i] = torch.zeros(st.k * st.k * 8 * st.cout,
+                                          dtype=torch.float32,
+                                          device=device)
             self.acts.append(torch.empty(M, st.cout, dtype=act_dtype,
                                          device=device))
             mo = B * st.oh * st.ow
@@ -71,9 +93,10 @@ class DeepWorkspace:
         # (wrot, the rotated/channel-transposed image the implicit
         # dgrad-as-conv consumes).  All slices of one flat buffer.
         self.cast_desc = {"R": [], "C": [], "K": [], "Cin": [], "w_off": [],
-                          "bf_off": [], "bfT_off": [], "rot_off": []}
+                          "bf_off": [], "bfT_off": [], "rot_off": [],
+                          "p8_off": []}
         off = 0
-        self.wbf, self.wbfT, self.wrot = [], [], []
+        self.wbf, self.wbfT, self.wrot, self.wp8 = [], [], [], []
         offs = []
         for i, st in enumerate(spec.stages):
             d = self.cast_desc
@@ -91,14 +114,23 @@ class DeepWorkspace:
             d["rot_off"].append(off)
             offs.append((off, st.cin * st.k * st.k * st.cout))
             off += st.cin * st.k * st.k * st.cout
+            if self.stage_pad8[i]:
+                d["p8_off"].append(off)
+                offs.append((off, st.cout * st.k * st.k * 8))
+                off += st.cout * st.k * st.k * 8
+            else:
+                d["p8_off"].append(-1)
+                offs.append((0, 0))
         self.wbuf = torch.zeros(off, dtype=torch.bfloat16, device=device)
         for i in range(len(spec.stages)):
-            o0, n0 = offs[3 * i]
-            o1, n1 = offs[3 * i + 1]
-            o2, n2 = offs[3 * i + 2]
+            o0, n0 = offs[4 * i]
+            o1, n1 = offs[4 * i + 1]
+            o2, n2 = offs[4 * i + 2]
+            o3, n3 = offs[4 * i + 3]
             self.wbf.append(self.wbuf[o0:o0 + n0])
             self.wbfT.append(self.wbuf[o1:o1 + n1])
             self.wrot.append(self.wbuf[o2:o2 + n2])
+            self.wp8.append(self.wbuf[o3:o3 + n3] if n3 else None)
         self.y = torch.empty(B, spec.n_classes, dtype=torch.float32,
                              device=device)
         self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
@@ -173,7 +205,7 @@ class DeepTrainer:
         self._C.deep_cast_all(self.model.params, self.ws.wbuf, d["R"],
                               d["C"], d["K"], d["Cin"], d["w_off"],
                               d["bf_off"], d["bfT_off"], d["rot_off"],
-                              native.current_stream_handle())
+                              d["p8_off"], native.current_stream_handle())
 
     def _hip_forward(self, x: torch.Tensor, labels: torch.Tensor, B: int,
                      mode: int):
@@ -189,20 +221,45 @@ class DeepTrainer:
             # see im2col8f in conv_kernels.hip.  Stage 0 (Cin=3)
             # materializes cols — also reused by its wgrad.
             implicit = w.stage_implicit[i]
-            if not implicit:
+            pad8 = w.stage_pad8[i]
+            M = B * st.h * st.w
+            if pad8:
+                # zero-pad Cin -> 8 so this stage rides the implicit fast
+                # path too (x8 is ~12x smaller than the cols it replaces)
+                self._C.deep_pad_channels(src, w.x8[i], M, st.cin, st_h)
+            elif not implicit:
                 self._C.deep_im2col(src, w.cols[i], B, st.h, st.w, st.cin,
                                     st.k, st.pad, st.kcp, st_h)
-            M = B * st.h * st.w
-            self._C.deep_gemm(src if implicit else w.cols[i],
-                              m.view(f"conv{i}_w"),
-                              m.view(f"conv{i}_b"), w.acts[i], M, st.kcp,
-                              st.cout, st.kcp, st.cout, 1, 1, st_h,
-                              w.wbfT[i],
-                              src if implicit else torch.empty(0),
-                              st.h, st.w, st.cin, st.k, st.pad)
-            self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
-                                  w.pouts[i], B, st.h, st.w, st.cout,
-                                  st.pool_k, st_h)
+            fuse_pool = w.stage_fusepool[i]
+            if pad8:
+                a_src, kdim, b_img, xc = (w.x8[i], st.k * st.k * 8,
+                                          w.wp8[i], 8)
+            elif implicit:
+                a_src, kdim, b_img, xc = src, st.kcp, w.wbfT[i], st.cin
+            else:
+                a_src, kdim, b_img, xc = (w.cols[i], st.kcp, w.wbfT[i],
+                                          st.cin)
+            # epilogue 3: the trainable-pool forward is computed inside
+            # the GEMM from the LDS-staged sigmoid tile (pw is the pool
+            # kernel view; its scalar bias follows contiguously in the
+            # flat param vector, same trick as deep_pool_fwd)
+            self._C.deep_gemm(a_src, m.view(f"conv{i}_w"),
+                              m.view(f"conv{i}_b"), w.acts[i], M, kdim,
+                              st.cout, kdim, st.cout, 1,
+                              3 if fuse_pool else 1, st_h,
+                              b_img,
+                              a_src if (implicit or pad8)
+                              else torch.empty(0),
+                              st.h, st.w, xc, st.k, st.pad,
+                              pw=m.view(f"pool{i}_w") if fuse_pool
+                              else torch.empty(0),
+                              pout=w.pouts[i] if fuse_pool
+                              else torch.empty(0),
+                              PK=st.pool_k if fuse_pool else 0)
+            if not fuse_pool:
+                self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
+                                      w.pouts[i], B, st.h, st.w, st.cout,
+                                      st.pool_k, st_h)
             src = w.pouts[i]
         self._C.deep_fc_fwd(w.pouts[-1], m.view("fc_w"), m.view("fc_b"),
                             labels, w.y, w.dz, w.loss_accum,
@@ -243,12 +300,26 @@ class DeepTrainer:
             # separate k_colsum/k_colsum_fin passes are gone.
             x_in = x if i == 0 else w.pouts[i - 1]
             implicit = w.stage_implicit[i]
-            self._C.deep_wgrad_gemm(x_in if implicit else w.cols[i], dapre,
-                                    m.grad_view(f"conv{i}_w"), M, st.kcp,
-                                    st.cout, ms, st_h,
-                                    x_in if implicit else torch.empty(0),
-                                    st.h, st.w, st.cin, st.k, st.pad,
-                                    db=m.grad_view(f"conv{i}_b"))
+            pad8 = w.stage_pad8[i]
+            if pad8:
+                # implicit against the padded x8 (still valid from this
+                # step's forward); the 8-padded dW lands in scratch and
+                # k_remap_dw8 folds it into the model's flat layout
+                self._C.deep_wgrad_gemm(w.x8[i], dapre, w.dw8[i], M,
+                                        st.k * st.k * 8, st.cout, ms, st_h,
+                                        w.x8[i], st.h, st.w, 8, st.k,
+                                        st.pad,
+                                        db=m.grad_view(f"conv{i}_b"))
+                self._C.deep_remap_dw8(w.dw8[i], m.grad_view(f"conv{i}_w"),
+                                       st.k * st.k, st.cin, st.cout, st_h)
+            else:
+                self._C.deep_wgrad_gemm(
+                    x_in if implicit else w.cols[i], dapre,
+                    m.grad_view(f"conv{i}_w"), M, st.kcp,
+                    st.cout, ms, st_h,
+                    x_in if implicit else torch.empty(0),
+                    st.h, st.w, st.cin, st.k, st.pad,
+                    db=m.grad_view(f"conv{i}_b"))
             if i > 0:
                 if implicit:
                     # dgrad-as-conv: implicit im2col of dapre against the
