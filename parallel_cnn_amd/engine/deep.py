@@ -48,17 +48,24 @@ class DeepWorkspace:
         # channel count is a multiple of 8 (16B spans inside one pixel)
         self.stage_implicit = [implicit and st.cin % 8 == 0
                                for st in spec.stages]
-        # Cin < 8 stages (the 3-channel input) run implicit too, against a
-        # zero-padded 8-channel copy of their input (k_pad_channels) and
-        # an 8-padded weight image; their weight grads come back through
-        # k_remap_dw8.  Kills the last materialized im2col.
-        self.stage_pad8 = [implicit and 0 < st.cin < 8
+        # Cin < 8 stages (the 3-channel input) CAN run implicit too,
+        # against a zero-padded 8-channel copy of their input
+        # (k_pad_channels) and an 8-padded weight image, grads remapped by
+        # k_remap_dw8.  Kills the im2col but grows stage-0 MFMA work
+        # K=96 -> 200; measured net-NEGATIVE at bs=256 (re-confirming the
+        # round-1 evaluation), so default OFF — PCNN_DEEP_PAD8=1 enables.
+        import os
+        pad8_on = os.environ.get("PCNN_DEEP_PAD8", "0") == "1"
+        fusepool_on = os.environ.get("PCNN_DEEP_FUSEPOOL", "1") == "1"
+        self.stage_pad8 = [implicit and pad8_on and 0 < st.cin < 8
                            for st in spec.stages]
+        self._fusepool_on = fusepool_on
         # pool-forward fuses into the conv GEMM epilogue when every pool
         # window lies inside one 64-row M-tile (true for all standard
         # shapes: W in {8,16,32}, pool 2x2) and the stage fits one n-tile
         self.stage_fusepool = [
-            implicit and st.cout <= 64 and st.w <= 64 and 64 % st.w == 0
+            implicit and fusepool_on and st.cout <= 64 and st.w <= 64
+            and 64 % st.w == 0
             and (64 // st.w) % st.pool_k == 0 and st.h % st.pool_k == 0
             and (st.h * st.w) % 64 == 0
             for st in spec.stages]
