@@ -52,12 +52,13 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
 int pcnn_deep_gemm(const void* A, const float* Bsrc, const float* bias,
                    void* C, long long M, int K, int N, int ldA, int ldC,
                    int b_kxn, int epilogue, int actf, void* stream);
-int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
+int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
                        const float* bias, void* C, long long M, int K, int N,
                        int ldA, int ldC, int b_kxn, int epilogue,
                        const void* imx, int XH, int XW, int XC, int XK,
                        int XP, const void* epi, const float* pw, void* pout,
-                       int PK, int actf, void* stream);
+                       int PK, float* c32, long long c32_cap, int actf,
+                       void* stream);
 int pcnn_deep_wgrad_gemm_ex2(const void* cols, const void* dpre, float* dW,
                              float* part, long long M, int KcP, int N,
                              int MS, const void* imx, int XH, int XW, int XC,
@@ -253,8 +254,8 @@ void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                int64_t b_kxn, int64_t epilogue, int64_t stream,
                at::Tensor Bpre, at::Tensor imx, int64_t XH, int64_t XW,
                int64_t XC, int64_t XK, int64_t XP, at::Tensor epi,
-               at::Tensor pw, at::Tensor pout, int64_t PK) {
-  check_hip(pcnn_deep_gemm_ex3(
+               at::Tensor pw, at::Tensor pout, int64_t PK, at::Tensor c32) {
+  check_hip(pcnn_deep_gemm_ex4(
                 A.data_ptr(), Bsrc.data_ptr<float>(),
                 Bpre.numel() ? Bpre.data_ptr() : nullptr,
                 bias.numel() ? bias.data_ptr<float>() : nullptr,
@@ -265,7 +266,8 @@ void deep_gemm(at::Tensor A, at::Tensor Bsrc, at::Tensor bias, at::Tensor C,
                 epi.numel() ? epi.data_ptr() : nullptr,
                 pw.numel() ? pw.data_ptr<float>() : nullptr,
                 pout.numel() ? pout.data_ptr() : nullptr, (int)PK,
-                act_flag(A), (void*)stream),
+                c32.numel() ? c32.data_ptr<float>() : nullptr,
+                (long long)c32.numel(), act_flag(A), (void*)stream),
             "deep_gemm");
 }
 
@@ -486,7 +488,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("XW") = 0, py::arg("XC") = 0, py::arg("XK") = 0,
         py::arg("XP") = 0, py::arg("epi") = at::empty({0}),
         py::arg("pw") = at::empty({0}), py::arg("pout") = at::empty({0}),
-        py::arg("PK") = 0);
+        py::arg("PK") = 0, py::arg("c32") = at::empty({0}));
   m.def("deep_cast_wt", &deep_cast_wt);
   m.def("deep_cast_all", &deep_cast_all);
   m.def("deep_pad_channels", &deep_pad_channels);

@@ -294,6 +294,14 @@ __device__ __forceinline__ bf16x8 frag_from_lds(const __bf16* row, int k0) {
 // TBM: M-tile (64 or 128).  At 128 each wave owns two 16-row fragments
 // (16 MFMAs per K-step — double the compute per barrier pair) at +9 KB
 // LDS; the launcher picks it for large-M calls.
+// Split-K (c32 != null): a launch whose (mtiles x ntiles) grid cannot
+// fill 256 CUs fully exposes the per-iteration stage/barrier stall (a
+// 64-WG stage-2 GEMM measured 47 us vs 20 us for the same FLOPs at 1024
+// WGs).  The grid gains a K-chunk axis: each WG covers tpc BKC-tiles of
+// K and stores its fp32 partial tile to slab `kchunk` of c32
+// [ks][M][N]; k_split_epi sums the slabs and applies the epilogue
+// (deterministic — no atomics, so graph replay and tests stay bitwise
+// reproducible).
 template <typename act_t, int TBM>
 __global__ __launch_bounds__(256) void k_gemm(
     const act_t* __restrict__ A, const float* __restrict__ Bsrc,
@@ -302,7 +310,8 @@ __global__ __launch_bounds__(256) void k_gemm(
     int b_kxn, int epilogue, const act_t* __restrict__ imx, int XH, int XW,
     int XC, int XK, int XP, const act_t* __restrict__ epi,
     unsigned long long fd_cin, unsigned long long fd_k,
-    const float* __restrict__ pw, act_t* __restrict__ pout, int PK) {
+    const float* __restrict__ pw, act_t* __restrict__ pout, int PK,
+    float* __restrict__ c32, int tpc) {
   __shared__ GemmLdsT<TBM, BKC> Lb[2];  // double-buffered tiles
   constexpr int RF = TBM / 64;        // row fragments per wave
   constexpr int TPR = 256 / TBM;      // staging threads per A row
@@ -314,8 +323,14 @@ __global__ __launch_bounds__(256) void k_gemm(
   const int lane = tid & 63;
   const int wv = tid >> 6;
   const int ntiles = (N + BN - 1) / BN;
-  const long long mtile = blockIdx.x / ntiles;
-  const int ntile = (int)(blockIdx.x % ntiles);
+  const long long mtiles = (M + TBM - 1) / TBM;
+  const long long mn = mtiles * ntiles;
+  const int kchunk = (int)(blockIdx.x / mn);
+  const long long rest = blockIdx.x % mn;
+  const long long mtile = rest / ntiles;
+  const int ntile = (int)(rest % ntiles);
+  const int k_lo = c32 ? kchunk * tpc * BKC : 0;
+  const int k_hi = c32 ? min(K, k_lo + tpc * BKC) : K;
   const long long m0 = mtile * TBM;
   const int n0 = ntile * BN;
   const int nf = min(BN, N - n0) / 16;  // fragments along N (N % 16 == 0)
@@ -452,12 +467,12 @@ __global__ __launch_bounds__(256) void k_gemm(
   // is written to Lb[1-p] and tile t+2's loads are issued — the
   // s_waitcnt for a tile's global loads lands ~one iteration after
   // issue instead of right after the MFMA block.
-  load_regs(0);
+  load_regs(k_lo);
   write_lds(Lb[0]);
-  if (BKC < K) load_regs(BKC);  // issue tile 1 (skip for single-tile K)
+  if (k_lo + BKC < k_hi) load_regs(k_lo + BKC);
   __syncthreads();              // Lb[0] visible
   int p = 0;
-  for (int kt = 0; kt < K; kt += BKC, p ^= 1) {
+  for (int kt = k_lo; kt < k_hi; kt += BKC, p ^= 1) {
     auto& L = Lb[p];
     // wave wv owns C rows [wv*16*RF, +16*RF); BKC/32 32-deep MFMA sub-steps
 #pragma unroll
@@ -478,11 +493,32 @@ __global__ __launch_bounds__(256) void k_gemm(
         }
       }
     }
-    if (kt + BKC < K) {
+    if (kt + BKC < k_hi) {
       write_lds(Lb[p ^ 1]);  // waits on tile t+1's loads here
-      if (kt + 2 * BKC < K) load_regs(kt + 2 * BKC);
+      if (kt + 2 * BKC < k_hi) load_regs(kt + 2 * BKC);
     }
     __syncthreads();  // reads of Lb[p] done AND Lb[1-p] complete
+  }
+
+  // split-K: store the fp32 partial tile to this chunk's slab and exit
+  if (c32 != nullptr) {
+    float* slab = c32 + (long long)kchunk * M * N;
+#pragma unroll
+    for (int rf = 0; rf < RF; ++rf) {
+      const int crow = (wv * RF + rf) * 16 + ((tid & 63) >> 4) * 4;
+#pragma unroll
+      for (int f = 0; f < BN / 16; ++f) {
+        if (f < nf) {
+          const int n = n0 + f * 16 + (tid & 15);
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const long long m = m0 + crow + r;
+            if (m < M && n < N) slab[m * N + n] = acc[rf][f][r];
+          }
+        }
+      }
+    }
+    return;
   }
 
   // epilogue: lane l, reg r -> C[row=(l>>4)*4+r][col=l&15] of its fragment
@@ -1281,6 +1317,107 @@ __global__ void k_cast_wt(const float* __restrict__ W,
   outT[(long long)c * R + r] = v;
 }
 
+// Combine split-K slabs (c32 [KS][M][N] fp32) and apply the GEMM
+// epilogue.  mode 0: plain store; 1: C = sigmoid(sum + bias); 2: C =
+// sum * epi * (1-epi); 3: mode 1 PLUS the trainable-pool forward (one
+// thread per pooled position x 8 channels — it writes the PK*PK acts it
+// combined and the pooled output; no tile-alignment requirement, the
+// pooling reads c32 global directly).
+template <typename act_t>
+__global__ void k_split_epi(const float* __restrict__ c32, int KS,
+                            act_t* __restrict__ C, long long M, int N,
+                            const float* __restrict__ bias,
+                            const act_t* __restrict__ epi, int epilogue,
+                            const float* __restrict__ pw,
+                            act_t* __restrict__ pout, int PK, int XH,
+                            int XW) {
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long MN = M * N;
+  if (epilogue == 3) {
+    const int OW = XW / PK, OH = XH / PK;
+    const int n8 = N / 8;
+    const long long total = (M / (PK * PK)) * n8;
+    if (idx >= total) return;
+    const int c0 = (int)(idx % n8) * 8;
+    long long pp = idx / n8;
+    const int q = (int)(pp % OW);
+    long long t = pp / OW;
+    const int p = (int)(t % OH);
+    const long long b = t / OH;
+    float pacc[8];
+    const float pb = pw[PK * PK];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) pacc[u] = pb;
+    for (int i = 0; i < PK; ++i)
+      for (int j = 0; j < PK; ++j) {
+        const long long m = ((b * XH + p * PK + i) * XW + q * PK + j);
+        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        for (int s = 0; s < KS; ++s) {
+          float a8[8];
+          ld8v(c32 + (long long)s * MN + m * N + c0, a8);
+#pragma unroll
+          for (int u = 0; u < 8; ++u) v8[u] += a8[u];
+        }
+        const float wv = pw[i * PK + j];
+        act_t o8[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const float a = sigmoidf_dev(v8[u] + bias[c0 + u]);
+          o8[u] = (act_t)a;
+          pacc[u] += wv * a;
+        }
+        if (sizeof(act_t) == 2)
+          *reinterpret_cast<uint4*>(C + m * N + c0) =
+              *reinterpret_cast<const uint4*>(o8);
+        else
+#pragma unroll
+          for (int u = 0; u < 8; ++u) C[m * N + c0 + u] = o8[u];
+      }
+    act_t po[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) po[u] = (act_t)sigmoidf_dev(pacc[u]);
+    if (sizeof(act_t) == 2)
+      *reinterpret_cast<uint4*>(pout + pp * N + c0) =
+          *reinterpret_cast<const uint4*>(po);
+    else
+#pragma unroll
+      for (int u = 0; u < 8; ++u) pout[pp * N + c0 + u] = po[u];
+    return;
+  }
+  // modes 0/1/2: one thread per 8 consecutive elements of the [M][N]
+  // stream (N % 8 == 0)
+  if (idx * 8 >= MN) return;
+  const long long e0 = idx * 8;
+  const int n = (int)(e0 % N);
+  float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int s = 0; s < KS; ++s) {
+    float a8[8];
+    ld8v(c32 + (long long)s * MN + e0, a8);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v8[u] += a8[u];
+  }
+  act_t o8[8];
+  if (epilogue == 1) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      o8[u] = (act_t)sigmoidf_dev(v8[u] + bias[n + u]);
+  } else if (epilogue == 2) {
+    float e8[8];
+    ld8v(epi + e0, e8);
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      o8[u] = (act_t)(v8[u] * e8[u] * (1.0f - e8[u]));
+  } else {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) o8[u] = (act_t)v8[u];
+  }
+  if (sizeof(act_t) == 2)
+    *reinterpret_cast<uint4*>(C + e0) = *reinterpret_cast<const uint4*>(o8);
+  else
+#pragma unroll
+    for (int u = 0; u < 8; ++u) C[e0 + u] = o8[u];
+}
+
 // Channel-pad: x[B*HW][Cin] -> x8[B*HW][8] with zeros in channels >= Cin
 // (Cin < 8).  Lets a Cin=3 input stage run the implicit-im2col GEMM fast
 // path (which needs Cin % 8 == 0) instead of materializing a 50 MB cols
@@ -1446,12 +1583,13 @@ int pcnn_deep_im2col(const void* x, void* cols, int B, int H, int W, int Cin,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
+int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
                        const float* bias, void* C, long long M, int K, int N,
                        int ldA, int ldC, int b_kxn, int epilogue,
                        const void* imx, int XH, int XW, int XC, int XK,
                        int XP, const void* epi, const float* pw, void* pout,
-                       int PK, int actf, void* stream) {
+                       int PK, float* c32, long long c32_cap, int actf,
+                       void* stream) {
   const int ntiles = (N + BN - 1) / BN;
   if (imx != nullptr &&
       ((XC % 8) != 0 || M >= (1LL << 26) || XC > 4096 || XK > 4096))
@@ -1463,34 +1601,58 @@ int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
     return -4;  // fused-pool preconditions (engine falls back)
   const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
   const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
-  // 128-row tiles measured neutral-to-negative at this family's shapes
-  // (the 2-barrier loop is staging/latency bound, not MFMA bound — see
-  // profiles/: MfmaUtil ~3%); the instantiation stays available but the
-  // selector is off.
-  const bool big = false && M >= 128 * 512;  // measured slower (twice)
-  const int tbm = big ? 128 : BM;
-  const long long mtiles = (M + tbm - 1) / tbm;
-  dim3 grid((unsigned)(mtiles * ntiles)), block(256);
-  if (big) {
-    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, 128>), grid, block,
-                                      0, (hipStream_t)stream,
-                                      (const act_t*)A, Bsrc,
-                                      (const __bf16*)Bpre, bias, (act_t*)C,
-                                      M, K, N, ldA, ldC, b_kxn, epilogue,
-                                      (const act_t*)imx, XH, XW, XC, XK,
-                                      XP, (const act_t*)epi, fd_cin, fd_k,
-                                      pw, (act_t*)pout, PK));
-  } else {
-    PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, BM>), grid, block,
-                                      0, (hipStream_t)stream,
-                                      (const act_t*)A, Bsrc,
-                                      (const __bf16*)Bpre, bias, (act_t*)C,
-                                      M, K, N, ldA, ldC, b_kxn, epilogue,
-                                      (const act_t*)imx, XH, XW, XC, XK,
-                                      XP, (const act_t*)epi, fd_cin, fd_k,
-                                      pw, (act_t*)pout, PK));
+  const long long mtiles = (M + BM - 1) / BM;
+  const long long mn = mtiles * ntiles;
+  // split-K policy: a grid under ~3 WGs/CU leaves most of the chip idle
+  // AND exposes the per-iteration stage/barrier latency (47 us measured
+  // at 64 WGs for 20 us of 1024-WG work); split K until ~1024 WGs.
+  int ks_eff = 1, tpc = 0;
+  const int ktiles = (K + BKC - 1) / BKC;
+  if (c32 != nullptr && mn < 768 && ktiles > 1 && (N % 8) == 0 &&
+      ldC == N) {
+    int ks = (int)(1024 / mn) + 1;
+    if (ks > ktiles) ks = ktiles;
+    tpc = (ktiles + ks - 1) / ks;
+    ks_eff = (ktiles + tpc - 1) / tpc;
+    if ((long long)ks_eff * M * N > c32_cap) {
+      ks_eff = 1;  // scratch too small — plain launch
+      tpc = 0;
+    }
+  }
+  const bool split = ks_eff > 1;
+  dim3 grid((unsigned)(mn * ks_eff)), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_gemm<act_t, BM>), grid, block,
+                                    0, (hipStream_t)stream,
+                                    (const act_t*)A, Bsrc,
+                                    (const __bf16*)Bpre, bias, (act_t*)C,
+                                    M, K, N, ldA, ldC, b_kxn, epilogue,
+                                    (const act_t*)imx, XH, XW, XC, XK,
+                                    XP, (const act_t*)epi, fd_cin, fd_k,
+                                    pw, (act_t*)pout, PK,
+                                    split ? c32 : nullptr, tpc));
+  if (split) {
+    const long long total = (epilogue == 3)
+                                ? (M / (PK * PK)) * (N / 8)
+                                : (M * N) / 8;
+    dim3 g2((unsigned)((total + 255) / 256));
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL(
+                            (k_split_epi<act_t>), g2, block, 0,
+                            (hipStream_t)stream, c32, ks_eff, (act_t*)C, M,
+                            N, bias, (const act_t*)epi, epilogue, pw,
+                            (act_t*)pout, PK, XH, XW));
   }
   return (int)hipGetLastError();
+}
+
+int pcnn_deep_gemm_ex3(const void* A, const float* Bsrc, const void* Bpre,
+                       const float* bias, void* C, long long M, int K, int N,
+                       int ldA, int ldC, int b_kxn, int epilogue,
+                       const void* imx, int XH, int XW, int XC, int XK,
+                       int XP, const void* epi, const float* pw, void* pout,
+                       int PK, int actf, void* stream) {
+  return pcnn_deep_gemm_ex4(A, Bsrc, Bpre, bias, C, M, K, N, ldA, ldC,
+                            b_kxn, epilogue, imx, XH, XW, XC, XK, XP, epi,
+                            pw, pout, PK, nullptr, 0, actf, stream);
 }
 
 int pcnn_deep_gemm_ex(const void* A, const float* Bsrc, const void* Bpre,

@@ -138,6 +138,27 @@ class DeepWorkspace:
             self.wbfT.append(self.wbuf[o1:o1 + n1])
             self.wrot.append(self.wbuf[o2:o2 + n2])
             self.wp8.append(self.wbuf[o3:o3 + n3] if n3 else None)
+        # split-K slab scratch (fp32): sized by replaying the launcher's
+        # split policy over every GEMM shape this batch size will launch
+        def split_need(M, N, K):
+            mn = ((M + 63) // 64) * ((N + 63) // 64)
+            ktiles = (K + 63) // 64
+            if mn >= 768 or ktiles <= 1 or N % 8:
+                return 0
+            ks = min(ktiles, 1024 // mn + 1)
+            tpc = (ktiles + ks - 1) // ks
+            return ((ktiles + tpc - 1) // tpc) * M * N
+
+        need = 0
+        for i, st in enumerate(spec.stages):
+            M = B * st.h * st.w
+            kfwd = st.k * st.k * 8 if self.stage_pad8[i] else st.kcp
+            need = max(need, split_need(M, st.cout, kfwd))
+            if i > 0:
+                need = max(need, split_need(M, st.cin,
+                                            st.k * st.k * st.cout))
+        self.c32 = (torch.empty(need, dtype=torch.float32, device=device)
+                    if need else torch.empty(0, device=device))
         self.y = torch.empty(B, spec.n_classes, dtype=torch.float32,
                              device=device)
         self.dz = torch.empty(B, spec.n_classes, dtype=torch.float32,
@@ -262,7 +283,8 @@ class DeepTrainer:
                               else torch.empty(0),
                               pout=w.pouts[i] if fuse_pool
                               else torch.empty(0),
-                              PK=st.pool_k if fuse_pool else 0)
+                              PK=st.pool_k if fuse_pool else 0,
+                              c32=w.c32)
             if not fuse_pool:
                 self._C.deep_pool_fwd(w.acts[i], m.view(f"pool{i}_w"),
                                       w.pouts[i], B, st.h, st.w, st.cout,
@@ -340,7 +362,7 @@ class DeepTrainer:
                                       kd, st.cin, kd, st.cin, 0, 2, st_h,
                                       w.wrot[i], dapre, st.h, st.w,
                                       st.cout, st.k, st.pad,
-                                      epi=w.pouts[i - 1])
+                                      epi=w.pouts[i - 1], c32=w.c32)
                 else:
                     # dgrad into the cols buffer (its forward use is done)
                     self._C.deep_gemm(dapre, m.view(f"conv{i}_w"),
