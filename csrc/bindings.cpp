@@ -93,6 +93,9 @@ int pcnn_deep_pool_bwd(const void* dppre, const void* a, const float* pw,
 int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
                          int H, int W, int C, int K, int G, int actf,
                          void* stream);
+int pcnn_deep_pool_wbwd(const void* dppre, const void* a, const float* pw,
+                        void* dapre, float* dpw, int B, int H, int W, int C,
+                        int K, int G, int actf, void* stream);
 int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
                      const int* labels, float* yg, float* dzg,
                      float* loss_accum, int* correct_accum, int B, int FCIN,
@@ -395,6 +398,18 @@ void deep_pool_wgrad(at::Tensor dppre, at::Tensor a, at::Tensor dpw,
             "deep_pool_wgrad");
 }
 
+void deep_pool_wbwd(at::Tensor dppre, at::Tensor a, at::Tensor pw,
+                    at::Tensor dapre, at::Tensor dpw, int64_t B, int64_t H,
+                    int64_t W, int64_t C, int64_t K, int64_t G,
+                    int64_t stream) {
+  check_hip(pcnn_deep_pool_wbwd(dppre.data_ptr(), a.data_ptr(),
+                                pw.data_ptr<float>(), dapre.data_ptr(),
+                                dpw.data_ptr<float>(), (int)B, (int)H,
+                                (int)W, (int)C, (int)K, (int)G,
+                                act_flag(a), (void*)stream),
+            "deep_pool_wbwd");
+}
+
 void deep_fc_fwd(at::Tensor flat, at::Tensor fw, at::Tensor fb,
                  at::Tensor labels, at::Tensor y, at::Tensor dz,
                  at::Tensor loss_accum, at::Tensor correct_accum, int64_t B,
@@ -505,6 +520,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_pool_fwd", &deep_pool_fwd);
   m.def("deep_pool_bwd", &deep_pool_bwd);
   m.def("deep_pool_wgrad", &deep_pool_wgrad);
+  m.def("deep_pool_wbwd", &deep_pool_wbwd);
   m.def("deep_fc_fwd", &deep_fc_fwd);
   m.def("deep_fc_bwd", &deep_fc_bwd);
   m.def("deep_fc_wgrad", &deep_fc_wgrad);

@@ -1200,6 +1200,84 @@ __global__ __launch_bounds__(256) void k_pool_wgrad8(
                     red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid]);
 }
 
+// Fused pool wgrad + backward (C % 8 == 0): the two kernels read the
+// same dppre and acts; one pass computes the K*K+1 weight-grad partials
+// AND writes the expanded conv preact grad dapre = d * pw[ij] * a *
+// (1-a) IN PLACE over the activation (same-thread read-then-write per
+// element, so the overwrite is safe).  Halves the pool backward's HBM
+// reads and removes a launch.
+template <typename act_t, int KK>
+__global__ __launch_bounds__(256) void k_pool_wbwd8(
+    const act_t* __restrict__ dppre, const act_t* __restrict__ a,
+    const float* __restrict__ pw, act_t* __restrict__ dapre,
+    float* __restrict__ dpw, int B, int H, int W, int C, int G) {
+  const int OH = H / KK, OW = W / KK;
+  const int C8 = C / 8;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  float acc[KK * KK];
+#pragma unroll
+  for (int w = 0; w < KK * KK; ++w) acc[w] = 0.f;
+  float bacc = 0.f;
+  const long long N8 = (long long)B * OH * OW * C8;
+  const long long stride = (long long)G * 256;
+
+  for (long long it = (long long)blockIdx.x * 256 + tid; it < N8;
+       it += stride) {
+    const int c0 = (int)(it % C8) * 8;
+    long long t = it / C8;
+    const int q = (int)(t % OW);
+    t /= OW;
+    const int p = (int)(t % OH);
+    const int b = (int)(t / OH);
+    float d[8];
+    ld8v(dppre + (((long long)b * OH + p) * OW + q) * C + c0, d);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) bacc += d[u];
+#pragma unroll
+    for (int i = 0; i < KK; ++i)
+#pragma unroll
+      for (int j = 0; j < KK; ++j) {
+        const long long off =
+            (((long long)b * H + p * KK + i) * W + q * KK + j) * C + c0;
+        float av[8];
+        ld8v(a + off, av);
+        const float wv = pw[i * KK + j];
+        act_t o8[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          acc[i * KK + j] += d[u] * av[u];
+          o8[u] = (act_t)(d[u] * wv * av[u] * (1.0f - av[u]));
+        }
+        if (sizeof(act_t) == 2)
+          *reinterpret_cast<uint4*>(dapre + off) =
+              *reinterpret_cast<const uint4*>(o8);
+        else
+#pragma unroll
+          for (int u = 0; u < 8; ++u) dapre[off + u] = o8[u];
+      }
+  }
+  __shared__ float red[4][KK * KK + 1];
+  const int wv = tid >> 6;
+#pragma unroll
+  for (int w = 0; w < KK * KK; ++w) {
+    float v = acc[w];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][w] = v;
+  }
+  {
+    float v = bacc;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (lane == 0) red[wv][KK * KK] = v;
+  }
+  __syncthreads();
+  if (tid <= KK * KK)
+    unsafeAtomicAdd(&dpw[tid],
+                    red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid]);
+}
+
 // ---------------------------------------------------------------------------
 // FC head (FCIN -> 10) + residual loss, general fan-in.  One block per
 // sample; 16 lanes per class with shuffle reduction (the LeNet pattern).
@@ -1223,7 +1301,20 @@ __global__ __launch_bounds__(256) void k_fc_fwd(
     const int l = tid & 15;
     const float* wk = fw + (long long)k * FCIN;
     float p = 0.f;
-    for (int m = l; m < FCIN; m += 16) p += wk[m] * ldf(xb + m);
+    if ((FCIN % 128) == 0) {
+      // 8-wide vector loads per lane: FCIN=1024 is 8 iterations of 16B
+      // loads instead of 64 dependent scalar loads (the serial chain was
+      // the whole kernel's latency at B blocks ~ 1/CU occupancy)
+      for (int m0 = l * 8; m0 < FCIN; m0 += 128) {
+        float xv[8], wv8[8];
+        ld8v(xb + m0, xv);
+        ld8v(wk + m0, wv8);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) p += wv8[u] * xv[u];
+      }
+    } else {
+      for (int m = l; m < FCIN; m += 16) p += wk[m] * ldf(xb + m);
+    }
 #pragma unroll
     for (int off = 8; off > 0; off >>= 1) p += __shfl_down(p, off, 16);
     if (l == 0) {
@@ -1861,6 +1952,20 @@ int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
                                           0, (hipStream_t)stream,
                                           (const act_t*)dppre,
                                           (const act_t*)a, dpw, B, H, W, C, K,
+                                          G));
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_pool_wbwd(const void* dppre, const void* a, const float* pw,
+                        void* dapre, float* dpw, int B, int H, int W, int C,
+                        int K, int G, int actf, void* stream) {
+  if (C % 8 != 0 || K != 2) return -2;  // engine falls back to 2 kernels
+  dim3 grid(G), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_pool_wbwd8<act_t, 2>), grid,
+                                          block, 0, (hipStream_t)stream,
+                                          (const act_t*)dppre,
+                                          (const act_t*)a, pw,
+                                          (act_t*)dapre, dpw, B, H, W, C,
                                           G));
   return (int)hipGetLastError();
 }

@@ -301,7 +301,10 @@ class DeepTrainer:
         nstage = len(spec.stages)
         self._C.deep_fc_bwd(w.dz, w.pouts[-1], m.view("fc_w"),
                             w.dppre[-1], B, spec.fc_in, spec.n_classes, st_h)
-        fs = max(1, min(32, B // 64))
+        # batch-slice count sized to ~4 WGs/CU (the owner-per-(k,m) grid
+        # alone is only ~40 WGs for the 10x1024 head)
+        bps = (spec.n_classes * spec.fc_in + 255 + spec.n_classes) // 256 + 1
+        fs = max(1, min(B, 1024 // bps))
         self._C.deep_fc_wgrad(w.dz, w.pouts[-1], m.grad_view("fc_w"),
                               m.grad_view("fc_b"), B, spec.fc_in,
                               spec.n_classes, fs, st_h)
@@ -312,14 +315,22 @@ class DeepTrainer:
             # atomics land on one cache line (~3ns/op serialized), so the
             # sweet spot trades compute depth against the atomic tail
             G = max(64, min(512, (B * st.oh * st.ow * st.cout) // (256 * 32)))
-            self._C.deep_pool_wgrad(w.dppre[i], w.acts[i],
-                                    m.grad_view(f"pool{i}_w"), B, st.h, st.w,
-                                    st.cout, st.pool_k, G, st_h)
-            # pool bwd writes the conv preact grad IN PLACE over the conv
-            # activation (elementwise same-index, safe)
-            self._C.deep_pool_bwd(w.dppre[i], w.acts[i],
-                                  m.view(f"pool{i}_w"), w.acts[i], B, st.h,
-                                  st.w, st.cout, st.pool_k, st_h)
+            if st.cout % 8 == 0 and st.pool_k == 2:
+                # fused: one pass computes the pool weight grads AND
+                # writes dapre in place over the activation
+                self._C.deep_pool_wbwd(w.dppre[i], w.acts[i],
+                                       m.view(f"pool{i}_w"), w.acts[i],
+                                       m.grad_view(f"pool{i}_w"), B, st.h,
+                                       st.w, st.cout, st.pool_k, G, st_h)
+            else:
+                self._C.deep_pool_wgrad(w.dppre[i], w.acts[i],
+                                        m.grad_view(f"pool{i}_w"), B, st.h,
+                                        st.w, st.cout, st.pool_k, G, st_h)
+                # pool bwd writes the conv preact grad IN PLACE over the
+                # conv activation (elementwise same-index, safe)
+                self._C.deep_pool_bwd(w.dppre[i], w.acts[i],
+                                      m.view(f"pool{i}_w"), w.acts[i], B,
+                                      st.h, st.w, st.cout, st.pool_k, st_h)
             dapre = w.acts[i]
             ms = self.ws.wgrad_ms(st, M)
             # measured (tools/deep_sweep.py): atomic combine beats the
