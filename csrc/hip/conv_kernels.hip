@@ -1424,55 +1424,92 @@ __global__ void k_split_epi(const float* __restrict__ c32, int KS,
                             int XW) {
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long MN = M * N;
+  // 2-in-flight slab reads: two independent 16B loads per iteration so
+  // the KS-deep sum is not one serial latency chain
+  auto slab_sum = [&](long long off, float* v8) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v8[u] = 0.f;
+    int s = 0;
+    for (; s + 2 <= KS; s += 2) {
+      float a8[8], b8[8];
+      ld8v(c32 + (long long)s * MN + off, a8);
+      ld8v(c32 + (long long)(s + 1) * MN + off, b8);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) v8[u] += a8[u] + b8[u];
+    }
+    if (s < KS) {
+      float a8[8];
+      ld8v(c32 + (long long)s * MN + off, a8);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) v8[u] += a8[u];
+    }
+  };
   if (epilogue == 3) {
-    const int OW = XW / PK, OH = XH / PK;
+    // one thread per CONV element x 8 channels (PK*PK more threads than
+    // pooled positions — a 32-WG combine at small M measured 19 us);
+    // window-aligned thread order so each pool window lives in one
+    // block, acts exchanged through LDS.
+    const int OW = XW / PK;
     const int n8 = N / 8;
-    const long long total = (M / (PK * PK)) * n8;
-    if (idx >= total) return;
-    const int c0 = (int)(idx % n8) * 8;
-    long long pp = idx / n8;
+    const int wsz = PK * PK * n8;      // threads per pool window
+    (void)wsz;
+    __shared__ float lds_a[256 * 8];
+    // NO early return before the barrier (divergent-barrier UB in the
+    // ragged last block) — inactive threads just skip the work
+    const bool active = idx * 8 < MN;
+    // idx = (pp * PK*PK + ij) * n8 + c8
+    const int c8 = (int)(idx % n8);
+    const long long t1 = idx / n8;
+    const int ij = (int)(t1 % (PK * PK));
+    const long long pp = t1 / (PK * PK);
+    const int i = ij / PK, j = ij % PK;
     const int q = (int)(pp % OW);
     long long t = pp / OW;
-    const int p = (int)(t % OH);
-    const long long b = t / OH;
-    float pacc[8];
-    const float pb = pw[PK * PK];
+    const int p = (int)(t % (XH / PK));
+    const long long b = t / (XH / PK);
+    const int c0 = c8 * 8;
+    const long long m = ((b * XH + p * PK + i) * XW + q * PK + j);
+    if (active) {
+      float v8[8];
+      slab_sum(m * N + c0, v8);
+      act_t o8[8];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) pacc[u] = pb;
-    for (int i = 0; i < PK; ++i)
-      for (int j = 0; j < PK; ++j) {
-        const long long m = ((b * XH + p * PK + i) * XW + q * PK + j);
-        float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-        for (int s = 0; s < KS; ++s) {
-          float a8[8];
-          ld8v(c32 + (long long)s * MN + m * N + c0, a8);
-#pragma unroll
-          for (int u = 0; u < 8; ++u) v8[u] += a8[u];
-        }
-        const float wv = pw[i * PK + j];
-        act_t o8[8];
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const float a = sigmoidf_dev(v8[u] + bias[c0 + u]);
-          o8[u] = (act_t)a;
-          pacc[u] += wv * a;
-        }
-        if (sizeof(act_t) == 2)
-          *reinterpret_cast<uint4*>(C + m * N + c0) =
-              *reinterpret_cast<const uint4*>(o8);
-        else
-#pragma unroll
-          for (int u = 0; u < 8; ++u) C[m * N + c0 + u] = o8[u];
+      for (int u = 0; u < 8; ++u) {
+        const float a = sigmoidf_dev(v8[u] + bias[c0 + u]);
+        o8[u] = (act_t)a;
+        lds_a[threadIdx.x * 8 + u] = a;
       }
-    act_t po[8];
+      if (sizeof(act_t) == 2)
+        *reinterpret_cast<uint4*>(C + m * N + c0) =
+            *reinterpret_cast<const uint4*>(o8);
+      else
 #pragma unroll
-    for (int u = 0; u < 8; ++u) po[u] = (act_t)sigmoidf_dev(pacc[u]);
-    if (sizeof(act_t) == 2)
-      *reinterpret_cast<uint4*>(pout + pp * N + c0) =
-          *reinterpret_cast<const uint4*>(po);
-    else
+        for (int u = 0; u < 8; ++u) C[m * N + c0 + u] = o8[u];
+    }
+    __syncthreads();
+    if (active && ij == 0) {
+      // window base thread pools its PK*PK acts from LDS
+      float pacc[8];
+      const float pb = pw[PK * PK];
 #pragma unroll
-      for (int u = 0; u < 8; ++u) pout[pp * N + c0 + u] = po[u];
+      for (int u = 0; u < 8; ++u) pacc[u] = pb;
+      const int base = (int)(threadIdx.x);  // = (local pp) * wsz + c8
+      for (int w2 = 0; w2 < PK * PK; ++w2) {
+        const float wv = pw[w2];
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          pacc[u] += wv * lds_a[(base + w2 * n8) * 8 + u];
+      }
+      act_t po[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) po[u] = (act_t)sigmoidf_dev(pacc[u]);
+      if (sizeof(act_t) == 2)
+        *reinterpret_cast<uint4*>(pout + pp * N + c0) =
+            *reinterpret_cast<const uint4*>(po);
+      else
+#pragma unroll
+        for (int u = 0; u < 8; ++u) pout[pp * N + c0 + u] = po[u];
+    }
     return;
   }
   // modes 0/1/2: one thread per 8 consecutive elements of the [M][N]
@@ -1480,13 +1517,8 @@ __global__ void k_split_epi(const float* __restrict__ c32, int KS,
   if (idx * 8 >= MN) return;
   const long long e0 = idx * 8;
   const int n = (int)(e0 % N);
-  float v8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int s = 0; s < KS; ++s) {
-    float a8[8];
-    ld8v(c32 + (long long)s * MN + e0, a8);
-#pragma unroll
-    for (int u = 0; u < 8; ++u) v8[u] += a8[u];
-  }
+  float v8[8];
+  slab_sum(e0, v8);
   act_t o8[8];
   if (epilogue == 1) {
 #pragma unroll
@@ -1688,7 +1720,8 @@ int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
   if (epilogue == 2 && epi == nullptr) return -3;
   if (epilogue == 3 &&
       (pw == nullptr || pout == nullptr || N > BN || PK < 1 || XW < 1 ||
-       64 % XW != 0 || (64 / XW) % PK != 0 || XH % PK != 0 || M % 64 != 0))
+       64 % XW != 0 || (64 / XW) % PK != 0 || XH % PK != 0 ||
+       M % 64 != 0 || (N % 8) != 0 || 256 % (PK * PK * (N / 8)) != 0))
     return -4;  // fused-pool preconditions (engine falls back)
   const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
   const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
@@ -1722,9 +1755,7 @@ int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
                                     pw, (act_t*)pout, PK,
                                     split ? c32 : nullptr, tpc));
   if (split) {
-    const long long total = (epilogue == 3)
-                                ? (M / (PK * PK)) * (N / 8)
-                                : (M * N) / 8;
+    const long long total = (M * N) / 8;
     dim3 g2((unsigned)((total + 255) / 256));
     PCNN_DISPATCH(actf, hipLaunchKernelGGL(
                             (k_split_epi<act_t>), g2, block, 0,
