@@ -96,10 +96,11 @@ int pcnn_deep_pool_wgrad(const void* dppre, const void* a, float* dpw, int B,
 int pcnn_deep_pool_wbwd(const void* dppre, const void* a, const float* pw,
                         void* dapre, float* dpw, int B, int H, int W, int C,
                         int K, int G, int actf, void* stream);
-int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
-                     const int* labels, float* yg, float* dzg,
-                     float* loss_accum, int* correct_accum, int B, int FCIN,
-                     int NCLS, int mode, int actf, void* stream);
+int pcnn_deep_fc_fwd2(const void* flat, const float* fw, const float* fb,
+                      const int* labels, float* yg, float* dzg,
+                      float* loss_accum, int* correct_accum, int B, int FCIN,
+                      int NCLS, int mode, void* dflat, int actf,
+                      void* stream);
 int pcnn_deep_fc_bwd(const float* dzg, const void* flat, const float* fw,
                      void* dflat, int B, int FCIN, int NCLS, int actf,
                      void* stream);
@@ -413,8 +414,9 @@ void deep_pool_wbwd(at::Tensor dppre, at::Tensor a, at::Tensor pw,
 void deep_fc_fwd(at::Tensor flat, at::Tensor fw, at::Tensor fb,
                  at::Tensor labels, at::Tensor y, at::Tensor dz,
                  at::Tensor loss_accum, at::Tensor correct_accum, int64_t B,
-                 int64_t FCIN, int64_t NCLS, int64_t mode, int64_t stream) {
-  check_hip(pcnn_deep_fc_fwd(
+                 int64_t FCIN, int64_t NCLS, int64_t mode, int64_t stream,
+                 at::Tensor dflat) {
+  check_hip(pcnn_deep_fc_fwd2(
                 flat.data_ptr(), fw.data_ptr<float>(), fb.data_ptr<float>(),
                 labels.data_ptr<int>(),
                 y.numel() ? y.data_ptr<float>() : nullptr,
@@ -422,7 +424,8 @@ void deep_fc_fwd(at::Tensor flat, at::Tensor fw, at::Tensor fb,
                 loss_accum.numel() ? loss_accum.data_ptr<float>() : nullptr,
                 correct_accum.numel() ? correct_accum.data_ptr<int>()
                                       : nullptr,
-                (int)B, (int)FCIN, (int)NCLS, (int)mode, act_flag(flat),
+                (int)B, (int)FCIN, (int)NCLS, (int)mode,
+                dflat.numel() ? dflat.data_ptr() : nullptr, act_flag(flat),
                 (void*)stream),
             "deep_fc_fwd");
 }
@@ -521,7 +524,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_pool_bwd", &deep_pool_bwd);
   m.def("deep_pool_wgrad", &deep_pool_wgrad);
   m.def("deep_pool_wbwd", &deep_pool_wbwd);
-  m.def("deep_fc_fwd", &deep_fc_fwd);
+  m.def("deep_fc_fwd", &deep_fc_fwd, py::arg("flat"), py::arg("fw"),
+        py::arg("fb"), py::arg("labels"), py::arg("y"), py::arg("dz"),
+        py::arg("loss_accum"), py::arg("correct_accum"), py::arg("B"),
+        py::arg("FCIN"), py::arg("NCLS"), py::arg("mode"),
+        py::arg("stream"), py::arg("dflat") = at::empty({0}));
   m.def("deep_fc_bwd", &deep_fc_bwd);
   m.def("deep_fc_wgrad", &deep_fc_wgrad);
   m.def("deep_update", &deep_update);

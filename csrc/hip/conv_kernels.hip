@@ -1283,15 +1283,20 @@ __global__ __launch_bounds__(256) void k_pool_wbwd8(
 // sample; 16 lanes per class with shuffle reduction (the LeNet pattern).
 // mode: 0 train (emit dz + loss), 1 eval (argmax + correct), 2 infer.
 // ---------------------------------------------------------------------------
+// dflat != null (train mode): the fc backward-data is fused in — after
+// the block's dz is in LDS, the same block computes dflat[b][m] =
+// (sum_k fw[k][m] dz[k]) * flat*(1-flat) for its sample (kills the
+// separate k_fc_bwd launch; fw re-read comes from L2).
 template <typename act_t>
 __global__ __launch_bounds__(256) void k_fc_fwd(
     const act_t* __restrict__ flat, const float* __restrict__ fw,
     const float* __restrict__ fb, const int* __restrict__ labels,
     float* __restrict__ yg, float* __restrict__ dzg,
     float* __restrict__ loss_accum, int* __restrict__ correct_accum, int B,
-    int FCIN, int NCLS, int mode) {
+    int FCIN, int NCLS, int mode, act_t* __restrict__ dflat) {
   __shared__ float ys[32];
   __shared__ float sq[32];
+  __shared__ float dzs[32];
   const int b = blockIdx.x;
   if (b >= B) return;
   const int tid = threadIdx.x;
@@ -1324,6 +1329,7 @@ __global__ __launch_bounds__(256) void k_fc_fwd(
       if (mode == 0) {
         const float d = (k == labels[b] ? 1.0f : 0.0f) - v;
         sq[k] = d * d;
+        dzs[k] = d;
         dzg[(long long)b * NCLS + k] = d;
       }
     }
@@ -1339,6 +1345,15 @@ __global__ __launch_bounds__(256) void k_fc_fwd(
     float s = 0.f;
     for (int k = 0; k < NCLS; ++k) s += sq[k];
     unsafeAtomicAdd(loss_accum, sqrtf(s));
+  }
+  if (mode == 0 && dflat != nullptr) {
+    for (int m = tid; m < FCIN; m += 256) {
+      float da = 0.f;
+      for (int k = 0; k < NCLS; ++k)
+        da += fw[(long long)k * FCIN + m] * dzs[k];
+      const float v = ldf(xb + m);
+      stf(dflat + (long long)b * FCIN + m, da * v * (1.0f - v));
+    }
   }
 }
 
@@ -2001,17 +2016,28 @@ int pcnn_deep_pool_wbwd(const void* dppre, const void* a, const float* pw,
   return (int)hipGetLastError();
 }
 
-int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
-                     const int* labels, float* yg, float* dzg,
-                     float* loss_accum, int* correct_accum, int B, int FCIN,
-                     int NCLS, int mode, int actf, void* stream) {
+int pcnn_deep_fc_fwd2(const void* flat, const float* fw, const float* fb,
+                      const int* labels, float* yg, float* dzg,
+                      float* loss_accum, int* correct_accum, int B, int FCIN,
+                      int NCLS, int mode, void* dflat, int actf,
+                      void* stream) {
   dim3 grid(B), block(256);
   PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_fc_fwd<act_t>), grid, block, 0,
                                           (hipStream_t)stream,
                                           (const act_t*)flat, fw, fb, labels,
                                           yg, dzg, loss_accum, correct_accum,
-                                          B, FCIN, NCLS, mode));
+                                          B, FCIN, NCLS, mode,
+                                          (act_t*)dflat));
   return (int)hipGetLastError();
+}
+
+int pcnn_deep_fc_fwd(const void* flat, const float* fw, const float* fb,
+                     const int* labels, float* yg, float* dzg,
+                     float* loss_accum, int* correct_accum, int B, int FCIN,
+                     int NCLS, int mode, int actf, void* stream) {
+  return pcnn_deep_fc_fwd2(flat, fw, fb, labels, yg, dzg, loss_accum,
+                           correct_accum, B, FCIN, NCLS, mode, nullptr,
+                           actf, stream);
 }
 
 int pcnn_deep_fc_bwd(const float* dzg, const void* flat, const float* fw,

@@ -30,13 +30,16 @@ MODE_TRAIN, MODE_EVAL, MODE_INFER = 0, 1, 2
 class DeepWorkspace:
     @staticmethod
     def wgrad_ms(st, M: int) -> int:
-        """K-slice count for the weight-grad GEMM: ~512 total workgroups
-        (measured optimum — per-WG fixed costs rise beyond it and the
-        fp32 atomic combine over the dW tile lines stays cheap), never
-        slicing thinner than one 64-row BK chunk."""
+        """M-slice count for the weight-grad GEMM.  Round 1 measured ~512
+        total WGs optimal; after the db fusion + hoisted implicit decode
+        the per-WG fixed cost shrank and ~1024 WGs (4/CU — enough
+        resident waves to hide the per-iteration stage stall) wins.
+        PCNN_WGRAD_WGS overrides for sweeps."""
+        import os
+        target = int(os.environ.get("PCNN_WGRAD_WGS", "1024"))
         ktiles = (st.kcp + 63) // 64
         ntiles = (st.cout + 63) // 64
-        return max(1, min(256, 512 // (ktiles * ntiles), M // 64))
+        return max(1, min(256, target // (ktiles * ntiles), M // 64))
 
     def __init__(self, model: DeepCNN, max_batch: int, device, act_dtype,
                  implicit: bool = True):
@@ -290,17 +293,20 @@ class DeepTrainer:
                                       w.pouts[i], B, st.h, st.w, st.cout,
                                       st.pool_k, st_h)
             src = w.pouts[i]
+        # train mode fuses the fc backward-data (dflat -> dppre[-1]) into
+        # the forward block — k_fc_bwd is not launched separately
         self._C.deep_fc_fwd(w.pouts[-1], m.view("fc_w"), m.view("fc_b"),
                             labels, w.y, w.dz, w.loss_accum,
                             w.correct_accum, B, spec.fc_in, spec.n_classes,
-                            mode, st_h)
+                            mode, st_h,
+                            dflat=w.dppre[-1] if mode == MODE_TRAIN
+                            else torch.empty(0))
 
     def _hip_backward(self, x: torch.Tensor, B: int):
         m, w, spec = self.model, self.ws, self.model.spec
         st_h = native.current_stream_handle()
         nstage = len(spec.stages)
-        self._C.deep_fc_bwd(w.dz, w.pouts[-1], m.view("fc_w"),
-                            w.dppre[-1], B, spec.fc_in, spec.n_classes, st_h)
+        # fc backward-data already produced by the fused forward (dflat)
         # batch-slice count sized to ~4 WGs/CU (the owner-per-(k,m) grid
         # alone is only ~40 WGs for the 10x1024 head)
         bps = (spec.n_classes * spec.fc_in + 255 + spec.n_classes) // 256 + 1
