@@ -32,11 +32,12 @@ class DeepWorkspace:
     def wgrad_ms(st, M: int) -> int:
         """M-slice count for the weight-grad GEMM.  Round 1 measured ~512
         total WGs optimal; after the db fusion + hoisted implicit decode
-        the per-WG fixed cost shrank and ~1024 WGs (4/CU — enough
-        resident waves to hide the per-iteration stage stall) wins.
-        PCNN_WGRAD_WGS overrides for sweeps."""
+        the per-WG fixed cost shrank — but ~512 total WGs REMAINS the
+        measured optimum (1024 lost 8% at bs=64, 2048 lost 19%: the
+        atomic-combine tail and per-WG prologue beat the extra
+        residency).  PCNN_WGRAD_WGS overrides for sweeps."""
         import os
-        target = int(os.environ.get("PCNN_WGRAD_WGS", "1024"))
+        target = int(os.environ.get("PCNN_WGRAD_WGS", "512"))
         ktiles = (st.kcp + 63) // 64
         ntiles = (st.cout + 63) // 64
         return max(1, min(256, target // (ktiles * ntiles), M // 64))
