@@ -174,16 +174,19 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
         }
       }
     } else {
+      // clamped unconditional loads + select-to-zero (same idiom as the
+      // vec path): 8 independent scalar loads issue back-to-back instead
+      // of a branchy serial chain (the old guarded form was this
+      // kernel's whole cost on the Cin=3 stage)
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        float val = 0.f;
-        if (kc + u < Kc) {
-          const int ih = oh + i - P;
-          const int iw = ow + j - P;
-          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-            val = ldf(xb + ((long long)ih * W + iw) * Cin + ci);
-        }
-        out[u] = (act_t)val;
+        const int ih = oh + i - P;
+        const int iw = ow + j - P;
+        const int ihc = min(max(ih, 0), H - 1);
+        const int iwc = min(max(iw, 0), W - 1);
+        const float val = ldf(xb + ((long long)ihc * W + iwc) * Cin + ci);
+        out[u] = (act_t)((kc + u < Kc && ih == ihc && iw == iwc) ? val
+                                                                 : 0.f);
         if (++ci >= Cin) {
           ci = 0;
           if (++j >= K) {

@@ -321,7 +321,10 @@ class DeepTrainer:
             # ~halved from the 2-items/thread target: the 17 per-WG
             # atomics land on one cache line (~3ns/op serialized), so the
             # sweet spot trades compute depth against the atomic tail
-            G = max(64, min(512, (B * st.oh * st.ow * st.cout) // (256 * 32)))
+            # ~2 pooled items (of 8 channels) per thread: the fused
+            # wgrad+bwd kernel writes K*K 16B stores per item, so deeper
+            # per-thread loops no longer amortize anything
+            G = max(64, min(512, (B * st.oh * st.ow * st.cout) // (256 * 16)))
             if st.cout % 8 == 0 and st.pool_k == 2:
                 # fused: one pass computes the pool weight grads AND
                 # writes dapre in place over the activation
