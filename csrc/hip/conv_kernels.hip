@@ -708,14 +708,26 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     }
   };
 
+  // XOR-swizzled transpose staging: the column write (one element per
+  // LDS row, all threads of a c-group on the same bank — any row pitch
+  // that keeps 16B-aligned fragment reads collapses mod 32) measured
+  // 22% LDSBankConflict.  Swizzling the 8-element m-group by the row
+  // index spreads the writes; fragment reads stay single 16B ds_reads
+  // (group-granular swizzle).
+  auto sw = [](int k, int m) {
+    // fold BOTH k&7 and k>>3 into the group swizzle: the simultaneous
+    // writers share k mod 8 (their k differ by 16), so low bits alone
+    // would leave them on one bank
+    return (((m >> 3) ^ (k & 7) ^ ((k >> 3) & 7)) & 7) * 8 + (m & 7);
+  };
   auto write_lds = [&](GemmLds& L) {
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       const int c = cq + h * 8;
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        L.As[c + u][row_s] = (__bf16)rc[h][u];
-        L.Bs[c + u][row_s] = (__bf16)rd[h][u];
+        L.As[c + u][sw(c + u, row_s)] = (__bf16)rc[h][u];
+        L.Bs[c + u][sw(c + u, row_s)] = (__bf16)rd[h][u];
         if (do_db) dbacc[h * 8 + u] += rd[h][u];
       }
     }
@@ -731,13 +743,15 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
     auto& L = Lb[p];
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const bf16x8 a0 = frag_from_lds(L.As[wv * 16 + (lane & 15)],
-                                      kk * 32 + (lane >> 4) * 8);
+      const int ar = wv * 16 + (lane & 15);
+      const int m8 = kk * 32 + (lane >> 4) * 8;  // 8-aligned: sw() keeps
+                                                 // the span one 16B read
+      const bf16x8 a0 = frag_from_lds(L.As[ar], sw(ar, m8));
 #pragma unroll
       for (int f = 0; f < BN / 16; ++f) {
         if (f < nf) {
-          const bf16x8 bf = frag_from_lds(L.Bs[f * 16 + (lane & 15)],
-                                          kk * 32 + (lane >> 4) * 8);
+          const int br = f * 16 + (lane & 15);
+          const bf16x8 bf = frag_from_lds(L.Bs[br], sw(br, m8));
           acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[f], 0,
                                                            0, 0);
         }
