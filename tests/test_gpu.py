@@ -469,3 +469,36 @@ def test_scale_check_preflight(device):
     assert "READY" in out.stdout
     assert "backend=nccl" in out.stdout
     assert "FAIL" not in out.stdout
+
+
+def test_native_cli_mnist_idx(device, tmp_path):
+    """Native CLI end-to-end off real IDX files on disk (the reference's
+    loaddata path, Sequential/Main.cpp:36-42, in the no-Python driver)."""
+    import os
+    import struct
+    import numpy as np
+    bin_path = os.path.join(os.path.dirname(__file__), "..", "tools",
+                            "pcnn_train")
+    if not os.path.exists(bin_path):
+        pytest.skip("native CLI not built")
+    rng = np.random.default_rng(7)
+    d = tmp_path / "data"
+    d.mkdir()
+    for iname, n, lname in [("train-images.idx3-ubyte", 256,
+                             "train-labels.idx1-ubyte"),
+                            ("t10k-images.idx3-ubyte", 128,
+                             "t10k-labels.idx1-ubyte")]:
+        imgs = rng.integers(0, 256, size=(n, 28, 28), dtype=np.uint8)
+        lbls = rng.integers(0, 10, size=n, dtype=np.uint8)
+        with open(d / iname, "wb") as f:
+            f.write(struct.pack(">iiii", 2051, n, 28, 28))
+            f.write(imgs.tobytes())
+        with open(d / lname, "wb") as f:
+            f.write(struct.pack(">ii", 2049, n))
+            f.write(lbls.tobytes())
+    out = subprocess.run(
+        [bin_path, "--data", "mnist", "--data-dir", str(d),
+         "--epochs", "1", "--batch-size", "32"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "Learning" in out.stdout and "Error Rate: " in out.stdout
