@@ -44,9 +44,13 @@ def main() -> int:
                    choices=["trainable", "max"])
     p.add_argument("--loss", default="residual",
                    choices=["residual", "softmax_ce"])
-    p.add_argument("--overlap-comm", action="store_true",
+    p.add_argument("--overlap-comm", default=None,
+                   action=argparse.BooleanOptionalAction,
                    help="two-bucket DP: overlap the fc/pool grad all-reduce "
-                        "with the conv wgrad")
+                        "with the conv wgrad (default: ON when world>1 — "
+                        "the 9.4 KB bucket is pure latency and the step is "
+                        "~19 us, so overlap is the right multi-GPU default; "
+                        "--no-overlap-comm forces it off)")
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
@@ -58,10 +62,12 @@ def main() -> int:
     cfg = TrainConfig(batch_size=args.batch_size, act_dtype=args.act_dtype,
                       device=args.device, log_interval=0, data="synthetic",
                       wgrad_chunk=args.wgrad_chunk, model=args.model,
-                      overlap_comm=args.overlap_comm, pool=args.pool,
+                      overlap_comm=bool(args.overlap_comm), pool=args.pool,
                       loss=args.loss)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
+    if args.overlap_comm is None and ctx.world_size > 1:
+        cfg.overlap_comm = True  # multi-GPU default (see --help)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
     if ctx.world_size == 1 and args.gpus > 1:
         raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
@@ -160,6 +166,7 @@ def main() -> int:
             "loss": args.loss,
             "backend": trainer.backend,
             "hipgraph": bool(getattr(trainer, "_graph", None)),
+            "overlap_comm": cfg.overlap_comm,
             "timed_blocks": len(blocks),
             "block_min_s": blocks[0],
             "block_max_s": blocks[-1],

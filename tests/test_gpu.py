@@ -235,7 +235,7 @@ def test_dp2_gloo_single_gpu(overlap, device):
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
          "--master-port", "29617", "bench.py", "--gpus", "2", "--steps",
          "20", "--warmup", "5", "--batch-size", "16"]
-        + (["--overlap-comm"] if overlap else []),
+        + (["--overlap-comm"] if overlap else ["--no-overlap-comm"]),
         capture_output=True, text=True, timeout=600,
         env={**__import__("os").environ, "PCNN_DIST_BACKEND": "gloo"})
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
