@@ -436,13 +436,20 @@ __global__ __launch_bounds__(256) void k_gemm(
     }
   };
 
+  // pack 8 fp32 -> one 16B bf16 LDS store (the row-major A/B images are
+  // contiguous along k and 16B aligned: kk % 8 == 0, row pitch 72*2 B)
+  auto st8 = [](__bf16* dst, const float* v) {
+    union {
+      __bf16 h[8];
+      uint4 u;
+    } t;
+#pragma unroll
+    for (int u2 = 0; u2 < 8; ++u2) t.h[u2] = (__bf16)v[u2];
+    *reinterpret_cast<uint4*>(dst) = t.u;
+  };
   auto write_lds = [&](GemmLdsT<TBM, BKC>& L) {
 #pragma unroll
-    for (int h = 0; h < NCH; ++h) {
-      const int kk = kq + h * 8;
-#pragma unroll
-      for (int u = 0; u < 8; ++u) L.As[row_a][kk + u] = (__bf16)ra[h][u];
-    }
+    for (int h = 0; h < NCH; ++h) st8(&L.As[row_a][kq + h * 8], ra[h]);
     if (Bpre == nullptr && b_kxn) {
 #pragma unroll
       for (int kb = 0; kb < BKC / 64; ++kb) {
@@ -457,11 +464,7 @@ __global__ __launch_bounds__(256) void k_gemm(
     } else {
       const int n = tid >> 2;
 #pragma unroll
-      for (int h = 0; h < NCHB; ++h) {
-        const int kk = bkq + h * 8;
-#pragma unroll
-        for (int u = 0; u < 8; ++u) L.Bs[n][kk + u] = (__bf16)rb[h][u];
-      }
+      for (int h = 0; h < NCHB; ++h) st8(&L.Bs[n][bkq + h * 8], rb[h]);
     }
   };
 
