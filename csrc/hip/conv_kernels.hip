@@ -347,13 +347,11 @@ __global__ __launch_bounds__(256) void k_gemm(
   const int row_a = tid / TPR;
   const int kq = (tid % TPR) * SPAN;
   const int bkq = (tid & 3) * SPANB;  // B staging: 4 threads per row
-  // Depth-2 software pipeline: TWO staging register sets, so a tile's
-  // global loads are issued two iterations before their s_waitcnt at
-  // write_lds (depth-1 left ~the whole load latency exposed per
-  // iteration at these small K-loops).  LDS stays double-buffered (a
-  // third buffer would halve WG residency).
-  float ra[2][NCH][8];
-  float rb[2][NCHB][8];
+  // T14-style software pipeline: next tile's global loads are issued into
+  // registers BEFORE the MFMA block; the LDS write happens after the read
+  // barrier.
+  float ra[NCH][8];
+  float rb[NCHB][8];
   const long long m_a = m0 + row_a;
   // implicit-A: decode this thread's im2col row position once (from the
   // CLAMPED row — loads are unconditional, so the address must stay
@@ -368,14 +366,14 @@ __global__ __launch_bounds__(256) void k_gemm(
     iKc = XK * XK * XC;
   }
 
-  auto load_regs = [&](int kt, int sset) {
+  auto load_regs = [&](int kt) {
 #pragma unroll
     for (int h = 0; h < NCH; ++h) {
       const int kk = kq + h * 8;
       const bool ok = m_a < M && (kt + kk) < K;
       if (imx != nullptr) {
         im2col8f(imx, XH, XW, XC, XK, XP, ib, ioh, iow, kt + kk, iKc,
-                 fd_cin, fd_k, ok, ra[sset][h]);
+                 fd_cin, fd_k, ok, ra[h]);
       } else {
         // clamped unconditional load + select-to-zero: a load behind a
         // thread-varying guard de-pipelines the whole K-loop (hipcc
@@ -383,10 +381,10 @@ __global__ __launch_bounds__(256) void k_gemm(
         // experiment).  K % 8 == 0 and K <= ldA keep the clamp in-bounds.
         const long long mm = m_a < M ? m_a : M - 1;
         const int kcl = (kt + kk) < K ? (kt + kk) : K - 8;
-        ld8v(A + mm * ldA + kcl, ra[sset][h]);
+        ld8v(A + mm * ldA + kcl, ra[h]);
         if (!ok) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) ra[sset][h][u] = 0.f;
+          for (int u = 0; u < 8; ++u) ra[h][u] = 0.f;
         }
       }
     }
@@ -400,10 +398,10 @@ __global__ __launch_bounds__(256) void k_gemm(
         const int kk = bkq + h * 8;
         const int kcl = (kt + kk) < K ? (kt + kk) : K - 8;
         ld8v(reinterpret_cast<const bf16*>(Bpre + (long long)nn * K + kcl),
-             rb[sset][h]);
+             rb[h]);
         if (!((n0 + n) < N && (kt + kk) < K)) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) rb[sset][h][u] = 0.f;
+          for (int u = 0; u < 8; ++u) rb[h][u] = 0.f;
         }
       }
     } else if (b_kxn) {
@@ -419,7 +417,7 @@ __global__ __launch_bounds__(256) void k_gemm(
 #pragma unroll
           for (int u = 0; u < 8; ++u) {
             const int n = n0 + nq + h * 8 + u;
-            rb[sset][kb * 2 + h][u] =
+            rb[kb * 2 + h][u] =
                 (kok && n < N) ? Bsrc[(long long)(kt + k) * N + n] : 0.f;
           }
       }
@@ -433,7 +431,7 @@ __global__ __launch_bounds__(256) void k_gemm(
         const float* src = Bsrc + (long long)(n0 + n) * K + kt + kk;
         const bool kok = ok && (kt + kk) < K;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) rb[sset][h][u] = kok ? src[u] : 0.f;
+        for (int u = 0; u < 8; ++u) rb[h][u] = kok ? src[u] : 0.f;
       }
     }
   };
@@ -449,10 +447,9 @@ __global__ __launch_bounds__(256) void k_gemm(
     for (int u2 = 0; u2 < 8; ++u2) t.h[u2] = (__bf16)v[u2];
     *reinterpret_cast<uint4*>(dst) = t.u;
   };
-  auto write_lds = [&](GemmLdsT<TBM, BKC>& L, int sset) {
+  auto write_lds = [&](GemmLdsT<TBM, BKC>& L) {
 #pragma unroll
-    for (int h = 0; h < NCH; ++h)
-      st8(&L.As[row_a][kq + h * 8], ra[sset][h]);
+    for (int h = 0; h < NCH; ++h) st8(&L.As[row_a][kq + h * 8], ra[h]);
     if (Bpre == nullptr && b_kxn) {
 #pragma unroll
       for (int kb = 0; kb < BKC / 64; ++kb) {
@@ -462,13 +459,12 @@ __global__ __launch_bounds__(256) void k_gemm(
         for (int h = 0; h < 2; ++h)
 #pragma unroll
           for (int u = 0; u < 8; ++u)
-            L.Bs[nq + h * 8 + u][k] = (__bf16)rb[sset][kb * 2 + h][u];
+            L.Bs[nq + h * 8 + u][k] = (__bf16)rb[kb * 2 + h][u];
       }
     } else {
       const int n = tid >> 2;
 #pragma unroll
-      for (int h = 0; h < NCHB; ++h)
-        st8(&L.Bs[n][bkq + h * 8], rb[sset][h]);
+      for (int h = 0; h < NCHB; ++h) st8(&L.Bs[n][bkq + h * 8], rb[h]);
     }
   };
 
@@ -477,12 +473,11 @@ __global__ __launch_bounds__(256) void k_gemm(
   // is written to Lb[1-p] and tile t+2's loads are issued — the
   // s_waitcnt for a tile's global loads lands ~one iteration after
   // issue instead of right after the MFMA block.
-  load_regs(k_lo, 0);
-  write_lds(Lb[0], 0);
-  if (k_lo + BKC < k_hi) load_regs(k_lo + BKC, 0);
-  if (k_lo + 2 * BKC < k_hi) load_regs(k_lo + 2 * BKC, 1);
+  load_regs(k_lo);
+  write_lds(Lb[0]);
+  if (k_lo + BKC < k_hi) load_regs(k_lo + BKC);
   __syncthreads();              // Lb[0] visible
-  int p = 0, sset = 0;
+  int p = 0;
   for (int kt = k_lo; kt < k_hi; kt += BKC, p ^= 1) {
     auto& L = Lb[p];
     // wave wv owns C rows [wv*16*RF, +16*RF); BKC/32 32-deep MFMA sub-steps
@@ -505,9 +500,8 @@ __global__ __launch_bounds__(256) void k_gemm(
       }
     }
     if (kt + BKC < k_hi) {
-      write_lds(Lb[p ^ 1], sset);  // waits on loads issued 2 iters ago
-      if (kt + 3 * BKC < k_hi) load_regs(kt + 3 * BKC, sset);
-      sset ^= 1;
+      write_lds(Lb[p ^ 1]);  // waits on tile t+1's loads here
+      if (kt + 2 * BKC < k_hi) load_regs(kt + 2 * BKC);
     }
     __syncthreads();  // reads of Lb[p] done AND Lb[1-p] complete
   }
