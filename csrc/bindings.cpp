@@ -72,6 +72,12 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
                        const long long* bf_off, const long long* bfT_off,
                        const long long* rot_off, const long long* p8_off,
                        void* stream);
+int pcnn_deep_update_cast(float* params, float* grads, long long n,
+                          float step, void* wbuf, int n_stages, const int* R,
+                          const int* C, const int* K, const int* Cin,
+                          const long long* w_off, const long long* bf_off,
+                          const long long* bfT_off, const long long* rot_off,
+                          const long long* p8_off, void* stream);
 int pcnn_deep_pad_channels(const void* x, void* x8, long long npix, int Cin,
                            int actf, void* stream);
 int pcnn_deep_remap_dw8(float* dW8, float* dW, int KK, int Cin, int Cout,
@@ -316,6 +322,37 @@ void deep_cast_all(at::Tensor params, at::Tensor wbuf,
             "deep_cast_all");
 }
 
+void deep_update_cast(at::Tensor params, at::Tensor grads, double step,
+                      at::Tensor wbuf, std::vector<int64_t> R,
+                      std::vector<int64_t> C, std::vector<int64_t> K,
+                      std::vector<int64_t> Cin, std::vector<int64_t> w_off,
+                      std::vector<int64_t> bf_off,
+                      std::vector<int64_t> bfT_off,
+                      std::vector<int64_t> rot_off,
+                      std::vector<int64_t> p8_off, int64_t stream) {
+  const size_t n = R.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "deep_update_cast: 1..8 stages");
+  int Ri[8], Ci[8], Ki[8], Cini[8];
+  long long wo[8], bo[8], bto[8], ro[8], po[8];
+  for (size_t s = 0; s < n; ++s) {
+    Ri[s] = (int)R[s];
+    Ci[s] = (int)C[s];
+    Ki[s] = (int)K[s];
+    Cini[s] = (int)Cin[s];
+    wo[s] = w_off[s];
+    bo[s] = bf_off[s];
+    bto[s] = bfT_off[s];
+    ro[s] = rot_off[s];
+    po[s] = p8_off.empty() ? -1 : p8_off[s];
+  }
+  check_hip(pcnn_deep_update_cast(params.data_ptr<float>(),
+                                  grads.data_ptr<float>(), params.numel(),
+                                  (float)step, wbuf.data_ptr(), (int)n, Ri,
+                                  Ci, Ki, Cini, wo, bo, bto, ro, po,
+                                  (void*)stream),
+            "deep_update_cast");
+}
+
 void deep_pad_channels(at::Tensor x, at::Tensor x8, int64_t npix,
                        int64_t Cin, int64_t stream) {
   check_hip(pcnn_deep_pad_channels(x.data_ptr(), x8.data_ptr(), npix,
@@ -510,6 +547,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_cast_wt", &deep_cast_wt);
   m.def("deep_cast_all", &deep_cast_all);
   m.def("deep_pad_channels", &deep_pad_channels);
+  m.def("deep_update_cast", &deep_update_cast);
   m.def("deep_remap_dw8", &deep_remap_dw8);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm, py::arg("cols"),
         py::arg("dpre"), py::arg("dW"), py::arg("M"), py::arg("KcP"),

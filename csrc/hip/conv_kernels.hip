@@ -588,6 +588,123 @@ __global__ __launch_bounds__(256) void k_gemm(
   }
 }
 
+// Small-K GEMM (K <= 128, N <= 64, materialized A, Bpre given): the
+// K-loop pipeline never warms at 2 iterations, so the general kernel
+// pays its whole prologue per 64-row tile.  Here the B image is staged
+// to LDS ONCE per workgroup and the WG grid-strides over M-tiles,
+// loading each tile's A fragments STRAIGHT into MFMA registers (16B
+// per lane, no LDS for A) with the next tile's loads issued before the
+// current tile's MFMA block — the M-axis, not the K-axis, carries the
+// pipeline.  Used by the Cin=3 stage-0 forward (K=96) and the
+// LeNet-shape probe.
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_gemm_smallk(
+    const act_t* __restrict__ A, const __bf16* __restrict__ Bpre,
+    const float* __restrict__ bias, act_t* __restrict__ C, long long M,
+    int K, int N, int ldA, int ldC, int epilogue,
+    const float* __restrict__ pw, act_t* __restrict__ pout, int PK, int XH,
+    int XW) {
+  __shared__ __bf16 Bs[64][136 + 8];
+  __shared__ float sc[64 * 64];  // mode-3 act tile
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  // stage B once: [N][K] bf16 rows, 16B chunks
+  for (int idx = tid * 8; idx < N * K; idx += 256 * 8) {
+    const int n = idx / K;
+    const int k = idx - n * K;
+    *reinterpret_cast<uint4*>(&Bs[n][k]) =
+        *reinterpret_cast<const uint4*>(Bpre + (long long)n * K + k);
+  }
+  __syncthreads();
+  const int nf = N / 16;
+  const int kf = (K + 31) / 32;      // MFMA K-steps (K % 8 == 0; tail
+                                     // spans read zero-padded A/B rows)
+  const long long mtiles = M / 64;   // M % 64 == 0 (launcher-enforced)
+  const int row = wv * 16 + (lane & 15);
+  const int kq = (lane >> 4) * 8;
+
+  bf16x8 a_cur[4], a_nxt[4];         // kf <= 4
+  auto load_a = [&](long long mt, bf16x8* dst) {
+    const act_t* src = A + (mt * 64 + row) * (long long)ldA + kq;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      if (kk < kf) {
+        // clamped-by-construction: kq + kk*32 + 8 <= ldA (ldA >= K,
+        // K-span padding rows are zero in the cols buffer)
+        if (kq + kk * 32 + 8 <= ldA)
+          dst[kk] = *reinterpret_cast<const bf16x8*>(src + kk * 32);
+        else
+          dst[kk] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+  };
+
+  long long t = blockIdx.x;
+  if (t < mtiles) load_a(t, a_cur);
+  for (; t < mtiles; t += gridDim.x) {
+    const long long tn = t + gridDim.x;
+    if (tn < mtiles) load_a(tn, a_nxt);
+    f32x4 acc[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      if (kk < kf) {
+#pragma unroll
+        for (int f = 0; f < 4; ++f)
+          if (f < nf) {
+            const bf16x8 bf =
+                frag_from_lds(Bs[f * 16 + (lane & 15)], kk * 32 + kq);
+            acc[f] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur[kk], bf,
+                                                        acc[f], 0, 0, 0);
+          }
+      }
+    // epilogue (same layouts as k_gemm)
+    const int ccol = lane & 15;
+    const int crow = wv * 16 + (lane >> 4) * 4;
+    const long long m0 = t * 64;
+#pragma unroll
+    for (int f = 0; f < 4; ++f)
+      if (f < nf) {
+        const int n = f * 16 + ccol;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[f][r];
+          if (epilogue >= 1) v = sigmoidf_dev(v + bias[n]);
+          stf(C + (m0 + crow + r) * ldC + n, v);
+          if (epilogue == 3) sc[(crow + r) * 64 + n] = v;
+        }
+      }
+    if (epilogue == 3) {
+      __syncthreads();
+      const int OW = XW / PK;
+      const long long row0 = m0 / XW;
+      const int npos = 64 / (PK * PK);
+      const float pb = pw[PK * PK];
+      for (int o = tid; o < npos * N; o += 256) {
+        const int lp = o / N;
+        const int n = o - lp * N;
+        const int pr = lp / OW;
+        const int pc = lp - pr * OW;
+        const long long crow0 = row0 + (long long)pr * PK;
+        const long long b = crow0 / XH;
+        const int h = (int)(crow0 - b * XH);
+        float a = pb;
+        for (int i = 0; i < PK; ++i)
+          for (int j = 0; j < PK; ++j)
+            a += pw[i * PK + j] *
+                 sc[((pr * PK + i) * XW + pc * PK + j) * 64 + n];
+        const long long pm = (b * (XH / PK) + h / PK) * OW + pc;
+        stf(pout + pm * ldC + n, sigmoidf_dev(a));
+      }
+      __syncthreads();  // sc reuse on the next m-tile
+    }
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) a_cur[kk] = a_nxt[kk];
+  }
+}
+
 // Weight-grad GEMM: dW[KcP][N] += cols^T[Kc x M-slice] @ dpre[M-slice x N].
 // Both operands are transpose-staged into the [row][k=m] LDS image; the
 // M dimension is the MFMA K axis.  Grid: (kc-tiles) x (n-tiles) x MS
@@ -1664,6 +1781,43 @@ __global__ void k_cast_wt_all(const float* __restrict__ params,
   }
 }
 
+// SGD update FUSED with the weight pre-cast: one pass updates every
+// parameter (p += step*g; g = 0) and, for conv-weight elements, emits
+// the next step's bf16 images (wbf/wbfT/wrot/wp8) from the freshly
+// computed value — the separate k_cast_wt_all launch at the next
+// forward is skipped (the engine tracks freshness).
+__global__ void k_update_cast_all(float* __restrict__ params,
+                                  float* __restrict__ grads, long long n,
+                                  float step, __bf16* __restrict__ wbuf,
+                                  CastDescs d) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const float pnew = params[i] + step * grads[i];
+  params[i] = pnew;
+  grads[i] = 0.f;
+  for (int s = 0; s < d.n; ++s) {
+    const long long q = i - d.w_off[s];
+    if (q < 0 || q >= d.cum[s + 1] - d.cum[s]) continue;
+    const int C = d.C[s];
+    const int kc = (int)(q / C);
+    const int c = (int)(q - (long long)kc * C);
+    const __bf16 v = (__bf16)pnew;
+    wbuf[d.bf_off[s] + q] = v;
+    wbuf[d.bfT_off[s] + (long long)c * d.R[s] + kc] = v;
+    const int Cin = d.Cin[s], K = d.K[s];
+    if (kc < K * K * Cin) {
+      const int ci = kc % Cin;
+      const int p2 = kc / Cin;
+      const int ki = p2 / K, kj = p2 - ki * K;
+      const int col = ((K - 1 - ki) * K + (K - 1 - kj)) * C + c;
+      wbuf[d.rot_off[s] + (long long)ci * (K * K * C) + col] = v;
+      if (d.p8_off[s] >= 0)
+        wbuf[d.p8_off[s] + (long long)c * (K * K * 8) + p2 * 8 + ci] = v;
+    }
+    break;
+  }
+}
+
 // Generic SGD apply + zero:  p += step*g; g = 0  over n params.
 __global__ void k_update_n(float* __restrict__ params,
                            float* __restrict__ grads, long long n,
@@ -1762,6 +1916,21 @@ int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
   const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
   const long long mtiles = (M + BM - 1) / BM;
   const long long mn = mtiles * ntiles;
+  // small-K fast path: B resident in LDS, A direct to MFMA registers,
+  // pipeline along M (the K-loop never warms at <= 2 iterations)
+  if (imx == nullptr && Bpre != nullptr && K <= 128 && (K % 32) == 0 &&
+      N <= 64 && (N % 16) == 0 && (M % 64) == 0 && ldA >= K &&
+      epilogue != 2) {
+    const long long g = mtiles < 1536 ? mtiles : 1536;
+    dim3 grid((unsigned)g), block(256);
+    PCNN_DISPATCH(actf, hipLaunchKernelGGL(
+                            (k_gemm_smallk<act_t>), grid, block, 0,
+                            (hipStream_t)stream, (const act_t*)A,
+                            (const __bf16*)Bpre, bias, (act_t*)C, M, K, N,
+                            ldA, ldC, epilogue, pw, (act_t*)pout, PK, XH,
+                            XW));
+    return (int)hipGetLastError();
+  }
   // split-K policy: a grid under ~3 WGs/CU leaves most of the chip idle
   // AND exposes the per-iteration stage/barrier latency (47 us measured
   // at 64 WGs for 20 us of 1024-WG work); split K until ~1024 WGs.
@@ -1864,6 +2033,35 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
   dim3 grid((unsigned)((d.cum[n_stages] + 255) / 256)), block(256);
   hipLaunchKernelGGL(k_cast_wt_all, grid, block, 0, (hipStream_t)stream,
                      params, (__bf16*)wbuf, d);
+  return (int)hipGetLastError();
+}
+
+int pcnn_deep_update_cast(float* params, float* grads, long long n,
+                          float step, void* wbuf, int n_stages, const int* R,
+                          const int* C, const int* K, const int* Cin,
+                          const long long* w_off, const long long* bf_off,
+                          const long long* bfT_off, const long long* rot_off,
+                          const long long* p8_off, void* stream) {
+  if (n_stages < 1 || n_stages > 8) return -2;
+  CastDescs d;
+  d.n = n_stages;
+  d.cum[0] = 0;
+  for (int s2 = 0; s2 < n_stages; ++s2) {
+    d.R[s2] = R[s2];
+    d.C[s2] = C[s2];
+    d.K[s2] = K[s2];
+    d.Cin[s2] = Cin[s2];
+    d.w_off[s2] = w_off[s2];
+    d.bf_off[s2] = bf_off[s2];
+    d.bfT_off[s2] = bfT_off[s2];
+    d.rot_off[s2] = rot_off[s2];
+    d.p8_off[s2] = p8_off ? p8_off[s2] : -1;
+    d.cum[s2 + 1] = d.cum[s2] + (long long)R[s2] * C[s2];
+  }
+  dim3 grid((unsigned)((n + 255) / 256)), block(256);
+  hipLaunchKernelGGL(k_update_cast_all, grid, block, 0,
+                     (hipStream_t)stream, params, grads, n, step,
+                     (__bf16*)wbuf, d);
   return (int)hipGetLastError();
 }
 

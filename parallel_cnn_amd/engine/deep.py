@@ -207,6 +207,10 @@ class DeepTrainer:
         self._samples_seen = 0
         self.global_step = 0
         self.epoch = 0              # completed-epoch cursor (exact resume)
+        # wbuf freshness: True when the bf16 weight images match params
+        # (the fused update+cast keeps them fresh across training steps;
+        # any external params mutation must invalidate)
+        self._wbuf_fresh = False
         self.timers = None  # set by enable_profiling
 
     def enable_profiling(self) -> None:
@@ -230,14 +234,36 @@ class DeepTrainer:
         return x.to(torch.float32), labels.to(torch.int64)
 
     # ------------------------------------------------------------- hip paths
-    def _hip_cast_weights(self):
+    def invalidate_weight_cache(self) -> None:
+        """Call after mutating model.params outside the engine (e.g.
+        checkpoint load): the bf16 weight images are re-cast on the next
+        forward."""
+        self._wbuf_fresh = False
+
+    def _hip_cast_weights(self, force: bool = False):
         # ONE launch casts every stage's weights into all three bf16
-        # images (wbf / wbfT / wrot) — was one k_cast_wt per stage
+        # images (wbf / wbfT / wrot) — skipped entirely when the fused
+        # update+cast already refreshed them at the end of the previous
+        # step (the common training case)
+        if self._wbuf_fresh and not force:
+            return
         d = self.ws.cast_desc
         self._C.deep_cast_all(self.model.params, self.ws.wbuf, d["R"],
                               d["C"], d["K"], d["Cin"], d["w_off"],
                               d["bf_off"], d["bfT_off"], d["rot_off"],
                               d["p8_off"], native.current_stream_handle())
+        self._wbuf_fresh = True
+
+    def _hip_update(self, scale: float) -> None:
+        """SGD update fused with the next step's weight cast."""
+        d = self.ws.cast_desc
+        self._C.deep_update_cast(self.model.params, self.model.grads,
+                                 self.cfg.dt * scale, self.ws.wbuf, d["R"],
+                                 d["C"], d["K"], d["Cin"], d["w_off"],
+                                 d["bf_off"], d["bfT_off"], d["rot_off"],
+                                 d["p8_off"],
+                                 native.current_stream_handle())
+        self._wbuf_fresh = True
 
     def _hip_forward(self, x: torch.Tensor, labels: torch.Tensor, B: int,
                      mode: int):
@@ -434,15 +460,16 @@ class DeepTrainer:
             self.model.grads.copy_(grads0)
             self.ws.loss_accum.zero_()
             self.ws.correct_accum.zero_()
+        # the captured graph contains NO cast (warmup left wbuf fresh at
+        # capture); refresh it for the restored params before any replay
+        self._hip_cast_weights(force=True)
         torch.cuda.synchronize()
 
     def _graph_body(self, B: int) -> None:
         self._hip_forward(self._gx, self._gl, B, MODE_TRAIN)
         self._hip_backward(self._gx, B)
         pdist.allreduce_grads(self.model.grads)
-        self._C.deep_update(self.model.params, self.model.grads,
-                            self.cfg.dt * self._scale(B),
-                            native.current_stream_handle())
+        self._hip_update(self._scale(B))
 
     def step_graph(self, x: torch.Tensor, labels: torch.Tensor) -> None:
         """Replay the captured step on a staged batch (two D2D copies)."""
@@ -475,10 +502,7 @@ class DeepTrainer:
                     with self.timers.phase("all-reduce"):
                         pdist.allreduce_grads(self.model.grads)
                     with self.timers.phase("update"):
-                        self._C.deep_update(self.model.params,
-                                            self.model.grads,
-                                            self.cfg.dt * scale,
-                                            native.current_stream_handle())
+                        self._hip_update(scale)
                 self._samples_seen += B * self.ctx.world_size
                 self.global_step += 1
                 return
@@ -486,9 +510,7 @@ class DeepTrainer:
             self._hip_backward(x, B)
             if apply_update:
                 pdist.allreduce_grads(self.model.grads)
-                self._C.deep_update(self.model.params, self.model.grads,
-                                    self.cfg.dt * scale,
-                                    native.current_stream_handle())
+                self._hip_update(scale)
         else:
             spec = self.model.spec
             xh = x.view(B, spec.in_h, spec.in_w, spec.in_ch)

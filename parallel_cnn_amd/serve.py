@@ -34,6 +34,8 @@ def create_app(cfg: Optional[TrainConfig] = None,
     trainer = (DeepTrainer(cfg) if cfg.model == "deepcnn" else Trainer(cfg))
     if ckpt:
         trainer.model.load(ckpt)
+        if hasattr(trainer, "invalidate_weight_cache"):
+            trainer.invalidate_weight_cache()
     spec_pixels = (32 * 32 * 3 if cfg.model == "deepcnn" else 784)
 
     app = FastAPI(title="parallel_cnn_amd inference")

@@ -55,6 +55,8 @@ def load_checkpoint(trainer, path: str) -> Optional[dict]:
     """Loads weights; restores global_step/epoch/RNG from the sidecar if
     present.  Returns the metadata dict (or None)."""
     trainer.model.load(path)
+    if hasattr(trainer, "invalidate_weight_cache"):
+        trainer.invalidate_weight_cache()
     meta_path = path + ".meta.json"
     if not os.path.exists(meta_path):
         return None
