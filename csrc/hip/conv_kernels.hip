@@ -26,6 +26,8 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
+#include <type_traits>
+
 #include "../lenet_dims.h"
 
 namespace pcnn_deep {
@@ -630,12 +632,24 @@ __global__ __launch_bounds__(256) void k_gemm_smallk(
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk)
       if (kk < kf) {
-        // clamped-by-construction: kq + kk*32 + 8 <= ldA (ldA >= K,
-        // K-span padding rows are zero in the cols buffer)
-        if (kq + kk * 32 + 8 <= ldA)
+        // in-bounds by construction: kq + kk*32 + 8 <= K <= ldA
+        // (K % 32 == 0, launcher-enforced)
+        if constexpr (std::is_same<act_t, bf16>::value) {
+          // one 16B load straight into the fragment
           dst[kk] = *reinterpret_cast<const bf16x8*>(src + kk * 32);
-        else
-          dst[kk] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        } else {
+          // fp32/fp16 activations: load + convert (the general kernels
+          // do the same at their LDS stage)
+          float v8[8];
+          ld8v(src + kk * 32, v8);
+          union {
+            __bf16 h[8];
+            bf16x8 v;
+          } t2;
+#pragma unroll
+          for (int u = 0; u < 8; ++u) t2.h[u] = (__bf16)v8[u];
+          dst[kk] = t2.v;
+        }
       }
   };
 
