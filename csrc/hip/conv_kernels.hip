@@ -81,6 +81,42 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
   out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
 }
 
+// Unaligned 8-element load -> fp32 (element-aligned only — the small-Cin
+// im2col spans start at arbitrary channel offsets).  The aligned(2/4)
+// vector types let LLVM emit unaligned-capable global loads (or split
+// them) instead of UB.
+__device__ __forceinline__ void ld8v_u(const bf16* p, float* out) {
+  typedef ushort4 __attribute__((aligned(2))) ushort4_u;
+  const ushort4_u a = *reinterpret_cast<const ushort4_u*>(p);
+  const ushort4_u b = *reinterpret_cast<const ushort4_u*>(p + 4);
+  const bf16* ea = reinterpret_cast<const bf16*>(&a);
+  const bf16* eb = reinterpret_cast<const bf16*>(&b);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    out[u] = (float)ea[u];
+    out[u + 4] = (float)eb[u];
+  }
+}
+__device__ __forceinline__ void ld8v_u(const fp16* p, float* out) {
+  typedef ushort4 __attribute__((aligned(2))) ushort4_u;
+  const ushort4_u a = *reinterpret_cast<const ushort4_u*>(p);
+  const ushort4_u b = *reinterpret_cast<const ushort4_u*>(p + 4);
+  const fp16* ea = reinterpret_cast<const fp16*>(&a);
+  const fp16* eb = reinterpret_cast<const fp16*>(&b);
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    out[u] = (float)ea[u];
+    out[u + 4] = (float)eb[u];
+  }
+}
+__device__ __forceinline__ void ld8v_u(const float* p, float* out) {
+  typedef float4 __attribute__((aligned(4))) float4_u;
+  const float4_u a = *reinterpret_cast<const float4_u*>(p);
+  const float4_u b = *reinterpret_cast<const float4_u*>(p + 4);
+  out[0] = a.x; out[1] = a.y; out[2] = a.z; out[3] = a.w;
+  out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
+}
+
 // Magic-multiply unsigned division (divisor known on the host): exact
 // for n < 2^38/d and n*M < 2^64, i.e. n < 2^26 with d <= 4096 — the
 // launcher checks the domain and falls back to the 64-bit-div kernel
@@ -176,24 +212,48 @@ __global__ void k_im2col(const act_t* __restrict__ x, act_t* __restrict__ cols,
         }
       }
     } else {
-      // clamped unconditional loads + select-to-zero (same idiom as the
-      // vec path): 8 independent scalar loads issue back-to-back instead
-      // of a branchy serial chain (the old guarded form was this
-      // kernel's whole cost on the Cin=3 stage)
+      // Small Cin: consecutive kc within one KERNEL ROW map to
+      // consecutive x memory (kc = (i*K+j)*Cin+ci, ci fastest, and
+      // NHWC x is (..., iw, ci)-contiguous) — so an 8-span that stays
+      // inside the row AND inside the image is ONE 16B load.  Border /
+      // row-crossing spans use clamped unconditional scalar loads +
+      // select-to-zero (a branchy guarded chain was this kernel's
+      // whole cost on the Cin=3 stage).
+      const int r0 = kc - (i * K + j) * Cin + j * Cin;  // = kc % rowc
+      const int jlast = j + (ci + 7) / Cin;
+      const int ih = oh + i - P;
+      if (r0 + 8 <= rowc && kc + 8 <= Kc && ih >= 0 && ih < H &&
+          ow + j - P >= 0 && ow + jlast - P < W) {
+        float v8[8];
+        ld8v_u(xb + ((long long)ih * W + (ow + j - P)) * Cin + ci, v8);
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int ih = oh + i - P;
-        const int iw = ow + j - P;
-        const int ihc = min(max(ih, 0), H - 1);
-        const int iwc = min(max(iw, 0), W - 1);
-        const float val = ldf(xb + ((long long)ihc * W + iwc) * Cin + ci);
-        out[u] = (act_t)((kc + u < Kc && ih == ihc && iw == iwc) ? val
-                                                                 : 0.f);
-        if (++ci >= Cin) {
-          ci = 0;
-          if (++j >= K) {
-            j = 0;
-            ++i;
+        for (int u = 0; u < 8; ++u) out[u] = (act_t)v8[u];
+        // advance (i, j, ci) by 8 positions
+        ci += 8;
+        j += ci / Cin;
+        ci %= Cin;
+        if (j >= K) {
+          i += j / K;
+          j %= K;
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int ih2 = oh + i - P;
+          const int iw = ow + j - P;
+          const int ihc = min(max(ih2, 0), H - 1);
+          const int iwc = min(max(iw, 0), W - 1);
+          const float val =
+              ldf(xb + ((long long)ihc * W + iwc) * Cin + ci);
+          out[u] = (act_t)((kc + u < Kc && ih2 == ihc && iw == iwc)
+                               ? val
+                               : 0.f);
+          if (++ci >= Cin) {
+            ci = 0;
+            if (++j >= K) {
+              j = 0;
+              ++i;
+            }
           }
         }
       }
