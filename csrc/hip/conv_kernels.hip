@@ -86,35 +86,19 @@ __device__ __forceinline__ void ld8v(const float* p, float* out) {
 // vector types let LLVM emit unaligned-capable global loads (or split
 // them) instead of UB.
 __device__ __forceinline__ void ld8v_u(const bf16* p, float* out) {
-  typedef ushort4 __attribute__((aligned(2))) ushort4_u;
-  const ushort4_u a = *reinterpret_cast<const ushort4_u*>(p);
-  const ushort4_u b = *reinterpret_cast<const ushort4_u*>(p + 4);
-  const bf16* ea = reinterpret_cast<const bf16*>(&a);
-  const bf16* eb = reinterpret_cast<const bf16*>(&b);
+  bf16 t[8];
+  __builtin_memcpy(t, p, 16);
 #pragma unroll
-  for (int u = 0; u < 4; ++u) {
-    out[u] = (float)ea[u];
-    out[u + 4] = (float)eb[u];
-  }
+  for (int u = 0; u < 8; ++u) out[u] = (float)t[u];
 }
 __device__ __forceinline__ void ld8v_u(const fp16* p, float* out) {
-  typedef ushort4 __attribute__((aligned(2))) ushort4_u;
-  const ushort4_u a = *reinterpret_cast<const ushort4_u*>(p);
-  const ushort4_u b = *reinterpret_cast<const ushort4_u*>(p + 4);
-  const fp16* ea = reinterpret_cast<const fp16*>(&a);
-  const fp16* eb = reinterpret_cast<const fp16*>(&b);
+  fp16 t[8];
+  __builtin_memcpy(t, p, 16);
 #pragma unroll
-  for (int u = 0; u < 4; ++u) {
-    out[u] = (float)ea[u];
-    out[u + 4] = (float)eb[u];
-  }
+  for (int u = 0; u < 8; ++u) out[u] = (float)t[u];
 }
 __device__ __forceinline__ void ld8v_u(const float* p, float* out) {
-  typedef float4 __attribute__((aligned(4))) float4_u;
-  const float4_u a = *reinterpret_cast<const float4_u*>(p);
-  const float4_u b = *reinterpret_cast<const float4_u*>(p + 4);
-  out[0] = a.x; out[1] = a.y; out[2] = a.z; out[3] = a.w;
-  out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
+  __builtin_memcpy(out, p, 32);
 }
 
 // Magic-multiply unsigned division (divisor known on the host): exact
