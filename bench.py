@@ -47,10 +47,12 @@ def main() -> int:
     p.add_argument("--overlap-comm", default=None,
                    action=argparse.BooleanOptionalAction,
                    help="two-bucket DP: overlap the fc/pool grad all-reduce "
-                        "with the conv wgrad (default: ON when world>1 — "
-                        "the 9.4 KB bucket is pure latency and the step is "
-                        "~19 us, so overlap is the right multi-GPU default; "
-                        "--no-overlap-comm forces it off)")
+                        "with the conv wgrad.  Default OFF: the 9.4 KB "
+                        "LeNet bucket is pure latency, and splitting it "
+                        "pays a second collective latency that the ~7 us "
+                        "of remaining wgrad cannot hide — ONE fused bucket "
+                        "is the design point (SURVEY 5.8); measure with "
+                        "--overlap-comm on multi-GPU nodes")
     args = p.parse_args()
 
     from parallel_cnn_amd.config import TrainConfig
@@ -66,8 +68,6 @@ def main() -> int:
                       loss=args.loss)
     device = cfg.resolved_device()
     ctx = pdist.init_from_env(device)
-    if args.overlap_comm is None and ctx.world_size > 1:
-        cfg.overlap_comm = True  # multi-GPU default (see --help)
     n_gpus = ctx.world_size if ctx.world_size > 1 else args.gpus
     if ctx.world_size == 1 and args.gpus > 1:
         raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
