@@ -258,3 +258,30 @@ def test_deep_training_converges_gpu(device):
         losses.append(t.train_epoch(x, y, log=lambda *a: None))
     torch.cuda.synchronize()
     assert losses[-1] < losses[0] * 0.99, losses
+
+
+def test_weight_cache_fresh_after_checkpoint_load(device, tmp_path):
+    """The fused update+cast keeps bf16 weight images fresh across steps;
+    a checkpoint load mutates params OUTSIDE the engine and must
+    invalidate them — a stale cache would silently serve the old
+    weights' predictions."""
+    from parallel_cnn_amd.utils.checkpoint import (load_checkpoint,
+                                                   save_checkpoint)
+    cfg = TrainConfig(batch_size=8, device="cuda", backend="hip",
+                      act_dtype="bf16", log_interval=0)
+    t = DeepTrainer(cfg)
+    x, y = synthetic_images(8, 32, 32, 3, seed=17)
+    t.step(*t.stage_batch(x, y))          # update+cast ran; cache fresh
+    torch.cuda.synchronize()
+    want = t.forward_logits(x)
+    path = str(tmp_path / "w.bin")
+    save_checkpoint(t, path)
+
+    t2 = DeepTrainer(TrainConfig(batch_size=8, device="cuda",
+                                 backend="hip", act_dtype="bf16",
+                                 log_interval=0, seed=99))
+    t2.step(*t2.stage_batch(x, y))        # t2's cache is fresh for ITS params
+    load_checkpoint(t2, path)             # must invalidate
+    got = t2.forward_logits(x)
+    assert torch.allclose(got, want, atol=1e-3), \
+        (got - want).abs().max().item()
