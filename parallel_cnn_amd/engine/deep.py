@@ -63,7 +63,6 @@ class DeepWorkspace:
         fusepool_on = os.environ.get("PCNN_DEEP_FUSEPOOL", "1") == "1"
         self.stage_pad8 = [implicit and pad8_on and 0 < st.cin < 8
                            for st in spec.stages]
-        self._fusepool_on = fusepool_on
         # pool-forward fuses into the conv GEMM epilogue when every pool
         # window lies inside one 64-row M-tile (true for all standard
         # shapes: W in {8,16,32}, pool 2x2) and the stage fits one n-tile

@@ -91,3 +91,24 @@ def test_deepcnn_mnist_mismatch_errors():
                    "--data", "mnist", "--epochs", "1"])
     assert out.returncode != 0
     assert "28x28x1" in (out.stdout + out.stderr)
+
+
+def test_cli_checkpoint_resume(tmp_path):
+    """CLI end-to-end exact resume: 2 straight epochs == 1 epoch -> save
+    -> --ckpt-load -> remaining epoch (the sidecar's epoch cursor)."""
+    ck1 = str(tmp_path / "full.bin")
+    ck2 = str(tmp_path / "half.bin")
+    ck3 = str(tmp_path / "resumed.bin")
+    base = ["--device", "cpu", "--train-count", "64", "--test-count", "32",
+            "--batch-size", "16", "--log-interval", "0", "--threshold", "0"]
+    out = run_cli(base + ["--epochs", "2", "--ckpt-save", ck1])
+    assert out.returncode == 0, out.stdout + out.stderr
+    out = run_cli(base + ["--epochs", "1", "--ckpt-save", ck2])
+    assert out.returncode == 0, out.stdout + out.stderr
+    out = run_cli(base + ["--epochs", "2", "--ckpt-load", ck2,
+                          "--ckpt-save", ck3])
+    assert out.returncode == 0, out.stdout + out.stderr
+    import numpy as np
+    full = np.fromfile(ck1, dtype="<f4")
+    resumed = np.fromfile(ck3, dtype="<f4")
+    assert np.array_equal(full, resumed)
