@@ -1968,7 +1968,7 @@ int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
   if (epilogue == 3 &&
       (pw == nullptr || pout == nullptr || N > BN || PK < 1 || XW < 1 ||
        64 % XW != 0 || (64 / XW) % PK != 0 || XH % PK != 0 ||
-       M % 64 != 0 || (N % 8) != 0 || 256 % (PK * PK * (N / 8)) != 0))
+       M % 64 != 0 || (N % 8) != 0))
     return -4;  // fused-pool preconditions (engine falls back)
   const unsigned long long fd_cin = fdiv_magic((unsigned)(XC > 0 ? XC : 1));
   const unsigned long long fd_k = fdiv_magic((unsigned)(XK > 0 ? XK : 1));
@@ -1994,8 +1994,13 @@ int pcnn_deep_gemm_ex4(const void* A, const float* Bsrc, const void* Bpre,
   // at 64 WGs for 20 us of 1024-WG work); split K until ~1024 WGs.
   int ks_eff = 1, tpc = 0;
   const int ktiles = (K + BKC - 1) / BKC;
+  // mode-3 split additionally needs pool windows block-aligned in the
+  // combine kernel (256 threads cover whole windows); widths like N=48
+  // fail it and simply run unsplit
+  const bool epi3_ok =
+      epilogue != 3 || 256 % (PK * PK * (N / 8)) == 0;
   if (c32 != nullptr && mn < 768 && ktiles > 1 && (N % 8) == 0 &&
-      ldC == N) {
+      ldC == N && epi3_ok) {
     int ks = (int)(1024 / mn) + 1;
     if (ks > ktiles) ks = ktiles;
     tpc = (ktiles + ks - 1) / ks;

@@ -285,3 +285,26 @@ def test_weight_cache_fresh_after_checkpoint_load(device, tmp_path):
     got = t2.forward_logits(x)
     assert torch.allclose(got, want, atol=1e-3), \
         (got - want).abs().max().item()
+
+
+def test_custom_channel_widths_gpu(device):
+    """Non-power-of-two widths (48 channels: N%16==0 but the split-epi
+    window alignment fails -> must run unsplit, not crash) train on the
+    hip path and match the oracle."""
+    cfg = TrainConfig(batch_size=8, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0,
+                      deep_channels="16,32,48")
+    t = DeepTrainer(cfg)
+    x, labels = synthetic_images(8, 32, 32, 3, seed=23, structured=False)
+    xb, lb = t.stage_batch(x, labels)
+    t.step(xb, lb)
+    torch.cuda.synchronize()
+    from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
+    ref = DeepCNN(seed=cfg.seed, spec=DeepCNNSpec(channels=(16, 32, 48)))
+    xh = x.view(8, 32, 32, 3)
+    acts, pouts, y = deep_ref.forward(xh, ref)
+    grads, loss = deep_ref.backward(xh, ref, acts, pouts, y, labels)
+    with torch.no_grad():
+        ref.params += t.cfg.dt * (1.0 / 8) * grads
+    diff = (t.model.params.cpu() - ref.params).abs().max().item()
+    assert diff < 5e-3, diff
