@@ -169,3 +169,20 @@ def test_deep_grad_accumulation_matches_big_batch():
     tb.step(*tb.stage_batch(x, y))
     assert torch.allclose(ta.model.params, tb.model.params, atol=1e-6)
     assert ta.model.grads.abs().sum() == 0  # consumed at the boundary
+
+
+def test_four_stage_spec():
+    """Deeper-than-default family: 4 stages halve 32 -> 2; the spec, the
+    oracle and the torchref engine all compose."""
+    from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
+    spec = DeepCNNSpec(channels=(16, 32, 32, 64))
+    assert [(s.h, s.w) for s in spec.stages] == [(32, 32), (16, 16),
+                                                (8, 8), (4, 4)]
+    assert spec.fc_in == 2 * 2 * 64
+    cfg = TrainConfig(backend="torchref", device="cpu", batch_size=4,
+                      log_interval=0, deep_channels="16,32,32,64")
+    t = DeepTrainer(cfg)
+    x, y = synthetic_images(8, 32, 32, 3, seed=2)
+    t.step(*t.stage_batch(x[:4], y[:4]))
+    l, n = t.consume_loss()
+    assert n == 4 and l > 0

@@ -308,3 +308,25 @@ def test_custom_channel_widths_gpu(device):
         ref.params += t.cfg.dt * (1.0 / 8) * grads
     diff = (t.model.params.cpu() - ref.params).abs().max().item()
     assert diff < 5e-3, diff
+
+
+def test_four_stage_hip_matches_oracle(device):
+    """4-stage family on the hip path (last stage h*w = 16 < 64: the
+    fused-pool tile alignment fails there and must fall back cleanly)."""
+    cfg = TrainConfig(batch_size=8, device="cuda", backend="hip",
+                      act_dtype="fp32", log_interval=0,
+                      deep_channels="16,32,32,64")
+    t = DeepTrainer(cfg)
+    x, labels = synthetic_images(8, 32, 32, 3, seed=29, structured=False)
+    t.step(*t.stage_batch(x, labels))
+    torch.cuda.synchronize()
+    from parallel_cnn_amd.models.deepcnn import DeepCNNSpec
+    ref = DeepCNN(seed=cfg.seed,
+                  spec=DeepCNNSpec(channels=(16, 32, 32, 64)))
+    xh = x.view(8, 32, 32, 3)
+    acts, pouts, y = deep_ref.forward(xh, ref)
+    grads, _ = deep_ref.backward(xh, ref, acts, pouts, y, labels)
+    with torch.no_grad():
+        ref.params += t.cfg.dt * (1.0 / 8) * grads
+    diff = (t.model.params.cpu() - ref.params).abs().max().item()
+    assert diff < 5e-3, diff
