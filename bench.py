@@ -162,6 +162,7 @@ def main() -> int:
             "per_gpu_batch": B,
             "input": "32x32x3" if args.model == "deepcnn" else "28x28x1",
             "parallelism": f"dp{n_gpus}",
+            "dist_backend": ctx.backend or "none",
             "pool": args.pool,
             "loss": args.loss,
             "backend": trainer.backend,

@@ -67,8 +67,23 @@ def init_from_env(device: str = "auto") -> DistContext:
     if use_gpu:
         torch.cuda.set_device(local_rank % n_gpus)
     if not dist.is_initialized():
-        dist.init_process_group(
-            backend=backend, timeout=datetime.timedelta(seconds=120))
+        try:
+            dist.init_process_group(
+                backend=backend, timeout=datetime.timedelta(seconds=120))
+        except Exception as e:
+            if backend != "gloo":
+                # a broken RCCL install/topology should degrade to a slow
+                # measured run, not a crashed one; the bench JSON reports
+                # which transport actually ran
+                import sys
+                print(f"[pcnn dist] {backend} init failed ({e}); "
+                      f"falling back to gloo", file=sys.stderr, flush=True)
+                backend = "gloo"
+                dist.init_process_group(
+                    backend=backend,
+                    timeout=datetime.timedelta(seconds=120))
+            else:
+                raise
     return DistContext(rank=rank, world_size=world, local_rank=local_rank,
                        backend=backend)
 

@@ -502,3 +502,28 @@ def test_native_cli_mnist_idx(device, tmp_path):
         capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "Learning" in out.stdout and "Error Rate: " in out.stdout
+
+
+def test_device_prefetcher_batches_match_slices(device):
+    """Regression for the round-1 prefetcher race (pinned staging buffer
+    overwritten while its H2D copy was in flight): every prefetched
+    batch must equal the corresponding host slice exactly."""
+    from parallel_cnn_amd.data.pipeline import DevicePrefetcher
+    torch.manual_seed(5)
+    N, B, P = 64 * 12, 64, S.IN_PIX
+    x = torch.rand(N, P)
+    labels = torch.randint(0, 10, (N,))
+    pf = DevicePrefetcher(x, labels, B, torch.device("cuda"),
+                          torch.bfloat16)
+    seen = 0
+    for i, (xb, lb) in enumerate(pf):
+        # force the host loop far ahead of the device to stress the
+        # in-flight-overwrite window the race fix guards
+        want_x = x[i * B:(i + 1) * B].to(torch.bfloat16)
+        want_l = labels[i * B:(i + 1) * B].to(torch.int32)
+        got_x = xb.cpu()
+        got_l = lb.cpu()
+        assert torch.equal(got_x, want_x), f"batch {i} torn"
+        assert torch.equal(got_l, want_l), f"labels {i} torn"
+        seen += 1
+    assert seen == N // B
