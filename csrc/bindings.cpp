@@ -72,6 +72,13 @@ int pcnn_deep_cast_all(const float* params, void* wbuf, int n_stages,
                        const long long* bf_off, const long long* bfT_off,
                        const long long* rot_off, const long long* p8_off,
                        void* stream);
+int pcnn_deep_wgrad_multi(int n_stages, const void* const* a,
+                          const void* const* dpre, float* const* dW,
+                          float* const* db, const long long* M,
+                          const int* KcP, const int* N, const int* MS,
+                          const int* implicit, const int* XH, const int* XW,
+                          const int* XC, const int* XK, const int* XP,
+                          int actf, void* stream);
 int pcnn_deep_update_cast(float* params, float* grads, long long n,
                           float step, void* wbuf, int n_stages, const int* R,
                           const int* C, const int* K, const int* Cin,
@@ -353,6 +360,45 @@ void deep_update_cast(at::Tensor params, at::Tensor grads, double step,
             "deep_update_cast");
 }
 
+void deep_wgrad_multi(std::vector<at::Tensor> a,
+                      std::vector<at::Tensor> dpre,
+                      std::vector<at::Tensor> dW, std::vector<at::Tensor> db,
+                      std::vector<int64_t> M, std::vector<int64_t> KcP,
+                      std::vector<int64_t> N, std::vector<int64_t> MS,
+                      std::vector<int64_t> implicit, std::vector<int64_t> XH,
+                      std::vector<int64_t> XW, std::vector<int64_t> XC,
+                      std::vector<int64_t> XK, std::vector<int64_t> XP,
+                      int64_t stream) {
+  const size_t n = a.size();
+  TORCH_CHECK(n >= 1 && n <= 8, "deep_wgrad_multi: 1..8 stages");
+  const void* ap[8];
+  const void* dp[8];
+  float* wp[8];
+  float* bp[8];
+  long long Mi[8];
+  int Ki[8], Ni[8], MSi[8], Ii[8], XHi[8], XWi[8], XCi[8], XKi[8], XPi[8];
+  for (size_t s = 0; s < n; ++s) {
+    ap[s] = a[s].data_ptr();
+    dp[s] = dpre[s].data_ptr();
+    wp[s] = dW[s].data_ptr<float>();
+    bp[s] = db[s].data_ptr<float>();
+    Mi[s] = M[s];
+    Ki[s] = (int)KcP[s];
+    Ni[s] = (int)N[s];
+    MSi[s] = (int)MS[s];
+    Ii[s] = (int)implicit[s];
+    XHi[s] = (int)XH[s];
+    XWi[s] = (int)XW[s];
+    XCi[s] = (int)XC[s];
+    XKi[s] = (int)XK[s];
+    XPi[s] = (int)XP[s];
+  }
+  check_hip(pcnn_deep_wgrad_multi((int)n, ap, dp, wp, bp, Mi, Ki, Ni, MSi,
+                                  Ii, XHi, XWi, XCi, XKi, XPi,
+                                  act_flag(dpre[0]), (void*)stream),
+            "deep_wgrad_multi");
+}
+
 void deep_pad_channels(at::Tensor x, at::Tensor x8, int64_t npix,
                        int64_t Cin, int64_t stream) {
   check_hip(pcnn_deep_pad_channels(x.data_ptr(), x8.data_ptr(), npix,
@@ -548,6 +594,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("deep_cast_all", &deep_cast_all);
   m.def("deep_pad_channels", &deep_pad_channels);
   m.def("deep_update_cast", &deep_update_cast);
+  m.def("deep_wgrad_multi", &deep_wgrad_multi);
   m.def("deep_remap_dw8", &deep_remap_dw8);
   m.def("deep_wgrad_gemm", &deep_wgrad_gemm, py::arg("cols"),
         py::arg("dpre"), py::arg("dW"), py::arg("M"), py::arg("KcP"),

@@ -777,8 +777,9 @@ __global__ __launch_bounds__(256) void k_gemm_smallk(
 // element is staged through write_lds exactly once per (kct == 0) WG,
 // accumulated per-thread and LDS-reduced after the MFMA loop.
 template <typename act_t>
-__global__ __launch_bounds__(256) void k_wgrad_gemm(
-    const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
+__device__ __forceinline__ void wgrad_body(
+    unsigned bidx, const act_t* __restrict__ cols,
+    const act_t* __restrict__ dpre,
     float* __restrict__ dW, float* __restrict__ part, long long M, int KcP,
     int N, int MS, const act_t* __restrict__ imx, int XH, int XW, int XC,
     int XK, int XP, float* __restrict__ db, unsigned long long fd_cin,
@@ -790,8 +791,8 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
   const int wv = tid >> 6;
   const int ntiles = (N + BN - 1) / BN;
   const int ktiles = (KcP + BM - 1) / BM;
-  const int kct = blockIdx.x % max(1, ktiles);
-  const int rest = blockIdx.x / max(1, ktiles);
+  const int kct = bidx % max(1, ktiles);
+  const int rest = bidx / max(1, ktiles);
   const int ntile = rest % ntiles;
   const int slice = rest / ntiles;
   const int kc0 = kct * BM;
@@ -982,6 +983,47 @@ __global__ __launch_bounds__(256) void k_wgrad_gemm(
       }
     }
   }
+}
+
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_wgrad_gemm(
+    const act_t* __restrict__ cols, const act_t* __restrict__ dpre,
+    float* __restrict__ dW, float* __restrict__ part, long long M, int KcP,
+    int N, int MS, const act_t* __restrict__ imx, int XH, int XW, int XC,
+    int XK, int XP, float* __restrict__ db, unsigned long long fd_cin,
+    unsigned long long fd_k, unsigned long long fd_xw,
+    unsigned long long fd_xh) {
+  wgrad_body<act_t>(blockIdx.x, cols, dpre, dW, part, M, KcP, N, MS, imx,
+                    XH, XW, XC, XK, XP, db, fd_cin, fd_k, fd_xw, fd_xh);
+}
+
+// All conv stages' weight-grad GEMMs in ONE launch: three sequential
+// ~500-WG launches each run at ~2 WGs/CU with the per-iteration stall
+// exposed; the combined grid fills the chip and lets the stages'
+// latencies hide each other.  Stage found by cumulative block offsets.
+struct WgDesc {
+  int n;                       // stages (<= 8)
+  long long blk_cum[9];        // cumulative grid blocks
+  const void* a[8];            // cols (materialized) or imx (implicit)
+  const void* dpre[8];
+  float* dW[8];
+  float* db[8];
+  long long M[8];
+  int KcP[8], N[8], MS[8], implicit[8];
+  int XH[8], XW[8], XC[8], XK[8], XP[8];
+  unsigned long long fd_cin[8], fd_k[8], fd_xw[8], fd_xh[8];
+};
+template <typename act_t>
+__global__ __launch_bounds__(256) void k_wgrad_multi(WgDesc d) {
+  int s = 0;
+  while (blockIdx.x >= d.blk_cum[s + 1]) ++s;
+  const unsigned bidx = (unsigned)(blockIdx.x - d.blk_cum[s]);
+  wgrad_body<act_t>(
+      bidx, (const act_t*)d.a[s], (const act_t*)d.dpre[s], d.dW[s],
+      nullptr, d.M[s], d.KcP[s], d.N[s], d.MS[s],
+      d.implicit[s] ? (const act_t*)d.a[s] : nullptr, d.XH[s], d.XW[s],
+      d.XC[s], d.XK[s], d.XP[s], d.db[s], d.fd_cin[s], d.fd_k[s],
+      d.fd_xw[s], d.fd_xh[s]);
 }
 
 // Fold MS weight-grad slabs into dW (+=): 8 fp32 per thread, the slab
@@ -2165,6 +2207,53 @@ int pcnn_deep_wgrad_gemm_ex(const void* cols, const void* dpre, float* dW,
                             int XP, int actf, void* stream) {
   return pcnn_deep_wgrad_gemm_ex2(cols, dpre, dW, part, M, KcP, N, MS, imx,
                                   XH, XW, XC, XK, XP, nullptr, actf, stream);
+}
+
+// n_stages wgrad GEMMs in one launch (see k_wgrad_multi).  a[s] is the
+// materialized cols when implicit[s]==0, else the NHWC source tensor.
+int pcnn_deep_wgrad_multi(int n_stages, const void* const* a,
+                          const void* const* dpre, float* const* dW,
+                          float* const* db, const long long* M,
+                          const int* KcP, const int* N, const int* MS,
+                          const int* implicit, const int* XH, const int* XW,
+                          const int* XC, const int* XK, const int* XP,
+                          int actf, void* stream) {
+  if (n_stages < 1 || n_stages > 8) return -2;
+  WgDesc d;
+  d.n = n_stages;
+  d.blk_cum[0] = 0;
+  for (int s = 0; s < n_stages; ++s) {
+    if (implicit[s] &&
+        ((XC[s] % 8) != 0 || M[s] >= (1LL << 26) || XC[s] > 4096 ||
+         XK[s] > 4096 || XW[s] > 4096 || XH[s] > 4096))
+      return -3;
+    d.a[s] = a[s];
+    d.dpre[s] = dpre[s];
+    d.dW[s] = dW[s];
+    d.db[s] = db[s];
+    d.M[s] = M[s];
+    d.KcP[s] = KcP[s];
+    d.N[s] = N[s];
+    d.MS[s] = MS[s];
+    d.implicit[s] = implicit[s];
+    d.XH[s] = XH[s];
+    d.XW[s] = XW[s];
+    d.XC[s] = XC[s];
+    d.XK[s] = XK[s];
+    d.XP[s] = XP[s];
+    d.fd_cin[s] = fdiv_magic((unsigned)(XC[s] > 0 ? XC[s] : 1));
+    d.fd_k[s] = fdiv_magic((unsigned)(XK[s] > 0 ? XK[s] : 1));
+    d.fd_xw[s] = fdiv_magic((unsigned)(XW[s] > 0 ? XW[s] : 1));
+    d.fd_xh[s] = fdiv_magic((unsigned)(XH[s] > 0 ? XH[s] : 1));
+    const long long blocks = (long long)((KcP[s] + BM - 1) / BM) *
+                             ((N[s] + BN - 1) / BN) * MS[s];
+    d.blk_cum[s + 1] = d.blk_cum[s] + blocks;
+  }
+  dim3 grid((unsigned)d.blk_cum[n_stages]), block(256);
+  PCNN_DISPATCH(actf, hipLaunchKernelGGL((k_wgrad_multi<act_t>), grid,
+                                          block, 0, (hipStream_t)stream,
+                                          d));
+  return (int)hipGetLastError();
 }
 
 int pcnn_deep_pad_channels(const void* x, void* x8, long long npix, int Cin,

@@ -340,12 +340,14 @@ class DeepTrainer:
         self._C.deep_fc_wgrad(w.dz, w.pouts[-1], m.grad_view("fc_w"),
                               m.grad_view("fc_b"), B, spec.fc_in,
                               spec.n_classes, fs, st_h)
+        # pass 1 — the dependency CHAIN, descending: pool wgrad+bwd of
+        # stage i (dapre in place over acts[i]), then dgrad into
+        # dppre[i-1].  Conv wgrads move to pass 2: each only needs its
+        # stage's dapre (stable once written) and its input activation,
+        # so all of them batch into ONE launch afterwards.
         for i in range(nstage - 1, -1, -1):
             st = spec.stages[i]
             M = B * st.h * st.w
-            # ~halved from the 2-items/thread target: the 17 per-WG
-            # atomics land on one cache line (~3ns/op serialized), so the
-            # sweet spot trades compute depth against the atomic tail
             # ~2 pooled items (of 8 channels) per thread: the fused
             # wgrad+bwd kernel writes K*K 16B stores per item, so deeper
             # per-thread loops no longer amortize anything
@@ -367,34 +369,7 @@ class DeepTrainer:
                                       m.view(f"pool{i}_w"), w.acts[i], B,
                                       st.h, st.w, st.cout, st.pool_k, st_h)
             dapre = w.acts[i]
-            ms = self.ws.wgrad_ms(st, M)
-            # measured (tools/deep_sweep.py): atomic combine beats the
-            # slab+reduce mode at every MS, and ~512 total WGs is optimal
-            # (per-WG fixed costs dominate beyond that).  The conv BIAS
-            # grad is folded into the wgrad GEMM (db argument) — the
-            # separate k_colsum/k_colsum_fin passes are gone.
-            x_in = x if i == 0 else w.pouts[i - 1]
             implicit = w.stage_implicit[i]
-            pad8 = w.stage_pad8[i]
-            if pad8:
-                # implicit against the padded x8 (still valid from this
-                # step's forward); the 8-padded dW lands in scratch and
-                # k_remap_dw8 folds it into the model's flat layout
-                self._C.deep_wgrad_gemm(w.x8[i], dapre, w.dw8[i], M,
-                                        st.k * st.k * 8, st.cout, ms, st_h,
-                                        w.x8[i], st.h, st.w, 8, st.k,
-                                        st.pad,
-                                        db=m.grad_view(f"conv{i}_b"))
-                self._C.deep_remap_dw8(w.dw8[i], m.grad_view(f"conv{i}_w"),
-                                       st.k * st.k, st.cin, st.cout, st_h)
-            else:
-                self._C.deep_wgrad_gemm(
-                    x_in if implicit else w.cols[i], dapre,
-                    m.grad_view(f"conv{i}_w"), M, st.kcp,
-                    st.cout, ms, st_h,
-                    x_in if implicit else torch.empty(0),
-                    st.h, st.w, st.cin, st.k, st.pad,
-                    db=m.grad_view(f"conv{i}_b"))
             if i > 0:
                 if implicit:
                     # dgrad-as-conv: implicit im2col of dapre against the
@@ -419,6 +394,49 @@ class DeepTrainer:
                                                w.dppre[i - 1], B, st.h,
                                                st.w, st.cin, st.k, st.pad,
                                                st.kcp, st_h)
+        # pass 2 — every stage's conv wgrad GEMM in ONE launch (the conv
+        # BIAS colsum stays folded in).  Three sequential ~500-WG
+        # launches each ran at ~2 WGs/CU with the per-iteration stall
+        # exposed; the combined grid fills the chip.  Measured MS optima
+        # per stage are unchanged (wgrad_ms).
+        da, dd, dwv, dbv = [], [], [], []
+        Ml, Kl, Nl, MSl, Il, XHl, XWl, XCl, XKl, XPl = ([] for _ in range(10))
+        for i, st in enumerate(spec.stages):
+            M = B * st.h * st.w
+            x_in = x if i == 0 else w.pouts[i - 1]
+            if w.stage_pad8[i]:
+                da.append(w.x8[i])
+                dwv.append(w.dw8[i])
+                Kl.append(st.k * st.k * 8)
+                Il.append(1)
+                XCl.append(8)
+            elif w.stage_implicit[i]:
+                da.append(x_in)
+                dwv.append(m.grad_view(f"conv{i}_w"))
+                Kl.append(st.kcp)
+                Il.append(1)
+                XCl.append(st.cin)
+            else:
+                da.append(w.cols[i])
+                dwv.append(m.grad_view(f"conv{i}_w"))
+                Kl.append(st.kcp)
+                Il.append(0)
+                XCl.append(st.cin)
+            dd.append(w.acts[i])
+            dbv.append(m.grad_view(f"conv{i}_b"))
+            Ml.append(M)
+            Nl.append(st.cout)
+            MSl.append(self.ws.wgrad_ms(st, M))
+            XHl.append(st.h)
+            XWl.append(st.w)
+            XKl.append(st.k)
+            XPl.append(st.pad)
+        self._C.deep_wgrad_multi(da, dd, dwv, dbv, Ml, Kl, Nl, MSl, Il,
+                                 XHl, XWl, XCl, XKl, XPl, st_h)
+        for i, st in enumerate(spec.stages):
+            if w.stage_pad8[i]:
+                self._C.deep_remap_dw8(w.dw8[i], m.grad_view(f"conv{i}_w"),
+                                       st.k * st.k, st.cin, st.cout, st_h)
 
     # ----------------------------------------------------------------- graph
     def enable_graph(self) -> None:
