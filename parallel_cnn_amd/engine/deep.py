@@ -328,10 +328,62 @@ class DeepTrainer:
                             dflat=w.dppre[-1] if mode == MODE_TRAIN
                             else torch.empty(0))
 
+    def _wgrad_side(self):
+        """Lazy side stream + events for the async-wgrad mode: weight
+        grads run on a second HIP stream, co-resident with the main
+        stream's pool/dgrad chain (chip fill via concurrency AND the
+        chain hides them).  PCNN_DEEP_WGRAD_MODE=multi keeps the
+        single-launch batched mode instead."""
+        if not hasattr(self, "_wside"):
+            self._wside = torch.cuda.Stream(device=self.device)
+            self._wev = [torch.cuda.Event() for _ in
+                         range(len(self.model.spec.stages) + 1)]
+            self._wdone = torch.cuda.Event()
+        return self._wside
+
+    def _launch_wgrad_stage(self, x: torch.Tensor, B: int, i: int,
+                            st_h: int) -> None:
+        """One stage's conv weight-grad GEMM (+fused bias colsum) on the
+        given stream (the async-wgrad side stream)."""
+        m, w, spec = self.model, self.ws, self.model.spec
+        st = spec.stages[i]
+        M = B * st.h * st.w
+        ms = self.ws.wgrad_ms(st, M)
+        x_in = x if i == 0 else w.pouts[i - 1]
+        if w.stage_pad8[i]:
+            self._C.deep_wgrad_gemm(w.x8[i], w.acts[i], w.dw8[i], M,
+                                    st.k * st.k * 8, st.cout, ms, st_h,
+                                    w.x8[i], st.h, st.w, 8, st.k, st.pad,
+                                    db=m.grad_view(f"conv{i}_b"))
+            self._C.deep_remap_dw8(w.dw8[i], m.grad_view(f"conv{i}_w"),
+                                   st.k * st.k, st.cin, st.cout, st_h)
+        elif w.stage_implicit[i]:
+            self._C.deep_wgrad_gemm(x_in, w.acts[i],
+                                    m.grad_view(f"conv{i}_w"), M, st.kcp,
+                                    st.cout, ms, st_h, x_in, st.h, st.w,
+                                    st.cin, st.k, st.pad,
+                                    db=m.grad_view(f"conv{i}_b"))
+        else:
+            self._C.deep_wgrad_gemm(w.cols[i], w.acts[i],
+                                    m.grad_view(f"conv{i}_w"), M, st.kcp,
+                                    st.cout, ms, st_h,
+                                    db=m.grad_view(f"conv{i}_b"))
+
     def _hip_backward(self, x: torch.Tensor, B: int):
         m, w, spec = self.model, self.ws, self.model.spec
         st_h = native.current_stream_handle()
         nstage = len(spec.stages)
+        import os
+        async_wg = (os.environ.get("PCNN_DEEP_WGRAD_MODE", "async")
+                    == "async")
+        side = self._wgrad_side() if async_wg else None
+        main_s = torch.cuda.current_stream() if async_wg else None
+        if async_wg:
+            # fc wgrad moves to the side stream: it only needs dz and the
+            # last pool output (both final after the forward)
+            self._wev[nstage].record(main_s)
+            side.wait_event(self._wev[nstage])
+        side_h = side.cuda_stream if async_wg else st_h
         # fc backward-data already produced by the fused forward (dflat)
         # batch-slice count sized to ~4 WGs/CU (the owner-per-(k,m) grid
         # alone is only ~40 WGs for the 10x1024 head)
@@ -339,7 +391,7 @@ class DeepTrainer:
         fs = max(1, min(B, 1024 // bps))
         self._C.deep_fc_wgrad(w.dz, w.pouts[-1], m.grad_view("fc_w"),
                               m.grad_view("fc_b"), B, spec.fc_in,
-                              spec.n_classes, fs, st_h)
+                              spec.n_classes, fs, side_h)
         # pass 1 — the dependency CHAIN, descending: pool wgrad+bwd of
         # stage i (dapre in place over acts[i]), then dgrad into
         # dppre[i-1].  Conv wgrads move to pass 2: each only needs its
@@ -370,6 +422,12 @@ class DeepTrainer:
                                       st.h, st.w, st.cout, st.pool_k, st_h)
             dapre = w.acts[i]
             implicit = w.stage_implicit[i]
+            if async_wg:
+                # this stage's dapre is final: its weight grad can run on
+                # the side stream, concurrent with the rest of the chain
+                self._wev[i].record(main_s)
+                side.wait_event(self._wev[i])
+                self._launch_wgrad_stage(x, B, i, side.cuda_stream)
             if i > 0:
                 if implicit:
                     # dgrad-as-conv: implicit im2col of dapre against the
@@ -394,11 +452,16 @@ class DeepTrainer:
                                                w.dppre[i - 1], B, st.h,
                                                st.w, st.cin, st.k, st.pad,
                                                st.kcp, st_h)
-        # pass 2 — every stage's conv wgrad GEMM in ONE launch (the conv
-        # BIAS colsum stays folded in).  Three sequential ~500-WG
-        # launches each ran at ~2 WGs/CU with the per-iteration stall
-        # exposed; the combined grid fills the chip.  Measured MS optima
-        # per stage are unchanged (wgrad_ms).
+        if async_wg:
+            # all wgrads are queued on the side stream; rejoin before the
+            # all-reduce / update touches the gradient bucket
+            self._wdone.record(side)
+            main_s.wait_event(self._wdone)
+            return
+        # pass 2 (multi mode) — every stage's conv wgrad GEMM in ONE
+        # launch (the conv BIAS colsum stays folded in).  Three
+        # sequential ~500-WG launches each ran at ~2 WGs/CU with the
+        # per-iteration stall exposed; the combined grid fills the chip.
         da, dd, dwv, dbv = [], [], [], []
         Ml, Kl, Nl, MSl, Il, XHl, XWl, XCl, XKl, XPl = ([] for _ in range(10))
         for i, st in enumerate(spec.stages):
