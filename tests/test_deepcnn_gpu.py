@@ -330,3 +330,35 @@ def test_four_stage_hip_matches_oracle(device):
         ref.params += t.cfg.dt * (1.0 / 8) * grads
     diff = (t.model.params.cpu() - ref.params).abs().max().item()
     assert diff < 5e-3, diff
+
+
+def test_async_wgrad_mode_matches_multi(device):
+    """PCNN_DEEP_WGRAD_MODE=async (side-stream weight grads) must produce
+    the same trajectory as the default batched mode (subprocess so the
+    env is read at workspace construction)."""
+    import os
+    code = (
+        "import torch;"
+        "from parallel_cnn_amd.config import TrainConfig;"
+        "from parallel_cnn_amd.engine.deep import DeepTrainer;"
+        "from parallel_cnn_amd.data.mnist import synthetic_images;"
+        "cfg = TrainConfig(batch_size=16, device='cuda', backend='hip',"
+        "act_dtype='fp32', log_interval=0);"
+        "t = DeepTrainer(cfg);"
+        "x, y = synthetic_images(32, 32, 32, 3, seed=43);"
+        "t.step(*t.stage_batch(x[:16], y[:16]));"
+        "t.step(*t.stage_batch(x[16:], y[16:]));"
+        "torch.cuda.synchronize();"
+        "print(float(t.model.params.double().abs().sum()),"
+        "float(t.model.params[:100].double().sum()))"
+    )
+    outs = {}
+    for mode in ("multi", "async"):
+        r = subprocess.run([sys.executable, "-c", code],
+                           capture_output=True, text=True, timeout=300,
+                           env={**os.environ,
+                                "PCNN_DEEP_WGRAD_MODE": mode})
+        assert r.returncode == 0, r.stdout + r.stderr
+        outs[mode] = [float(v) for v in r.stdout.split()]
+    for a, b in zip(outs["multi"], outs["async"]):
+        assert abs(a - b) < 1e-3 * max(1.0, abs(a)), outs

@@ -39,6 +39,7 @@ KERNEL_MAP = [
     # DeepCNN path
     ("k_gemm_smallk", "conv fwd (stage 0, small-K GEMM)", ""),
     ("k_gemm<", "conv fwd GEMM + dgrad-as-conv", "implicit im2col"),
+    ("k_wgrad_multi", "ALL conv weight grads (one launch)", ""),
     ("k_wgrad_gemm", "conv weight grad (+bias colsum)", ""),
     ("k_split_epi", "GEMM split-K combine + epilogues", ""),
     ("k_im2col", "im2col (stage 0)", ""),
