@@ -155,7 +155,7 @@ def main() -> int:
         "data": "synthetic",
         "config": {
             "model": ("DeepCNN 32x32x3 (3x(conv5x5+trainable-pool2x2) -> "
-                      "fc1024x10), im2col+MFMA GEMM path"
+                      "fc1024x10), implicit-im2col+MFMA GEMM path"
                       if args.model == "deepcnn" else
                       "LeNet-5 28x28x1 (conv6x5x5 -> trainable-pool4x4 -> fc216x10)"),
             "global_batch": global_batch,

@@ -3,15 +3,17 @@
 Backends: "hip" (gfx950 kernels, csrc/hip/conv_kernels.hip) and "torchref"
 (the fp32 oracle, also the CPU execution path for this family).
 
-Step sequence (hip, deep_implicit=True — the default): one fused weight
-cast; per stage {implicit-im2col GEMM+bias+sigmoid (stage 0 with Cin=3
-materializes cols), pool}; fc fwd (+ residual loss); backward {fc
-bwd/wgrad; per stage: pool wgrad, pool bwd (in-place over the conv
-activation), implicit wgrad GEMM with the bias colsum folded in, and for
-i>0 an implicit dgrad-as-conv GEMM (rotated weight image, sigmoid-bwd
-epilogue) writing dppre[i-1] directly}; one fused DP all-reduce of the
-flat gradient bucket; SGD update.  deep_implicit=False keeps the round-1
-materialized-cols path (dgrad into dcols + col2im).
+Step sequence (hip, deep_implicit=True — the default), 15 kernels:
+per stage {implicit-im2col GEMM + bias + sigmoid + fused trainable-pool
+epilogue (stage 0 with Cin=3 materializes cols; under-filled grids
+split-K with a deterministic combine)}; fc fwd (+ residual loss + fused
+fc backward-data); backward {fc wgrad; the dependency chain of fused
+pool wgrad+bwd (dapre in place) and implicit dgrad-as-conv GEMMs
+(rotated weight image, sigmoid-bwd epilogue, writing dppre[i-1]
+directly); then EVERY stage's conv wgrad GEMM in ONE k_wgrad_multi
+launch with the bias colsums folded in}; one fused DP all-reduce of the
+flat gradient bucket; SGD update fused with the next step's weight cast.
+deep_implicit=False keeps the round-1 materialized-cols path.
 """
 from __future__ import annotations
 
