@@ -34,12 +34,15 @@ class DeepWorkspace:
     def wgrad_ms(st, M: int) -> int:
         """M-slice count for the weight-grad GEMM.  Round 1 measured ~512
         total WGs optimal; after the db fusion + hoisted implicit decode
-        the per-WG fixed cost shrank — but ~512 total WGs REMAINS the
-        measured optimum (1024 lost 8% at bs=64, 2048 lost 19%: the
-        atomic-combine tail and per-WG prologue beat the extra
-        residency).  PCNN_WGRAD_WGS overrides for sweeps."""
+        the per-WG fixed cost shrank, and under the BATCHED k_wgrad_multi
+        launch the optimum is batch-dependent: ~256 per-stage target WGs
+        at small batches (341k vs 332k at bs=64 — the combined grid
+        already fills), ~512 at bs>=256 (667k vs 617k).
+        PCNN_WGRAD_WGS overrides for sweeps."""
         import os
-        target = int(os.environ.get("PCNN_WGRAD_WGS", "512"))
+        B = max(1, M // (st.h * st.w))
+        default = "256" if B <= 128 else "512"
+        target = int(os.environ.get("PCNN_WGRAD_WGS", default))
         ktiles = (st.kcp + 63) // 64
         ntiles = (st.cout + 63) // 64
         return max(1, min(256, target // (ktiles * ntiles), M // 64))
