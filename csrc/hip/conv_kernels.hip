@@ -296,7 +296,7 @@ __device__ __forceinline__ void im2col8f(const act_t* __restrict__ x, int H,
 // ---------------------------------------------------------------------------
 
 constexpr int BM = 64, BN = 64, BK = 64;   // wgrad M-chunk per barrier pair
-constexpr int BKC = 64;  // k_gemm K-step (measured: 128 halves the
+constexpr int BKC = 32;  // k_gemm K-step (measured: 128 halves the
                          // iteration count but costs 187 VGPRs -> occupancy
                          // 2 waves/SIMD and loses 15% end-to-end; 64 keeps
                          // 68 VGPRs / occupancy 5)
@@ -397,7 +397,8 @@ __global__ __launch_bounds__(256) void k_gemm(
   // registers BEFORE the MFMA block; the LDS write happens after the read
   // barrier.
   float ra[NCH][8];
-  float rb[NCHB][8];
+  float rb[(NCHB > 2 ? NCHB : 2)][8];  // the b_kxn staging needs 2 chunks
+                                       // even when BKC < 64
   const long long m_a = m0 + row_a;
   // implicit-A: decode this thread's im2col row position once (from the
   // CLAMPED row — loads are unconditional, so the address must stay
@@ -452,10 +453,12 @@ __global__ __launch_bounds__(256) void k_gemm(
       }
     } else if (b_kxn) {
       // Bsrc[K][N]: read rows k (coalesced along n), write transposed;
-      // each thread covers BKC/64 k-rows 64 apart
+      // each thread covers ceil(BKC/64) k-rows 64 apart (rows >= BKC
+      // skipped when BKC < 64)
 #pragma unroll
-      for (int kb = 0; kb < BKC / 64; ++kb) {
+      for (int kb = 0; kb < (BKC + 63) / 64; ++kb) {
         const int k = (tid >> 2) + kb * 64;
+        if (k >= BKC) continue;
         const int nq = (tid & 3) * 16;
         const bool kok = (kt + k) < K;
 #pragma unroll
@@ -498,8 +501,9 @@ __global__ __launch_bounds__(256) void k_gemm(
     for (int h = 0; h < NCH; ++h) st8(&L.As[row_a][kq + h * 8], ra[h]);
     if (Bpre == nullptr && b_kxn) {
 #pragma unroll
-      for (int kb = 0; kb < BKC / 64; ++kb) {
+      for (int kb = 0; kb < (BKC + 63) / 64; ++kb) {
         const int k = (tid >> 2) + kb * 64;
+        if (k >= BKC) continue;
         const int nq = (tid & 3) * 16;
 #pragma unroll
         for (int h = 0; h < 2; ++h)
