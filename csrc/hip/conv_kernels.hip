@@ -789,15 +789,7 @@ __device__ __forceinline__ void wgrad_body(
     int XK, int XP, float* __restrict__ db, unsigned long long fd_cin,
     unsigned long long fd_k, unsigned long long fd_xw,
     unsigned long long fd_xh) {
-  // WBK=32 M-chunks: half the LDS of the BK=64 shape (20.5 KB vs 37)
-  // raises residency 4 -> 7 WGs/CU — the co-resident waves cover the
-  // per-chunk stage/barrier stall (same lever as k_gemm's BKC=32)
-  constexpr int WBK = 32;
-  struct WgLds {
-    __bf16 As[64][WBK + 8];
-    __bf16 Bs[64][WBK + 8];
-  };
-  __shared__ WgLds Lb[2];  // double-buffered M-chunks
+  __shared__ GemmLds Lb[2];  // double-buffered M-chunks
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -818,32 +810,35 @@ __device__ __forceinline__ void wgrad_body(
 #pragma unroll
   for (int f = 0; f < BN / 16; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
 
-  const int row_s = tid >> 3;        // source row m (32 rows, 8 thr each)
-  const int cq = (tid & 7) * 8;      // 8 columns per thread (one chunk)
+  const int row_s = tid >> 2;        // source row m (64 rows, 4 thr each)
+  const int cq = (tid & 3) * 16;     // 16 columns per thread
   // T14-style pipeline (see k_gemm): prefetch the next M-chunk into
   // registers under the MFMAs, write to LDS after the read barrier.
-  float rc[8];  // cols prefetch
-  float rd[8];  // dpre prefetch
+  float rc[2][8];  // cols prefetch
+  float rd[2][8];  // dpre prefetch
 
   const int iKc = XK * XK * XC;
-  // implicit-A: this thread's kc span is FIXED — hoist the (i, j, ci)
-  // decode out of the M-walk (two magic divisions, once)
-  int hi_i = 0, hi_j = 0, hi_ci = 0;
-  bool hi_in = false;
+  // implicit-A: this thread's kc spans are FIXED — hoist the (i, j, ci)
+  // decode out of the M-walk (two magic divisions per span, once)
+  int hi_i[2], hi_j[2], hi_ci[2];
+  bool hi_in[2];
   if (imx != nullptr) {
-    const int kc = kc0 + cq;
-    const unsigned p = fdiv((unsigned)kc, fd_cin);
-    hi_ci = kc - (int)p * XC;
-    const unsigned pi = fdiv(p, fd_k);
-    hi_j = (int)(p - pi * (unsigned)XK);
-    hi_i = (int)pi;
-    hi_in = kc < iKc;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int kc = kc0 + cq + h * 8;
+      const unsigned p = fdiv((unsigned)kc, fd_cin);
+      hi_ci[h] = kc - (int)p * XC;
+      const unsigned pi = fdiv(p, fd_k);
+      hi_j[h] = (int)(p - pi * (unsigned)XK);
+      hi_i[h] = (int)pi;
+      hi_in[h] = kc < iKc;
+    }
   }
-  // fused colsum partials: thread owns columns [n0+cq, +8)
-  float dbacc[8];
+  // fused colsum partials: thread owns columns [n0+cq, +16)
+  float dbacc[16];
   const bool do_db = db != nullptr && kct == 0;
 #pragma unroll
-  for (int u = 0; u < 8; ++u) dbacc[u] = 0.f;
+  for (int u = 0; u < 16; ++u) dbacc[u] = 0.f;
 
   auto load_regs = [&](long long mt) {
     const long long m = mt + row_s;
@@ -861,35 +856,36 @@ __device__ __forceinline__ void wgrad_body(
       ioh = (int)(bhq - bq * (unsigned)XH);
       ib = (int)bq;
     }
-    {
-      const int c = cq;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int c = cq + h * 8;
       if (imx != nullptr) {
-        const int ih = ioh + hi_i - XP;
-        const int iw = iow + hi_j - XP;
+        const int ih = ioh + hi_i[h] - XP;
+        const int iw = iow + hi_j[h] - XP;
         const int ihc = min(max(ih, 0), XH - 1);
         const int iwc = min(max(iw, 0), XW - 1);
-        ld8v(imx + (((long long)ib * XH + ihc) * XW + iwc) * XC + hi_ci,
-             rc);
-        if (!(ok && hi_in && ih == ihc && iw == iwc)) {
+        ld8v(imx + (((long long)ib * XH + ihc) * XW + iwc) * XC + hi_ci[h],
+             rc[h]);
+        if (!(ok && hi_in[h] && ih == ihc && iw == iwc)) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) rc[u] = 0.f;
+          for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
         }
       } else {
         const bool cok = (kc0 + c + 8) <= KcP;
         const int ccl = cok ? kc0 + c : KcP - 8;
-        ld8v(cols + mm * KcP + ccl, rc);
+        ld8v(cols + mm * KcP + ccl, rc[h]);
         if (!(ok && cok)) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) rc[u] = 0.f;
+          for (int u = 0; u < 8; ++u) rc[h][u] = 0.f;
         }
       }
       {
         const bool nok = (n0 + c + 8) <= N;
         const int ncl = nok ? n0 + c : N - 8;
-        ld8v(dpre + mm * N + ncl, rd);
+        ld8v(dpre + mm * N + ncl, rd[h]);
         if (!(ok && nok)) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) rd[u] = 0.f;
+          for (int u = 0; u < 8; ++u) rd[h][u] = 0.f;
         }
       }
     }
@@ -902,34 +898,37 @@ __device__ __forceinline__ void wgrad_body(
   // index spreads the writes; fragment reads stay single 16B ds_reads
   // (group-granular swizzle).
   auto sw = [](int k, int m) {
-    // group swizzle over the 4 m-groups of the 32-wide chunk: the
-    // simultaneous writers (same m, k stride 8) fold k>>3 and k>>5 so
-    // they spread across groups; residual 2-way is inherent (8 writers,
-    // 4 groups)
-    return ((((m >> 3) ^ (k >> 3) ^ (k >> 5)) & 3)) * 8 + (m & 7);
+    // fold BOTH k&7 and k>>3 into the group swizzle: the simultaneous
+    // writers share k mod 8 (their k differ by 16), so low bits alone
+    // would leave them on one bank
+    return (((m >> 3) ^ (k & 7) ^ ((k >> 3) & 7)) & 7) * 8 + (m & 7);
   };
-  auto write_lds = [&](WgLds& L) {
-    const int c = cq;
+  auto write_lds = [&](GemmLds& L) {
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      L.As[c + u][sw(c + u, row_s)] = (__bf16)rc[u];
-      L.Bs[c + u][sw(c + u, row_s)] = (__bf16)rd[u];
-      if (do_db) dbacc[u] += rd[u];
+    for (int h = 0; h < 2; ++h) {
+      const int c = cq + h * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        L.As[c + u][sw(c + u, row_s)] = (__bf16)rc[h][u];
+        L.Bs[c + u][sw(c + u, row_s)] = (__bf16)rd[h][u];
+        if (do_db) dbacc[h * 8 + u] += rd[h][u];
+      }
     }
   };
 
   // Double-buffered pipeline, one barrier per M-chunk (see k_gemm)
   load_regs(m_lo);
   write_lds(Lb[0]);
-  if (m_lo + WBK < m_hi) load_regs(m_lo + WBK);
+  if (m_lo + BK < m_hi) load_regs(m_lo + BK);
   __syncthreads();
   int p = 0;
-  for (long long mt = m_lo; mt < m_hi; mt += WBK, p ^= 1) {
+  for (long long mt = m_lo; mt < m_hi; mt += BK, p ^= 1) {
     auto& L = Lb[p];
-    {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
       const int ar = wv * 16 + (lane & 15);
-      const int m8 = (lane >> 4) * 8;  // 8-aligned: sw() keeps the span
-                                       // one 16B read
+      const int m8 = kk * 32 + (lane >> 4) * 8;  // 8-aligned: sw() keeps
+                                                 // the span one 16B read
       const bf16x8 a0 = frag_from_lds(L.As[ar], sw(ar, m8));
 #pragma unroll
       for (int f = 0; f < BN / 16; ++f) {
@@ -941,9 +940,9 @@ __device__ __forceinline__ void wgrad_body(
         }
       }
     }
-    if (mt + WBK < m_hi) {
+    if (mt + BK < m_hi) {
       write_lds(Lb[p ^ 1]);
-      if (mt + 2 * WBK < m_hi) load_regs(mt + 2 * WBK);
+      if (mt + 2 * BK < m_hi) load_regs(mt + 2 * BK);
     }
     __syncthreads();
   }
@@ -952,13 +951,13 @@ __device__ __forceinline__ void wgrad_body(
   // 64 staging rows (LDS reuse of the now-idle A tile) and combine into
   // db with one atomic per column per (kct==0, slice) WG.
   if (do_db) {
-    float* sc = reinterpret_cast<float*>(&Lb[0]);  // [32 rows][64 cols] fp32
+    float* sc = reinterpret_cast<float*>(&Lb[0]);  // [64 rows][64 cols] fp32
 #pragma unroll
-    for (int u = 0; u < 8; ++u) sc[row_s * 64 + cq + u] = dbacc[u];
+    for (int u = 0; u < 16; ++u) sc[row_s * 64 + cq + u] = dbacc[u];
     __syncthreads();
     if (tid < 64 && (n0 + tid) < N) {
       float s = 0.f;
-      for (int r = 0; r < 32; ++r) s += sc[r * 64 + tid];
+      for (int r = 0; r < 64; ++r) s += sc[r * 64 + tid];
       if (s != 0.f) unsafeAtomicAdd(&db[n0 + tid], s);
     }
     __syncthreads();  // scratch reads done before any later reuse
