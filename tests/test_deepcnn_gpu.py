@@ -362,3 +362,14 @@ def test_async_wgrad_mode_matches_multi(device):
         outs[mode] = [float(v) for v in r.stdout.split()]
     for a, b in zip(outs["multi"], outs["async"]):
         assert abs(a - b) < 1e-3 * max(1.0, abs(a)), outs
+
+
+def test_fuzz_corner_configs(device):
+    """Four fixed corner configs through tools/fuzz_deep (random channel
+    sets, tiny batches, both engine modes) — the broader sweep lives in
+    tools/, this keeps a slice of it in the suite."""
+    out = subprocess.run(
+        [sys.executable, "tools/fuzz_deep.py", "--n", "4", "--seed", "11"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "4/4 configs passed" in out.stdout
