@@ -3,7 +3,7 @@
 Backends: "hip" (gfx950 kernels, csrc/hip/conv_kernels.hip) and "torchref"
 (the fp32 oracle, also the CPU execution path for this family).
 
-Step sequence (hip, deep_implicit=True — the default), 15 kernels:
+Step sequence (hip, deep_implicit=True — the default), 17 launches:
 per stage {implicit-im2col GEMM + bias + sigmoid + fused trainable-pool
 epilogue (stage 0 with Cin=3 materializes cols; under-filled grids
 split-K with a deterministic combine)}; fc fwd (+ residual loss + fused
